@@ -1,0 +1,117 @@
+"""Flagship benchmark: INTELLECT-1 10B DiLoCo H=100 training throughput.
+
+Driver contract: `python bench.py --gpus N --steps K --warmup W` (N>1 via
+torch.distributed.run, one rank per GPU over RCCL). Each rank is one DiLoCo
+worker (worker_size=1, weak scaling: fixed per-GPU batch); the outer int8
+ring all-reduce runs every H=100 inner steps (and once during warmup so the
+path is exercised even when K < H). Rank 0 prints ONE JSON line with the
+whole-job tokens/sec.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--model", default="intellect_10b")
+    ap.add_argument("--seq-len", type=int, default=2048)
+    ap.add_argument("--micro-batch", type=int, default=8)
+    ap.add_argument("--h", type=int, default=100, help="DiLoCo inner steps per outer sync")
+    ap.add_argument("--no-outer-warmup", action="store_true")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    n_gpus = world if world > 1 else args.gpus
+    if world == 1:
+        os.environ.setdefault("RANK", "0")
+        os.environ.setdefault("WORLD_SIZE", "1")
+    rank = int(os.environ.get("RANK", 0))
+
+    from prime_amd.train import Trainer
+    from prime_amd.utils.config import (
+        DataSection, DilocoConfig, MetricsConfig, ModelConfig, TrainConfig,
+    )
+    from prime_amd.utils.metrics import mfu as mfu_of
+
+    cfg = TrainConfig(
+        run_name="bench",
+        steps=args.steps,
+        model=ModelConfig(
+            name=args.model, seq_len=args.seq_len,
+            activation_checkpointing=True,
+        ),
+        data=DataSection(kind="synthetic", micro_batch_size=args.micro_batch),
+        diloco=DilocoConfig(H=args.h, quant_int8=True, outer_device="auto"),
+        metrics=MetricsConfig(log_interval=10**9),
+    )
+    trainer = Trainer(cfg, run_dir=os.environ.get("BENCH_RUN_DIR", "/tmp/prime_amd_bench"))
+    dev = trainer.device
+
+    # ---- warmup (incl. one outer sync so ring/host-offload path is hot)
+    for _ in range(args.warmup):
+        trainer.train_step()
+    if not args.no_outer_warmup:
+        trainer.diloco.outer_step()
+    trainer.mesh.barrier()
+    if dev.type == "cuda":
+        torch.cuda.synchronize()
+
+    # ---- timed region
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        trainer.train_step()
+    if dev.type == "cuda":
+        torch.cuda.synchronize()
+    trainer.mesh.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if trainer.mesh.initialized:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], device=dev if dev.type == "cuda" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t)
+
+    tokens_per_step_per_gpu = args.micro_batch * args.seq_len
+    total_tokens = tokens_per_step_per_gpu * args.steps * n_gpus
+    tps = total_tokens / elapsed
+    tps_gpu = tps / n_gpus
+    mfu = mfu_of(tps_gpu, trainer.flops_per_token)
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "tokens_per_sec",
+            "value": tps,
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": 1000.0 * elapsed / args.steps,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "mfu": mfu,
+            "config": {
+                "model": args.model,
+                "global_batch": args.micro_batch * n_gpus,
+                "seq_len": args.seq_len,
+                "parallelism": f"diloco{n_gpus}_h{args.h}_int8ring",
+            },
+        }), flush=True)
+    trainer.close()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,143 @@
+"""Async distributed checkpointing.
+
+GPU path: device tensors are staged into reusable PINNED host buffers with
+hipMemcpyAsync on a dedicated side stream (no stall of the compute stream),
+then a background thread waits on the HIP event and persists with
+torch.save. Checkpoints are taken at outer-step boundaries so every worker
+snapshot is outer-consistent; retention keeps the newest `keep` tags.
+
+Layout:  <path>/step_<outer>/worker<id>.pt   + meta.json
+         <path>/latest -> step_<outer>       (symlink)
+Each DiLoCo worker's leader rank persists {master32/theta_outer/outer_buf,
+inner m/v/step, dataloader, config}; non-leader ranks of a sharded worker
+persist their own shard file.
+"""
+from __future__ import annotations
+
+import json
+import os
+import shutil
+import threading
+import time
+from pathlib import Path
+
+import torch
+
+
+class _PinnedStager:
+    """Reusable pinned-host mirror of a dict of device tensors."""
+
+    def __init__(self):
+        self.buffers: dict[str, torch.Tensor] = {}
+
+    def stage(self, state: dict[str, torch.Tensor], stream: torch.cuda.Stream) -> dict:
+        out = {}
+        with torch.cuda.stream(stream):
+            for k, t in state.items():
+                buf = self.buffers.get(k)
+                if buf is None or buf.shape != t.shape or buf.dtype != t.dtype:
+                    buf = torch.empty_like(t, device="cpu", pin_memory=True)
+                    self.buffers[k] = buf
+                buf.copy_(t, non_blocking=True)
+                out[k] = buf
+        return out
+
+
+class CheckpointManager:
+    def __init__(self, path: str | Path, keep: int = 3, async_save: bool = True,
+                 worker_id: int = 0, is_leader: bool = True):
+        self.root = Path(path)
+        self.keep = keep
+        self.async_save = async_save
+        self.worker_id = worker_id
+        self.is_leader = is_leader
+        self._stager = _PinnedStager()
+        self._stream = None
+        self._pending: threading.Thread | None = None
+        self.root.mkdir(parents=True, exist_ok=True)
+
+    # ------------------------------------------------------------- save
+    def save(self, outer_step: int, tensors: dict[str, torch.Tensor],
+             meta: dict) -> None:
+        """Snapshot `tensors` (+ JSON-serializable `meta`)."""
+        self.wait()
+        tag_dir = self.root / f"step_{outer_step}"
+        tag_dir.mkdir(parents=True, exist_ok=True)
+        fname = tag_dir / f"worker{self.worker_id}.pt"
+
+        on_gpu = any(t.is_cuda for t in tensors.values())
+        if on_gpu and self.async_save:
+            if self._stream is None:
+                self._stream = torch.cuda.Stream()
+            self._stream.wait_stream(torch.cuda.current_stream())
+            host = self._stager.stage(
+                {k: t for k, t in tensors.items() if t.is_cuda}, self._stream
+            )
+            host.update({k: t for k, t in tensors.items() if not t.is_cuda})
+            ev = torch.cuda.Event()
+            ev.record(self._stream)
+
+            def _persist():
+                ev.synchronize()
+                torch.save({"tensors": host, "meta": meta}, fname)
+                self._finalize(tag_dir, outer_step, meta)
+
+            self._pending = threading.Thread(target=_persist, daemon=True)
+            self._pending.start()
+        else:
+            host = {k: t.detach().cpu() for k, t in tensors.items()}
+            torch.save({"tensors": host, "meta": meta}, fname)
+            self._finalize(tag_dir, outer_step, meta)
+
+    def _finalize(self, tag_dir: Path, outer_step: int, meta: dict) -> None:
+        if self.is_leader and self.worker_id == 0:
+            (tag_dir / "meta.json").write_text(
+                json.dumps({"outer_step": outer_step, "time": time.time(), **{
+                    k: v for k, v in meta.items()
+                    if isinstance(v, (int, float, str, bool, type(None)))
+                }})
+            )
+            latest = self.root / "latest"
+            tmp = self.root / ".latest.tmp"
+            if tmp.is_symlink() or tmp.exists():
+                tmp.unlink()
+            tmp.symlink_to(tag_dir.name)
+            os.replace(tmp, latest)
+            self._retain()
+
+    def _retain(self) -> None:
+        tags = sorted(
+            (d for d in self.root.iterdir() if d.is_dir() and d.name.startswith("step_")),
+            key=lambda d: int(d.name.split("_")[1]),
+        )
+        for d in tags[: -self.keep] if self.keep > 0 else []:
+            shutil.rmtree(d, ignore_errors=True)
+
+    def wait(self) -> None:
+        if self._pending is not None:
+            self._pending.join()
+            self._pending = None
+
+    # ------------------------------------------------------------- load
+    def latest_tag(self) -> Path | None:
+        latest = self.root / "latest"
+        if latest.exists():
+            return latest.resolve()
+        tags = sorted(
+            (d for d in self.root.iterdir() if d.is_dir() and d.name.startswith("step_")),
+            key=lambda d: int(d.name.split("_")[1]),
+        )
+        return tags[-1] if tags else None
+
+    def load(self, tag: str | Path | None = None, map_location="cpu") -> dict | None:
+        tag_dir = Path(tag) if tag else self.latest_tag()
+        if tag_dir is None or not tag_dir.exists():
+            return None
+        fname = tag_dir / f"worker{self.worker_id}.pt"
+        if not fname.exists():
+            # joining worker: adopt any peer's snapshot (live recovery fallback)
+            cands = sorted(tag_dir.glob("worker*.pt"))
+            if not cands:
+                return None
+            fname = cands[0]
+        return torch.load(fname, map_location=map_location, weights_only=False)

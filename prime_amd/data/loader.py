@@ -1,0 +1,93 @@
+"""Data pipeline: synthetic token stream (benchmarks — no network in the
+build/bench environment) and a memory-mapped token-file dataset for real
+runs. Both are stateful (resumable from checkpoints) and shard by
+(worker_id, n_workers, local DP rank) so every DiLoCo worker sees a
+disjoint stream.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from pathlib import Path
+
+import numpy as np
+import torch
+
+
+@dataclass
+class DataConfig:
+    kind: str = "synthetic"        # synthetic | token_file
+    path: str | None = None        # .bin/.npy uint16|uint32 token file
+    seq_len: int = 2048
+    micro_batch_size: int = 4
+    seed: int = 1234
+
+
+class SyntheticTokens:
+    """Deterministic pseudo-random token batches; infinite, resumable."""
+
+    def __init__(self, cfg: DataConfig, vocab_size: int, shard: int, n_shards: int):
+        self.cfg = cfg
+        self.vocab = vocab_size
+        self.shard = shard
+        self.n_shards = n_shards
+        self.batch_idx = 0
+
+    def state_dict(self) -> dict:
+        return {"batch_idx": self.batch_idx}
+
+    def load_state_dict(self, sd: dict) -> None:
+        self.batch_idx = int(sd["batch_idx"])
+
+    def next_batch(self, device: torch.device) -> tuple[torch.Tensor, torch.Tensor]:
+        g = torch.Generator()
+        g.manual_seed(self.cfg.seed + self.batch_idx * self.n_shards + self.shard)
+        self.batch_idx += 1
+        b, s = self.cfg.micro_batch_size, self.cfg.seq_len
+        toks = torch.randint(0, self.vocab, (b, s + 1), generator=g)
+        x = toks[:, :-1].to(device, non_blocking=True)
+        y = toks[:, 1:].to(device, non_blocking=True)
+        return x, y
+
+
+class TokenFileDataset:
+    """Memory-mapped flat token file; strided contiguous windows per shard."""
+
+    def __init__(self, cfg: DataConfig, vocab_size: int, shard: int, n_shards: int):
+        if not cfg.path:
+            raise ValueError("token_file dataset requires data.path")
+        p = Path(cfg.path)
+        dtype = np.uint16 if vocab_size <= 65535 else np.uint32
+        self.tokens = np.memmap(p, dtype=dtype, mode="r")
+        self.cfg = cfg
+        self.shard = shard
+        self.n_shards = n_shards
+        self.batch_idx = 0
+        self.windows = (len(self.tokens) - 1) // cfg.seq_len
+
+    def state_dict(self) -> dict:
+        return {"batch_idx": self.batch_idx}
+
+    def load_state_dict(self, sd: dict) -> None:
+        self.batch_idx = int(sd["batch_idx"])
+
+    def next_batch(self, device: torch.device) -> tuple[torch.Tensor, torch.Tensor]:
+        b, s = self.cfg.micro_batch_size, self.cfg.seq_len
+        xs, ys = [], []
+        for i in range(b):
+            w = (self.batch_idx * self.n_shards * b + self.shard * b + i) % self.windows
+            start = w * s
+            chunk = torch.from_numpy(self.tokens[start : start + s + 1].astype(np.int64))
+            xs.append(chunk[:-1])
+            ys.append(chunk[1:])
+        self.batch_idx += 1
+        x = torch.stack(xs).to(device, non_blocking=True)
+        y = torch.stack(ys).to(device, non_blocking=True)
+        return x, y
+
+
+def build_dataloader(cfg: DataConfig, vocab_size: int, shard: int, n_shards: int):
+    if cfg.kind == "synthetic":
+        return SyntheticTokens(cfg, vocab_size, shard, n_shards)
+    if cfg.kind == "token_file":
+        return TokenFileDataset(cfg, vocab_size, shard, n_shards)
+    raise ValueError(f"unknown data.kind '{cfg.kind}'")

@@ -1,0 +1,140 @@
+"""Llama-family transformer assembled from the gfx950 fused ops.
+
+MI355X-first design decisions:
+  - qkv and gate/up projections are each ONE hipBLASLt GEMM (fused weights);
+    everything between GEMMs (RMSNorm, RoPE, flash attention, SwiGLU,
+    cross-entropy) is a hand-written HIP kernel (prime_amd.ops).
+  - bf16 parameters/activations, fp32 only where numerics need it
+    (norm stats, softmax, loss).
+  - per-layer activation checkpointing (non-reentrant) for the big configs:
+    288 GB HBM3E prefers recompute over host offload.
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+from torch.utils.checkpoint import checkpoint
+
+from .. import ops
+from .configs import LlamaConfig, get_config
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, dim: int, eps: float):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.eps = eps
+
+    def forward(self, x):
+        return ops.rmsnorm(x, self.weight, self.eps)
+
+
+class Attention(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        hd = cfg.head_dim
+        self.wqkv = nn.Linear(cfg.dim, (cfg.n_heads + 2 * cfg.n_kv_heads) * hd, bias=False)
+        self.wo = nn.Linear(cfg.n_heads * hd, cfg.dim, bias=False)
+
+    def forward(self, x, cos, sin):
+        B, S, _ = x.shape
+        cfg = self.cfg
+        hd = cfg.head_dim
+        qkv = self.wqkv(x)
+        q, k, v = qkv.split(
+            [cfg.n_heads * hd, cfg.n_kv_heads * hd, cfg.n_kv_heads * hd], dim=-1
+        )
+        q = q.view(B, S, cfg.n_heads, hd)
+        k = k.view(B, S, cfg.n_kv_heads, hd)
+        v = v.view(B, S, cfg.n_kv_heads, hd)
+        q = ops.apply_rope(q, cos, sin)
+        k = ops.apply_rope(k, cos, sin)
+        o = ops.flash_attention(q, k, v, causal=True)
+        return self.wo(o.reshape(B, S, cfg.n_heads * hd))
+
+
+class MLP(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.w_gateup = nn.Linear(cfg.dim, 2 * cfg.intermediate, bias=False)
+        self.w_down = nn.Linear(cfg.intermediate, cfg.dim, bias=False)
+
+    def forward(self, x):
+        return self.w_down(ops.swiglu(self.w_gateup(x)))
+
+
+class Block(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.attn_norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.attn = Attention(cfg)
+        self.mlp_norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.mlp = MLP(cfg)
+
+    def forward(self, x, cos, sin):
+        x = x + self.attn(self.attn_norm(x), cos, sin)
+        x = x + self.mlp(self.mlp_norm(x))
+        return x
+
+
+class Llama(nn.Module):
+    def __init__(self, cfg: LlamaConfig, activation_checkpointing: bool = False):
+        super().__init__()
+        self.cfg = cfg
+        self.activation_checkpointing = activation_checkpointing
+        self.tok_embeddings = nn.Embedding(cfg.vocab_size, cfg.dim)
+        self.layers = nn.ModuleList(Block(cfg) for _ in range(cfg.n_layers))
+        self.norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+        if cfg.tie_embeddings:
+            self.lm_head.weight = self.tok_embeddings.weight
+        cos, sin = ops.reference.rope_tables(cfg.head_dim, cfg.max_seq, cfg.rope_theta)
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+        self.apply(self._init)
+        # scaled init for residual-out projections (GPT-2 style)
+        std = 0.02 / math.sqrt(2 * cfg.n_layers)
+        for blk in self.layers:
+            nn.init.normal_(blk.attn.wo.weight, std=std)
+            nn.init.normal_(blk.mlp.w_down.weight, std=std)
+
+    @staticmethod
+    def _init(m):
+        if isinstance(m, (nn.Linear, nn.Embedding)):
+            nn.init.normal_(m.weight, std=0.02)
+
+    def reset_rope(self, device=None) -> None:
+        """Recompute fp32 RoPE tables (call after .to(dtype=bf16), which
+        would otherwise truncate the tables to bf16)."""
+        cfg = self.cfg
+        cos, sin = ops.reference.rope_tables(
+            cfg.head_dim, cfg.max_seq, cfg.rope_theta,
+            device=device or self.rope_cos.device,
+        )
+        self.rope_cos = cos
+        self.rope_sin = sin
+
+    def forward(self, tokens: torch.Tensor) -> torch.Tensor:
+        """tokens [B,S] -> hidden states [B,S,dim] (pre-lm_head)."""
+        x = self.tok_embeddings(tokens)
+        cos, sin = self.rope_cos, self.rope_sin
+        for blk in self.layers:
+            if self.activation_checkpointing and self.training:
+                x = checkpoint(blk, x, cos, sin, use_reentrant=False)
+            else:
+                x = blk(x, cos, sin)
+        return self.norm(x)
+
+    def loss(self, tokens: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:
+        """Fused lm_head + cross-entropy. tokens/targets: [B,S]."""
+        h = self.forward(tokens)
+        logits = self.lm_head(h).flatten(0, 1)
+        return ops.cross_entropy(logits, targets.flatten(), ignore_index=-100)
+
+
+def build_model(name: str, activation_checkpointing: bool = False, **overrides) -> Llama:
+    return Llama(get_config(name, **overrides), activation_checkpointing)

@@ -1,0 +1,92 @@
+"""ctypes binding to the in-tree gfx950 kernel library (libprime_hip.so).
+
+Policy (matches the build contract): on a machine WITH a GPU the HIP library
+is mandatory — a missing/unbuildable .so raises instead of silently falling
+back to eager PyTorch. On CPU-only machines the ops use their PyTorch
+reference paths and the library is never required.
+"""
+from __future__ import annotations
+
+import ctypes
+from typing import Optional
+
+import torch
+
+from .build import LIB_PATH, build
+
+_LIB: Optional[ctypes.CDLL] = None
+_TRIED = False
+
+_SIGS = {
+    # name -> argtypes (all return c_int hipError_t)
+    "prime_rmsnorm_fwd": [ctypes.c_void_p] * 5 + [ctypes.c_int64] * 2 + [ctypes.c_double],
+    "prime_rmsnorm_bwd": [ctypes.c_void_p] * 7 + [ctypes.c_int64] * 2 + [ctypes.c_double],
+    "prime_rope": [ctypes.c_void_p] * 5 + [ctypes.c_int64] * 4 + [ctypes.c_int, ctypes.c_int64],
+    "prime_swiglu_fwd": [ctypes.c_void_p] * 3 + [ctypes.c_int64] * 2,
+    "prime_swiglu_bwd": [ctypes.c_void_p] * 4 + [ctypes.c_int64] * 2,
+    "prime_adamw": [ctypes.c_void_p] * 6 + [ctypes.c_int64] + [ctypes.c_double] * 5 + [ctypes.c_int64],
+    "prime_pseudograd": [ctypes.c_void_p] * 4 + [ctypes.c_int64],
+    "prime_quant_int8": [ctypes.c_void_p] * 4 + [ctypes.c_int64],
+    "prime_dequant_int8": [ctypes.c_void_p] * 4 + [ctypes.c_int64] * 2,
+    "prime_nesterov_outer": [ctypes.c_void_p] * 6 + [ctypes.c_int64] + [ctypes.c_double] * 2,
+    "prime_cross_entropy": [ctypes.c_void_p] * 5 + [ctypes.c_int64] * 2 + [ctypes.c_double] + [ctypes.c_int64] * 2,
+    "prime_flash_fwd": [ctypes.c_void_p] * 6 + [ctypes.c_int64] * 5 + [ctypes.c_double, ctypes.c_int64],
+    "prime_attn_delta": [ctypes.c_void_p] * 4 + [ctypes.c_int64] * 2,
+    "prime_flash_bwd_dq": [ctypes.c_void_p] * 9 + [ctypes.c_int64] * 5 + [ctypes.c_double, ctypes.c_int64],
+    "prime_flash_bwd_dkv": [ctypes.c_void_p] * 11 + [ctypes.c_int64] * 5 + [ctypes.c_double, ctypes.c_int64],
+    "prime_mfma_probe": [ctypes.c_void_p] * 4,
+}
+
+
+def _load() -> ctypes.CDLL:
+    if not LIB_PATH.exists():
+        # last resort: try building (hipcc cross-compiles without a GPU)
+        build(verbose=True)
+    lib = ctypes.CDLL(str(LIB_PATH))
+    for name, argtypes in _SIGS.items():
+        fn = getattr(lib, name)
+        fn.argtypes = argtypes
+        fn.restype = ctypes.c_int
+    return lib
+
+
+def lib() -> ctypes.CDLL:
+    """Return the kernel library, loading (and if needed building) it."""
+    global _LIB, _TRIED
+    if _LIB is None:
+        if _TRIED:
+            raise RuntimeError("prime_amd HIP kernel library failed to load earlier")
+        _TRIED = True
+        try:
+            _LIB = _load()
+        except Exception as e:  # noqa: BLE001
+            raise RuntimeError(
+                "prime_amd requires its HIP kernel library on GPU machines; "
+                f"build failed or .so missing at {LIB_PATH}: {e}"
+            ) from e
+    return _LIB
+
+
+def have_lib() -> bool:
+    try:
+        lib()
+        return True
+    except RuntimeError:
+        return False
+
+
+def stream_of(t: torch.Tensor) -> ctypes.c_void_p:
+    """Current HIP stream for the tensor's device, as a raw handle."""
+    s = torch.cuda.current_stream(t.device).cuda_stream
+    return ctypes.c_void_p(s)
+
+
+def check(ret: int, what: str) -> None:
+    if ret != 0:
+        raise RuntimeError(f"HIP kernel {what} failed with hipError_t={ret}")
+
+
+def ptr(t: Optional[torch.Tensor]) -> ctypes.c_void_p:
+    if t is None:
+        return ctypes.c_void_p(0)
+    return ctypes.c_void_p(t.data_ptr())

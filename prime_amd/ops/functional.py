@@ -1,0 +1,315 @@
+"""Autograd-wrapped fused ops.
+
+Dispatch rule: CUDA tensors run the hand-written gfx950 HIP kernels
+(mandatory — no silent eager fallback on GPU); CPU tensors run the fp32
+PyTorch reference (prime_amd/ops/reference.py) so the gloo plumbing config
+works without a GPU.
+"""
+from __future__ import annotations
+
+import torch
+
+from . import reference as ref
+from ._lib import check, lib, ptr, stream_of
+
+
+def _is_hip(t: torch.Tensor) -> bool:
+    return t.is_cuda
+
+
+# --------------------------------------------------------------- RMSNorm
+class _RMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, w: torch.Tensor, eps: float):
+        if not _is_hip(x):
+            ctx.save_for_backward(x, w)
+            ctx.eps = eps
+            return ref.rmsnorm(x, w, eps)
+        x = x.contiguous()
+        shp = x.shape
+        D = shp[-1]
+        R = x.numel() // D
+        y = torch.empty_like(x)
+        rstd = torch.empty(R, device=x.device, dtype=torch.float32)
+        check(
+            lib().prime_rmsnorm_fwd(stream_of(x), ptr(x), ptr(w), ptr(y), ptr(rstd), R, D, eps),
+            "rmsnorm_fwd",
+        )
+        ctx.save_for_backward(x, w, rstd)
+        ctx.eps = eps
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        if len(ctx.saved_tensors) == 2:  # CPU path
+            x, w = ctx.saved_tensors
+            x = x.detach().float().requires_grad_(True)
+            w2 = w.detach().float().requires_grad_(True)
+            with torch.enable_grad():
+                y = ref.rmsnorm(x, w2, ctx.eps)
+            gx, gw = torch.autograd.grad(y, [x, w2], dy.float())
+            return gx.to(dy.dtype), gw.to(w.dtype), None
+        x, w, rstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        D = x.shape[-1]
+        R = x.numel() // D
+        dx = torch.empty_like(x)
+        dw = torch.zeros(D, device=x.device, dtype=torch.float32)
+        check(
+            lib().prime_rmsnorm_bwd(
+                stream_of(x), ptr(dy), ptr(x), ptr(w), ptr(rstd), ptr(dx), ptr(dw), R, D, ctx.eps
+            ),
+            "rmsnorm_bwd",
+        )
+        return dx, dw.to(w.dtype), None
+
+
+def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    return _RMSNorm.apply(x, w, eps)
+
+
+# ------------------------------------------------------------------ RoPE
+class _Rope(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, cos, sin, pos_offset: int):
+        if not _is_hip(x):
+            ctx.cpu = (cos, sin, pos_offset)
+            return ref.apply_rope(x, cos, sin, pos_offset)
+        ctx.cpu = None
+        x = x.contiguous()
+        B, S, H, D = x.shape
+        y = torch.empty_like(x)
+        check(
+            lib().prime_rope(
+                stream_of(x), ptr(x), ptr(y), ptr(cos), ptr(sin), B * S * H, H, S, D, 0, pos_offset
+            ),
+            "rope_fwd",
+        )
+        ctx.save_for_backward(cos, sin)
+        ctx.meta = (B, S, H, D, pos_offset)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        if ctx.cpu is not None:
+            cos, sin, off = ctx.cpu
+            # inverse rotation
+            return ref.apply_rope(dy, cos, -sin, off), None, None, None
+        cos, sin = ctx.saved_tensors
+        B, S, H, D, off = ctx.meta
+        dy = dy.contiguous()
+        dx = torch.empty_like(dy)
+        check(
+            lib().prime_rope(
+                stream_of(dy), ptr(dy), ptr(dx), ptr(cos), ptr(sin), B * S * H, H, S, D, 1, off
+            ),
+            "rope_bwd",
+        )
+        return dx, None, None, None
+
+
+def apply_rope(x, cos, sin, pos_offset: int = 0):
+    """x: [B,S,H,D] bf16; cos/sin: [S_max, D/2] fp32 tables."""
+    return _Rope.apply(x, cos, sin, pos_offset)
+
+
+# ---------------------------------------------------------------- SwiGLU
+class _SwiGLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gu):
+        if not _is_hip(gu):
+            ctx.save_for_backward(gu)
+            return ref.swiglu(gu)
+        gu = gu.contiguous()
+        I = gu.shape[-1] // 2
+        R = gu.numel() // (2 * I)
+        out = torch.empty(*gu.shape[:-1], I, device=gu.device, dtype=gu.dtype)
+        check(lib().prime_swiglu_fwd(stream_of(gu), ptr(gu), ptr(out), R, I), "swiglu_fwd")
+        ctx.save_for_backward(gu)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        (gu,) = ctx.saved_tensors
+        if not _is_hip(gu):
+            g = gu.detach().float().requires_grad_(True)
+            with torch.enable_grad():
+                y = ref.swiglu(g)
+            (gx,) = torch.autograd.grad(y, [g], dout.float())
+            return gx.to(gu.dtype)
+        dout = dout.contiguous()
+        I = gu.shape[-1] // 2
+        R = gu.numel() // (2 * I)
+        dgu = torch.empty_like(gu)
+        check(lib().prime_swiglu_bwd(stream_of(gu), ptr(dout), ptr(gu), ptr(dgu), R, I), "swiglu_bwd")
+        return dgu
+
+
+def swiglu(gu: torch.Tensor) -> torch.Tensor:
+    """gu: [..., 2I] (gate ‖ up) -> [..., I] = silu(gate) * up."""
+    return _SwiGLU.apply(gu)
+
+
+# ------------------------------------------------------- flash attention
+class _FlashAttention(torch.autograd.Function):
+    """Operates in [B,H,S,D] domain; the public wrapper transposes."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal: bool, scale: float):
+        # q: [B,H,S,D] contiguous; k,v: [B,Hkv,S,D] contiguous
+        B, H, S, D = q.shape
+        Hkv = k.shape[1]
+        vt = v.transpose(-1, -2).contiguous()  # [B,Hkv,D,S]
+        o = torch.empty_like(q)
+        lse = torch.empty(B, H, S, device=q.device, dtype=torch.float32)
+        check(
+            lib().prime_flash_fwd(
+                stream_of(q), ptr(q), ptr(k), ptr(vt), ptr(o), ptr(lse),
+                B, H, Hkv, S, D, scale, int(causal),
+            ),
+            "flash_fwd",
+        )
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.meta = (causal, scale)
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        causal, scale = ctx.meta
+        B, H, S, D = q.shape
+        Hkv = k.shape[1]
+        do = do.contiguous()
+        delta = torch.empty(B * H * S, device=q.device, dtype=torch.float32)
+        check(
+            lib().prime_attn_delta(stream_of(q), ptr(do), ptr(o), ptr(delta), B * H * S, D),
+            "attn_delta",
+        )
+        kt = k.transpose(-1, -2).contiguous()
+        dq = torch.empty_like(q)
+        check(
+            lib().prime_flash_bwd_dq(
+                stream_of(q), ptr(q), ptr(k), ptr(kt), ptr(v), ptr(do), ptr(lse), ptr(delta),
+                ptr(dq), B, H, Hkv, S, D, scale, int(causal),
+            ),
+            "flash_bwd_dq",
+        )
+        qt = q.transpose(-1, -2).contiguous()
+        dot = do.transpose(-1, -2).contiguous()
+        dk = torch.empty_like(k)
+        dv = torch.empty_like(v)
+        check(
+            lib().prime_flash_bwd_dkv(
+                stream_of(q), ptr(q), ptr(qt), ptr(k), ptr(v), ptr(do), ptr(dot), ptr(lse),
+                ptr(delta), ptr(dk), ptr(dv), B, H, Hkv, S, D, scale, int(causal),
+            ),
+            "flash_bwd_dkv",
+        )
+        return dq, dk, dv, None, None
+
+
+def flash_attention(q, k, v, causal: bool = True) -> torch.Tensor:
+    """q: [B,S,H,D]; k,v: [B,S,Hkv,D] (GQA). Returns [B,S,H,D]."""
+    if not _is_hip(q):
+        return ref.attention(q, k, v, causal)
+    D = q.shape[-1]
+    scale = D**-0.5
+    qc = q.transpose(1, 2).contiguous()
+    kc = k.transpose(1, 2).contiguous()
+    vc = v.transpose(1, 2).contiguous()
+    o = _FlashAttention.apply(qc, kc, vc, causal, scale)
+    return o.transpose(1, 2)
+
+
+# --------------------------------------------------------- cross entropy
+class _CrossEntropy(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, targets, ignore_index: int):
+        # logits [R, V] bf16, targets [R] int32/int64
+        R, V = logits.shape
+        logits = logits.contiguous()
+        t32 = targets.to(torch.int32).contiguous()
+        loss = torch.empty(R, device=logits.device, dtype=torch.float32)
+        dlogits = torch.empty_like(logits)
+        check(
+            lib().prime_cross_entropy(
+                stream_of(logits), ptr(logits), ptr(t32), ptr(loss), ptr(dlogits),
+                R, V, 1.0, ignore_index, 1,
+            ),
+            "cross_entropy",
+        )
+        n_valid = (targets != ignore_index).sum().clamp(min=1)
+        ctx.save_for_backward(dlogits, n_valid)
+        return loss.sum() / n_valid.float()
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        dlogits, n_valid = ctx.saved_tensors
+        g = dlogits * (grad_out.float() / n_valid.float()).to(dlogits.dtype)
+        return g, None, None
+
+
+def cross_entropy(logits, targets, ignore_index: int = -100) -> torch.Tensor:
+    """Fused CE: mean loss over non-ignored tokens; grad computed in-kernel."""
+    if not _is_hip(logits):
+        return ref.cross_entropy(logits, targets, ignore_index)
+    return _CrossEntropy.apply(logits, targets, ignore_index)
+
+
+# ------------------------------------------------- raw (non-autograd) ops
+def fused_adamw(p32, p16, grad, m, v, *, lr, beta1, beta2, eps, wd, step):
+    check(
+        lib().prime_adamw(
+            stream_of(p32), ptr(p32), ptr(p16), ptr(grad), ptr(m), ptr(v),
+            p32.numel(), lr, beta1, beta2, eps, wd, step,
+        ),
+        "adamw",
+    )
+
+
+def pseudograd(outer32, master32, out_delta):
+    check(
+        lib().prime_pseudograd(
+            stream_of(outer32), ptr(outer32), ptr(master32), ptr(out_delta), outer32.numel()
+        ),
+        "pseudograd",
+    )
+
+
+QBLK = 1024
+
+
+def quant_int8(x32: torch.Tensor):
+    n = x32.numel()
+    nblk = (n + QBLK - 1) // QBLK
+    q = torch.empty(n, device=x32.device, dtype=torch.int8)
+    scales = torch.empty(nblk, device=x32.device, dtype=torch.float32)
+    check(lib().prime_quant_int8(stream_of(x32), ptr(x32), ptr(q), ptr(scales), n), "quant_int8")
+    return q, scales
+
+
+def dequant_int8(q, scales, out, accumulate: bool = False):
+    check(
+        lib().prime_dequant_int8(
+            stream_of(out), ptr(q), ptr(scales), ptr(out), q.numel(), int(accumulate)
+        ),
+        "dequant_int8",
+    )
+    return out
+
+
+def nesterov_outer(theta32, master32, inner16, buf32, delta32, *, lr, mu):
+    check(
+        lib().prime_nesterov_outer(
+            stream_of(theta32), ptr(theta32), ptr(master32), ptr(inner16), ptr(buf32),
+            ptr(delta32), theta32.numel(), lr, mu,
+        ),
+        "nesterov_outer",
+    )
+
+
+def mfma_probe(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
+    """16x16x32 bf16 MFMA single-tile matmul (layout verification)."""
+    C = torch.empty(16, 16, device=A.device, dtype=torch.float32)
+    check(lib().prime_mfma_probe(stream_of(A), ptr(A), ptr(B), ptr(C)), "mfma_probe")
+    return C

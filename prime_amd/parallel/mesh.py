@@ -1,0 +1,116 @@
+"""ElasticDeviceMesh — the process-group topology for DiLoCo training.
+
+Two nested layers:
+  - LOCAL  (intra-worker): the ranks of one DiLoCo worker, RCCL over xGMI.
+    Carries the inner-loop collectives (DP grad all-reduce / FSDP
+    reduce-scatter + all-gather).
+  - OUTER  (cross-worker): one group per local index, linking rank i of
+    every worker. Carries the every-H int8 pseudo-gradient ring.
+
+Static mode: the full world is one torchrun job; groups are carved out of
+it. Elastic mode (ElasticRegistry): workers register in a TCPStore with
+heartbeats; membership changes are observed at outer-step boundaries and
+the outer layer is rebuilt (prime's ElasticDeviceMesh semantics: dynamic
+join/leave, eviction of dead peers, live checkpoint recovery for joiners —
+see prime_amd/parallel/elastic.py).
+"""
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass
+
+import torch
+import torch.distributed as dist
+
+from . import ring
+
+
+def _env_int(name: str, default: int) -> int:
+    return int(os.environ.get(name, default))
+
+
+@dataclass
+class MeshConfig:
+    worker_size: int = 1          # GPUs per DiLoCo worker
+    backend: str | None = None    # default: nccl(=RCCL) on GPU, gloo on CPU
+    quant_outer: bool = True      # int8 ring for the outer all-reduce
+    timeout_s: float = 600.0
+
+
+class ElasticDeviceMesh:
+    def __init__(self, cfg: MeshConfig | None = None):
+        self.cfg = cfg or MeshConfig()
+        self.rank = _env_int("RANK", 0)
+        self.world_size = _env_int("WORLD_SIZE", 1)
+        self.local_rank = _env_int("LOCAL_RANK", self.rank)
+        cuda = torch.cuda.is_available()
+        self.backend = self.cfg.backend or ("nccl" if cuda else "gloo")
+        if cuda:
+            torch.cuda.set_device(self.local_rank % torch.cuda.device_count())
+        self.device = (
+            torch.device("cuda", torch.cuda.current_device()) if cuda else torch.device("cpu")
+        )
+
+        ws = self.cfg.worker_size
+        if self.world_size % ws != 0:
+            raise ValueError(f"world_size {self.world_size} not divisible by worker_size {ws}")
+        self.n_workers = self.world_size // ws
+        self.worker_id = self.rank // ws
+        self.worker_rank = self.rank % ws  # rank inside the worker
+
+        if self.world_size > 1 and not dist.is_initialized():
+            dist.init_process_group(
+                backend=self.backend,
+                rank=self.rank,
+                world_size=self.world_size,
+            )
+        self.initialized = dist.is_initialized()
+
+        # local group (ranks of my worker)
+        self.local_group = None
+        self.outer_group = None
+        if self.initialized:
+            if ws > 1:
+                for w in range(self.n_workers):
+                    g = dist.new_group(list(range(w * ws, (w + 1) * ws)))
+                    if w == self.worker_id:
+                        self.local_group = g
+            if self.n_workers > 1:
+                for i in range(ws):
+                    g = dist.new_group([w * ws + i for w in range(self.n_workers)])
+                    if i == self.worker_rank:
+                        self.outer_group = g
+
+    # ------------------------------------------------------------- helpers
+    @property
+    def is_leader(self) -> bool:
+        return self.worker_rank == 0
+
+    def local_allreduce_grad(self, flat_grad: torch.Tensor) -> None:
+        """Inner-loop DP gradient average across the worker's ranks."""
+        if self.local_group is None:
+            return
+        dist.all_reduce(flat_grad, group=self.local_group)
+        flat_grad.div_(self.cfg.worker_size)
+
+    def outer_allreduce_avg(self, delta32: torch.Tensor) -> None:
+        """Cross-worker pseudo-gradient average (int8 ring by default)."""
+        if self.outer_group is None and self.n_workers == 1:
+            return
+        group = self.outer_group
+        if self.cfg.quant_outer:
+            ring.ring_allreduce_int8(delta32, group=group, average=True)
+        else:
+            ring.allreduce_fp32(delta32, group=group, average=True)
+
+    def barrier(self) -> None:
+        if self.initialized:
+            if self.device.type == "cuda":
+                dist.barrier(device_ids=[self.device.index])
+            else:
+                dist.barrier()
+
+    def destroy(self) -> None:
+        if self.initialized and dist.is_initialized():
+            dist.destroy_process_group()
+            self.initialized = False

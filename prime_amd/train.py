@@ -1,0 +1,270 @@
+"""Training loop: DiLoCo inner/outer over the fused-kernel Llama.
+
+torchrun-able (one process per GPU, RANK/WORLD_SIZE/LOCAL_RANK from env) and
+callable in-process for tests. The same loop runs the CPU/gloo plumbing
+config (BASELINE.json config 1) and the MI355X configs.
+"""
+from __future__ import annotations
+
+import math
+import os
+import time
+from pathlib import Path
+
+import torch
+
+from .ckpt.manager import CheckpointManager
+from .data import DataConfig, build_dataloader
+from .models import build_model
+from .parallel.diloco import DilocoOptimizer
+from .parallel.flat import FlatParamSpace, FusedAdamW
+from .parallel.mesh import ElasticDeviceMesh, MeshConfig
+from .utils.config import TrainConfig
+from .utils.logging import JsonLogger
+from .utils.metrics import MetricsWriter, model_flops_per_token, mfu
+
+
+def lr_at(step: int, cfg) -> float:
+    """Warmup + cosine decay to min_lr_ratio."""
+    if step < cfg.warmup_steps:
+        return cfg.lr * (step + 1) / max(1, cfg.warmup_steps)
+    if not cfg.lr_decay_steps:
+        return cfg.lr
+    t = min(1.0, (step - cfg.warmup_steps) / max(1, cfg.lr_decay_steps - cfg.warmup_steps))
+    return cfg.lr * (cfg.min_lr_ratio + (1 - cfg.min_lr_ratio) * 0.5 * (1 + math.cos(math.pi * t)))
+
+
+class Trainer:
+    def __init__(self, cfg: TrainConfig, run_dir: str | Path | None = None):
+        self.cfg = cfg
+        self.run_dir = Path(run_dir) if run_dir else Path("runs") / cfg.run_name
+        self.run_dir.mkdir(parents=True, exist_ok=True)
+
+        self.mesh = ElasticDeviceMesh(MeshConfig(
+            worker_size=cfg.parallel.worker_size,
+            backend=cfg.parallel.backend,
+            quant_outer=cfg.diloco.quant_int8,
+        ))
+        dev = cfg.device or ("cuda" if torch.cuda.is_available() else "cpu")
+        self.device = self.mesh.device if dev == "cuda" else torch.device(dev)
+
+        self.log = JsonLogger(log_path=self.run_dir / f"rank{self.mesh.rank}.log",
+                              rank=self.mesh.rank)
+        self.metrics = MetricsWriter(
+            self.run_dir / "metrics.jsonl" if self.mesh.rank == 0 else None
+        )
+
+        torch.manual_seed(1234)
+        self.model = build_model(
+            cfg.model.name,
+            activation_checkpointing=cfg.model.activation_checkpointing,
+            **cfg.model.overrides,
+        )
+        self.model_cfg = self.model.cfg
+        if self.device.type == "cuda":
+            self.model = self.model.to(self.device, dtype=torch.bfloat16)
+        else:
+            self.model = self.model.to(self.device)
+        self.model.reset_rope(self.device)  # tables stay fp32
+
+        # broadcast initial params so every rank starts identical
+        self.flat = FlatParamSpace(self.model)
+        if self.mesh.initialized:
+            import torch.distributed as dist
+            dist.broadcast(self.flat.flat_w, src=0)
+            self.flat.master32.copy_(self.flat.flat_w.float())
+
+        self.inner = FusedAdamW(
+            self.flat, lr=cfg.optim.lr, betas=tuple(cfg.optim.betas),
+            eps=cfg.optim.eps, weight_decay=cfg.optim.weight_decay,
+        )
+        self.diloco = DilocoOptimizer(
+            self.flat, self.mesh, self.inner,
+            outer_lr=cfg.diloco.outer_lr,
+            outer_momentum=cfg.diloco.outer_momentum,
+            H=cfg.diloco.H if cfg.diloco.enabled else 10**9,
+            outer_device=cfg.diloco.outer_device,
+        )
+
+        data_cfg = DataConfig(
+            kind=cfg.data.kind, path=cfg.data.path, seq_len=cfg.model.seq_len,
+            micro_batch_size=cfg.data.micro_batch_size, seed=cfg.data.seed,
+        )
+        self.data = build_dataloader(
+            data_cfg, self.model_cfg.vocab_size,
+            shard=self.mesh.rank, n_shards=max(1, self.mesh.world_size),
+        )
+
+        self.ckpt = None
+        if cfg.checkpoint.interval > 0 or cfg.checkpoint.resume:
+            self.ckpt = CheckpointManager(
+                cfg.checkpoint.path or (self.run_dir / "ckpt"),
+                keep=cfg.checkpoint.keep, async_save=cfg.checkpoint.async_save,
+                worker_id=self.mesh.worker_id, is_leader=self.mesh.is_leader,
+            )
+        self.step_count = 0
+        if cfg.checkpoint.resume and self.ckpt:
+            self._resume(cfg.checkpoint.resume)
+
+        self.tokens_per_step = (
+            cfg.data.micro_batch_size * cfg.data.grad_accum * cfg.model.seq_len
+            * max(1, self.mesh.world_size)
+        )
+        self.flops_per_token = model_flops_per_token(self.model_cfg, cfg.model.seq_len)
+
+    # ---------------------------------------------------------------- steps
+    def train_step(self) -> torch.Tensor:
+        cfg = self.cfg
+        self.inner.lr = lr_at(self.step_count, cfg.optim)
+        self.flat.zero_grad()
+        loss_acc = None
+        for _ in range(cfg.data.grad_accum):
+            x, y = self.data.next_batch(self.device)
+            loss = self.model.loss(x, y) / cfg.data.grad_accum
+            loss.backward()
+            loss_acc = loss.detach() if loss_acc is None else loss_acc + loss.detach()
+        self.mesh.local_allreduce_grad(self.flat.flat_grad)
+        if cfg.optim.grad_clip > 0:
+            self.flat.clip_grad_norm_(cfg.optim.grad_clip)
+        did_outer = self.diloco.step()
+        self.step_count += 1
+        if did_outer:
+            self._maybe_checkpoint()
+        return loss_acc
+
+    def _maybe_checkpoint(self) -> None:
+        if not self.ckpt or self.cfg.checkpoint.interval <= 0:
+            return
+        if self.diloco.outer_step_count % self.cfg.checkpoint.interval != 0:
+            return
+        if not self.mesh.is_leader:
+            return
+        self.save_checkpoint()
+
+    def save_checkpoint(self) -> None:
+        t0 = time.perf_counter()
+        tensors = {
+            "master32": self.flat.master32,
+            "theta_outer": self.diloco.theta_outer,
+            "outer_buf": self.diloco.outer_buf,
+            "adam_m": self.inner.m,
+            "adam_v": self.inner.v,
+        }
+        meta = {
+            "inner_step": self.diloco.inner_step_count,
+            "outer_step": self.diloco.outer_step_count,
+            "adam_step": self.inner.step_count,
+            "step_count": self.step_count,
+            "data_state": self.data.state_dict(),
+            "model": self.model_cfg.name,
+        }
+        self.ckpt.save(self.diloco.outer_step_count, tensors, meta)
+        self.log.info(
+            f"checkpoint staged at outer step {self.diloco.outer_step_count} "
+            f"({time.perf_counter() - t0:.3f}s to stage)", type_="checkpoint",
+        )
+
+    def _resume(self, tag: str) -> None:
+        payload = self.ckpt.load(None if tag == "latest" else tag,
+                                 map_location=self.device)
+        if payload is None:
+            self.log.warning(f"no checkpoint found for resume='{tag}'")
+            return
+        t = payload["tensors"]
+        m = payload["meta"]
+        self.flat.load_flat_(t["master32"].to(self.flat.master32.device))
+        self.diloco.theta_outer.copy_(t["theta_outer"])
+        self.diloco.outer_buf.copy_(t["outer_buf"])
+        self.inner.m.copy_(t["adam_m"])
+        self.inner.v.copy_(t["adam_v"])
+        self.inner.step_count = m["adam_step"]
+        self.diloco.inner_step_count = m["inner_step"]
+        self.diloco.outer_step_count = m["outer_step"]
+        self.step_count = m["step_count"]
+        self.data.load_state_dict(m["data_state"])
+        self.log.info(f"resumed from outer step {m['outer_step']} (inner {m['inner_step']})")
+
+    # ----------------------------------------------------------------- run
+    def run(self) -> dict:
+        cfg = self.cfg
+        self.log.info(
+            f"starting run '{cfg.run_name}': model={self.model_cfg.name} "
+            f"({self.model_cfg.n_params()/1e6:.1f}M params), "
+            f"workers={self.mesh.n_workers}x{cfg.parallel.worker_size}, "
+            f"H={cfg.diloco.H if cfg.diloco.enabled else 'off'}, device={self.device}"
+        )
+        t_start = time.perf_counter()
+        window_t0, window_steps = t_start, 0
+        last_loss = float("nan")
+        loss = None
+        for _ in range(cfg.steps):
+            loss = self.train_step()
+            window_steps += 1
+            if self.step_count % cfg.metrics.log_interval == 0:
+                if self.device.type == "cuda":
+                    torch.cuda.synchronize()
+                now = time.perf_counter()
+                dt = now - window_t0
+                tps = self.tokens_per_step * window_steps / dt
+                tps_gpu = tps / max(1, self.mesh.world_size)
+                last_loss = float(loss)
+                self.metrics.write(
+                    self.step_count, loss=last_loss, tokens_per_sec=tps,
+                    tokens_per_sec_per_gpu=tps_gpu,
+                    mfu=mfu(tps_gpu, self.flops_per_token),
+                    lr=self.inner.lr, ms_per_step=1000 * dt / window_steps,
+                    outer_steps=self.diloco.outer_step_count,
+                )
+                self.log.progress(
+                    f"step {self.step_count}/{cfg.steps} loss={last_loss:.4f} "
+                    f"tok/s={tps:,.0f} mfu={mfu(tps_gpu, self.flops_per_token):.3f}"
+                )
+                window_t0, window_steps = now, 0
+        if self.device.type == "cuda":
+            torch.cuda.synchronize()
+        total_t = time.perf_counter() - t_start
+        if loss is not None:
+            last_loss = float(loss)
+        if self.ckpt:
+            self.ckpt.wait()
+        result = {
+            "steps": self.step_count,
+            "outer_steps": self.diloco.outer_step_count,
+            "loss": last_loss,
+            "total_time_s": total_t,
+            "tokens_per_sec": self.tokens_per_step * cfg.steps / total_t,
+        }
+        self.log.info(f"run complete: {result}", type_="result")
+        return result
+
+    def close(self) -> None:
+        if self.ckpt:
+            self.ckpt.wait()
+        self.metrics.close()
+        self.log.close()
+        self.mesh.destroy()
+
+
+def train_from_config(cfg: TrainConfig, run_dir=None) -> dict:
+    tr = Trainer(cfg, run_dir)
+    try:
+        return tr.run()
+    finally:
+        tr.close()
+
+
+def main() -> None:
+    import argparse
+
+    from .utils.config import load_config
+
+    ap = argparse.ArgumentParser(description="prime_amd trainer (torchrun-able)")
+    ap.add_argument("config", help="TOML run config")
+    ap.add_argument("--run-dir", default=None)
+    args = ap.parse_args()
+    cfg = load_config(args.config)
+    train_from_config(cfg, args.run_dir)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,59 @@
+import torch
+
+from prime_amd.ckpt import CheckpointManager
+from prime_amd.utils.config import (
+    CheckpointConfig, DilocoConfig, MetricsConfig, ModelConfig, TrainConfig,
+)
+from prime_amd.train import Trainer
+
+
+def test_save_load_roundtrip(tmp_path):
+    mgr = CheckpointManager(tmp_path / "ck", keep=2, async_save=False)
+    t = {"a": torch.arange(10, dtype=torch.float32), "b": torch.ones(4)}
+    mgr.save(1, t, {"step": 1})
+    out = mgr.load()
+    assert out is not None
+    torch.testing.assert_close(out["tensors"]["a"], t["a"])
+    assert out["meta"]["step"] == 1
+
+
+def test_retention(tmp_path):
+    mgr = CheckpointManager(tmp_path / "ck", keep=2, async_save=False)
+    for s in (1, 2, 3, 4):
+        mgr.save(s, {"x": torch.tensor([float(s)])}, {})
+    dirs = sorted(
+        d.name for d in (tmp_path / "ck").iterdir()
+        if d.is_dir() and not d.is_symlink()
+    )
+    assert dirs == ["step_3", "step_4"]
+    assert (tmp_path / "ck" / "latest").resolve().name == "step_4"
+
+
+def _train_cfg(steps, tmp_path, resume=None):
+    cfg = TrainConfig(
+        run_name="ckpt_test",
+        steps=steps,
+        model=ModelConfig(name="llama_test", seq_len=64),
+        diloco=DilocoConfig(H=2),
+        checkpoint=CheckpointConfig(
+            interval=1, path=str(tmp_path / "ck"), async_save=False, resume=resume
+        ),
+        metrics=MetricsConfig(log_interval=100),
+    )
+    cfg.data.micro_batch_size = 2
+    return cfg
+
+
+def test_trainer_resume(tmp_path):
+    tr = Trainer(_train_cfg(4, tmp_path), run_dir=tmp_path / "runA")
+    tr.run()
+    w_end = tr.flat.flat_w.clone()
+    inner_end = tr.diloco.inner_step_count
+    tr.close()
+
+    tr2 = Trainer(_train_cfg(2, tmp_path, resume="latest"), run_dir=tmp_path / "runB")
+    assert tr2.diloco.inner_step_count == inner_end
+    torch.testing.assert_close(tr2.flat.flat_w, w_end)
+    res = tr2.run()  # continues training
+    assert res["steps"] == inner_end + 2
+    tr2.close()

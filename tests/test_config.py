@@ -1,0 +1,44 @@
+import pytest
+
+from prime_amd.utils.config import (
+    ConfigError,
+    TrainConfig,
+    default_config_toml,
+    load_config,
+)
+
+
+def test_template_roundtrip(tmp_path):
+    p = tmp_path / "cfg.toml"
+    p.write_text(default_config_toml("llama_150m"))
+    cfg = load_config(p)
+    assert cfg.model.name == "llama_150m"
+    assert cfg.diloco.H == 100
+    assert cfg.diloco.quant_int8 is True
+
+
+def test_unknown_key_rejected(tmp_path):
+    p = tmp_path / "bad.toml"
+    p.write_text('steps = 10\n[model]\nnam = "typo"\n')
+    with pytest.raises(ConfigError) as ei:
+        load_config(p)
+    assert "model.nam" in str(ei.value)
+
+
+def test_toml_syntax_error(tmp_path):
+    p = tmp_path / "broken.toml"
+    p.write_text("steps = [unclosed")
+    with pytest.raises(ConfigError):
+        load_config(p)
+
+
+def test_missing_file():
+    with pytest.raises(ConfigError):
+        load_config("/nonexistent/cfg.toml")
+
+
+def test_defaults():
+    cfg = TrainConfig()
+    assert cfg.parallel.worker_size == 1
+    assert cfg.optim.lr == pytest.approx(3e-4)
+    assert cfg.checkpoint.interval == 0

@@ -1,0 +1,215 @@
+"""GPU numerics tests: every HIP kernel vs the PyTorch fp32 reference.
+
+All inputs are RANDOM and ASYMMETRIC (guide §3: symmetric inputs silently
+pass transposed MFMA layouts)."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def _bf(x):
+    return x.to(DEV, dtype=torch.bfloat16)
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _seed():
+    torch.manual_seed(1234)
+
+
+def test_hip_lib_loads():
+    from prime_amd.ops import have_lib
+
+    assert have_lib(), "HIP kernel library must load on GPU machines"
+
+
+def test_mfma_layout_vs_matmul():
+    """16x16x32 MFMA fragment-map verification with asymmetric random A,B."""
+    from prime_amd import ops
+
+    A = torch.randn(16, 32).clamp(-2, 2)
+    B = torch.randn(32, 16).clamp(-2, 2)
+    got = ops.mfma_probe(_bf(A).contiguous(), _bf(B).contiguous())
+    torch.cuda.synchronize()
+    want = _bf(A).float() @ _bf(B).float()
+    torch.testing.assert_close(got.cpu(), want.cpu(), atol=1e-3, rtol=1e-3)
+
+
+def test_rmsnorm_fwd_bwd():
+    from prime_amd import ops
+
+    x = _bf(torch.randn(33, 1024)).requires_grad_(True)
+    w = _bf(torch.randn(1024)).requires_grad_(True)
+    y = ops.rmsnorm(x, w)
+    dy = _bf(torch.randn_like(y.detach()))
+    y.backward(dy)
+
+    xr = x.detach().float().cpu().requires_grad_(True)
+    wr = w.detach().float().cpu().requires_grad_(True)
+    yr = ops.reference.rmsnorm(xr, wr)
+    yr.backward(dy.float().cpu())
+
+    torch.testing.assert_close(y.detach().float().cpu(), yr.detach(), atol=3e-2, rtol=3e-2)
+    torch.testing.assert_close(x.grad.float().cpu(), xr.grad, atol=3e-2, rtol=3e-2)
+    torch.testing.assert_close(w.grad.float().cpu(), wr.grad, atol=0.1, rtol=3e-2)
+
+
+def test_rope_fwd_bwd():
+    from prime_amd import ops
+    from prime_amd.ops import reference as ref
+
+    B, S, H, D = 2, 128, 4, 128
+    cos, sin = ref.rope_tables(D, 256, device=DEV)
+    x = _bf(torch.randn(B, S, H, D)).requires_grad_(True)
+    y = ops.apply_rope(x, cos, sin)
+    dy = _bf(torch.randn_like(y.detach()))
+    y.backward(dy)
+
+    xr = x.detach().float().cpu().requires_grad_(True)
+    yr = ref.apply_rope(xr, cos.cpu(), sin.cpu())
+    yr.backward(dy.float().cpu())
+    torch.testing.assert_close(y.detach().float().cpu(), yr.detach(), atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(x.grad.float().cpu(), xr.grad, atol=2e-2, rtol=2e-2)
+
+
+def test_swiglu_fwd_bwd():
+    from prime_amd import ops
+
+    gu = _bf(torch.randn(65, 512)).requires_grad_(True)
+    y = ops.swiglu(gu)
+    dy = _bf(torch.randn_like(y.detach()))
+    y.backward(dy)
+
+    gr = gu.detach().float().cpu().requires_grad_(True)
+    yr = ops.reference.swiglu(gr)
+    yr.backward(dy.float().cpu())
+    torch.testing.assert_close(y.detach().float().cpu(), yr.detach(), atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(gu.grad.float().cpu(), gr.grad, atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.parametrize("causal", [True, False])
+@pytest.mark.parametrize("gqa", [False, True])
+def test_flash_attention_fwd_bwd(causal, gqa):
+    from prime_amd import ops
+
+    B, S, H, D = 2, 256, 4, 128
+    Hkv = 2 if gqa else H
+    q = _bf(0.5 * torch.randn(B, S, H, D)).requires_grad_(True)
+    k = _bf(0.5 * torch.randn(B, S, Hkv, D)).requires_grad_(True)
+    v = _bf(0.5 * torch.randn(B, S, Hkv, D)).requires_grad_(True)
+    o = ops.flash_attention(q, k, v, causal=causal)
+    do = _bf(torch.randn_like(o.detach()))
+    o.backward(do)
+
+    qr = q.detach().float().cpu().requires_grad_(True)
+    kr = k.detach().float().cpu().requires_grad_(True)
+    vr = v.detach().float().cpu().requires_grad_(True)
+    orf = ops.reference.attention(qr, kr, vr, causal=causal)
+    orf.backward(do.float().cpu())
+
+    torch.testing.assert_close(o.detach().float().cpu(), orf.detach(), atol=3e-2, rtol=3e-2)
+    torch.testing.assert_close(q.grad.float().cpu(), qr.grad, atol=5e-2, rtol=5e-2)
+    torch.testing.assert_close(k.grad.float().cpu(), kr.grad, atol=5e-2, rtol=5e-2)
+    torch.testing.assert_close(v.grad.float().cpu(), vr.grad, atol=5e-2, rtol=5e-2)
+
+
+def test_flash_attention_head_dim_64():
+    from prime_amd import ops
+
+    B, S, H, D = 1, 128, 2, 64
+    q = _bf(torch.randn(B, S, H, D))
+    k = _bf(torch.randn(B, S, H, D))
+    v = _bf(torch.randn(B, S, H, D))
+    o = ops.flash_attention(q, k, v, causal=True)
+    want = ops.reference.attention(
+        q.float().cpu(), k.float().cpu(), v.float().cpu(), causal=True
+    )
+    torch.testing.assert_close(o.float().cpu(), want, atol=3e-2, rtol=3e-2)
+
+
+def test_cross_entropy_fwd_bwd():
+    from prime_amd import ops
+    import torch.nn.functional as F
+
+    R, V = 64, 32000
+    logits = _bf(torch.randn(R, V)).requires_grad_(True)
+    tgt = torch.randint(0, V, (R,), device=DEV)
+    tgt[5] = -100
+    loss = ops.cross_entropy(logits, tgt)
+    loss.backward()
+
+    lr = logits.detach().float().cpu().requires_grad_(True)
+    want = F.cross_entropy(lr, tgt.cpu(), ignore_index=-100)
+    want.backward()
+    torch.testing.assert_close(loss.float().cpu(), want.detach(), atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(
+        logits.grad.float().cpu(), lr.grad, atol=2e-3, rtol=5e-2
+    )
+
+
+def test_fused_adamw_vs_reference():
+    from prime_amd import ops
+    from prime_amd.ops import reference as ref
+
+    N = 4096 + 3  # exercise the scalar tail
+    p32 = torch.randn(N, device=DEV)
+    p16 = p32.bfloat16()
+    g = _bf(torch.randn(N))
+    m = torch.zeros(N, device=DEV)
+    v = torch.zeros(N, device=DEV)
+
+    p32r, mr, vr = p32.cpu().clone(), m.cpu().clone(), v.cpu().clone()
+    for step in (1, 2, 3):
+        ops.fused_adamw(p32, p16, g, m, v, lr=1e-2, beta1=0.9, beta2=0.95,
+                        eps=1e-8, wd=0.1, step=step)
+        ref.adamw_step(p32r, g.float().cpu(), mr, vr, 1e-2, 0.9, 0.95, 1e-8, 0.1, step)
+    torch.testing.assert_close(p32.cpu(), p32r, atol=1e-5, rtol=1e-5)
+    torch.testing.assert_close(m.cpu(), mr, atol=1e-5, rtol=1e-5)
+    torch.testing.assert_close(p16.float().cpu(), p32r, atol=1e-2, rtol=1e-2)
+
+
+def test_quant_int8_gpu_vs_reference():
+    from prime_amd import ops
+    from prime_amd.ops import reference as ref
+
+    x = torch.randn(ops.QBLK * 7, device=DEV) * 0.01
+    q, s = ops.quant_int8(x)
+    qr, sr = ref.quant_int8_blockwise(x.cpu())
+    torch.testing.assert_close(s.cpu(), sr, atol=1e-7, rtol=1e-5)
+    # rounding ties may differ by 1 lsb
+    assert (q.cpu().int() - qr.int()).abs().max() <= 1
+    out = torch.empty_like(x)
+    ops.dequant_int8(q, s, out)
+    torch.testing.assert_close(out, x, atol=float(x.abs().max() / 127), rtol=0)
+
+
+def test_nesterov_outer_gpu():
+    from prime_amd import ops
+
+    N = 2048
+    theta = torch.randn(N, device=DEV)
+    master = torch.randn(N, device=DEV)
+    p16 = torch.empty(N, device=DEV, dtype=torch.bfloat16)
+    buf = torch.randn(N, device=DEV)
+    delta = torch.randn(N, device=DEV)
+    theta0, buf0 = theta.clone(), buf.clone()
+
+    ops.nesterov_outer(theta, master, p16, buf, delta, lr=0.7, mu=0.9)
+    want_buf = 0.9 * buf0 + delta
+    want_theta = theta0 - 0.7 * (delta + 0.9 * want_buf)
+    torch.testing.assert_close(buf.cpu(), want_buf.cpu(), atol=1e-5, rtol=1e-5)
+    torch.testing.assert_close(theta.cpu(), want_theta.cpu(), atol=1e-5, rtol=1e-5)
+    torch.testing.assert_close(master.cpu(), want_theta.cpu(), atol=1e-5, rtol=1e-5)
+    torch.testing.assert_close(p16.float().cpu(), want_theta.cpu(), atol=1e-2, rtol=1e-2)
+
+
+def test_pseudograd_gpu():
+    from prime_amd import ops
+
+    outer = torch.randn(1024, device=DEV)
+    master = torch.randn(1024, device=DEV)
+    delta = torch.empty(1024, device=DEV)
+    ops.pseudograd(outer, master, delta)
+    torch.testing.assert_close(delta, outer - master)
