@@ -66,11 +66,16 @@ class DilocoOptimizer:
         else:
             self.chunk = (chunk_elems // _RING_ALIGN) * _RING_ALIGN
             self._n_comm = (n + self.chunk - 1) // self.chunk * self.chunk
-            self.theta_outer = torch.zeros(self._n_comm, dtype=torch.float32,
-                                           pin_memory=True)
+
+            def _host_buf(k: int) -> torch.Tensor:
+                try:
+                    return torch.zeros(k, dtype=torch.float32, pin_memory=True)
+                except RuntimeError:  # pinned pool exhausted: pageable fallback
+                    return torch.zeros(k, dtype=torch.float32)
+
+            self.theta_outer = _host_buf(self._n_comm)
             self.theta_outer[:n].copy_(flat.master32.to("cpu"))
-            self.outer_buf = torch.zeros(self._n_comm, dtype=torch.float32,
-                                         pin_memory=True)
+            self.outer_buf = _host_buf(self._n_comm)
             dev = flat.device
             self._g_theta = torch.zeros(self.chunk, device=dev, dtype=torch.float32)
             self._g_buf = torch.zeros(self.chunk, device=dev, dtype=torch.float32)

@@ -55,16 +55,17 @@ class Trainer:
         )
 
         torch.manual_seed(1234)
-        self.model = build_model(
-            cfg.model.name,
-            activation_checkpointing=cfg.model.activation_checkpointing,
-            **cfg.model.overrides,
-        )
+        # construct directly on the target device: CPU-side init of a 10B
+        # model takes minutes; on-GPU init is seconds
+        with torch.device(self.device):
+            self.model = build_model(
+                cfg.model.name,
+                activation_checkpointing=cfg.model.activation_checkpointing,
+                **cfg.model.overrides,
+            )
         self.model_cfg = self.model.cfg
         if self.device.type == "cuda":
-            self.model = self.model.to(self.device, dtype=torch.bfloat16)
-        else:
-            self.model = self.model.to(self.device)
+            self.model = self.model.to(dtype=torch.bfloat16)
         self.model.reset_rope(self.device)  # tables stay fp32
 
         # broadcast initial params so every rank starts identical
