@@ -41,15 +41,16 @@ class _RMSNorm(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy: torch.Tensor):
-        if len(ctx.saved_tensors) == 2:  # CPU path
-            x, w = ctx.saved_tensors
+        saved = ctx.saved_tensors  # read once: ckpt unpack hooks fire per access
+        if len(saved) == 2:  # CPU path
+            x, w = saved
             x = x.detach().float().requires_grad_(True)
             w2 = w.detach().float().requires_grad_(True)
             with torch.enable_grad():
                 y = ref.rmsnorm(x, w2, ctx.eps)
             gx, gw = torch.autograd.grad(y, [x, w2], dy.float())
             return gx.to(dy.dtype), gw.to(w.dtype), None
-        x, w, rstd = ctx.saved_tensors
+        x, w, rstd = saved
         dy = dy.contiguous()
         D = x.shape[-1]
         R = x.numel() // D

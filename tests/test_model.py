@@ -31,7 +31,15 @@ def test_activation_checkpointing_same_loss():
     m2.train()
     x = torch.randint(0, 256, (2, 64))
     y = torch.randint(0, 256, (2, 64))
-    torch.testing.assert_close(m1.loss(x, y), m2.loss(x, y), atol=1e-5, rtol=1e-5)
+    l1 = m1.loss(x, y)
+    l2 = m2.loss(x, y)
+    torch.testing.assert_close(l1, l2, atol=1e-5, rtol=1e-5)
+    # backward through the checkpointed graph (catches saved_tensors re-access)
+    l1.backward()
+    l2.backward()
+    g1 = m1.layers[0].attn.wqkv.weight.grad
+    g2 = m2.layers[0].attn.wqkv.weight.grad
+    torch.testing.assert_close(g1, g2, atol=1e-5, rtol=1e-4)
 
 
 def test_flat_param_space_views():
