@@ -1,28 +1,36 @@
-// Flash attention (causal/non-causal, GQA) for gfx950 — MFMA
-// v_mfma_f32_16x16x32_bf16 tiles, online softmax, FA2-style split backward
-// (dQ kernel with q-outer loop; dK/dV kernel with kv-outer loop; no
-// atomics — each accumulator lives in registers).
+// Flash attention v2 (causal/non-causal, GQA) for gfx950 — MFMA
+// v_mfma_f32_16x16x32_bf16 tiles, online softmax, FA2-style split backward.
 //
-// Layouts (chosen so EVERY global operand-fragment load is one contiguous
-// 16-byte bf16x8 per lane — see fragment maps below):
-//   fwd:  Q[B,H,S,D], K[B,Hkv,S,D], Vt[B,Hkv,D,S]  ->  O[B,H,S,D], lse[B,H,S]
-//   bwd:  additionally Kt[B,Hkv,D,S], Qt[B,H,D,S], dO[B,H,S,D], dOt[B,H,D,S],
-//         delta[B,H,S] (rowsum(dO*O), precomputed)
-// S must be a multiple of 64 (checked host-side).
+// v2 design (profiled v1 at ~70 TF: redundant per-wave global reads and
+// unswizzled LDS dominated):
+//  - K/V(+transposed) tiles are staged COOPERATIVELY once per block with
+//    __builtin_amdgcn_global_load_lds width-16 (no VGPR round trip), with
+//    the T2 XOR-16B swizzle applied via the pre-swizzled SOURCE address
+//    (global_load_lds writes linearly: guide §5.5 T2 + rule 21).
+//  - all operand-fragment LDS reads are ld8 (16 B) at the swizzled address:
+//    2-way bank aliasing max (free on CDNA4).
+//  - per-wave P / dS buffers are swizzled the same way (v1 left a 16-way
+//    conflict on the PV A-fragment reads).
+//  - kernels are STRIDE-AWARE over [B,S,H,D] inputs: q/k/v are consumed
+//    directly as views into the packed qkv GEMM output, no transposes or
+//    .contiguous() copies (v1 spent 8% of step time in copyBuffer).
+//  - lse/delta layout [B,S,H] (contiguous with the row order of bshd).
 //
-// Fragment maps for mfma_f32_16x16x32_bf16 (verified on HW by
-// tests/test_ops_gpu.py::test_mfma_layout against torch.matmul):
+// Fragment maps for mfma_f32_16x16x32_bf16 (HW-verified by
+// tests/test_ops_gpu.py::test_mfma_layout_vs_matmul, asymmetric inputs):
 //   A[16][32]: lane l holds A[l%16][(l/16)*8 + j]          j = 0..7
 //   B[32][16]: lane l holds B[(l/16)*8 + j][l%16]
 //   C[16][16]: lane l holds C[(l/16)*4 + r][l%16]          r = 0..3
 //
 // Workgroup = 4 waves; each wave owns 16 q-rows (fwd/dQ) or 16 k-rows
-// (dK/dV); tile = 64 x 64. P/dS cross-layout hops (C-layout -> A-layout)
-// bounce through a per-wave LDS [16][64] bf16 buffer.
+// (dK/dV); tile = 64 x 64. S % 64 == 0 (checked host-side), D in {64,128}.
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) short short8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((address_space(3))) char lds_char;
+typedef __attribute__((address_space(3))) void lds_void;
+typedef __attribute__((address_space(1))) const void g_void;
 
 __device__ __forceinline__ short8 ld8(const bf16* p) {
   return *reinterpret_cast<const short8*>(p);
@@ -32,7 +40,6 @@ __device__ __forceinline__ f32x4 mfma16(short8 a, short8 b, f32x4 c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
-// reduce over the 16-lane group (low 4 bits of lane id)
 __device__ __forceinline__ float grp16_max(float x) {
 #pragma unroll
   for (int off = 1; off < 16; off <<= 1) x = fmaxf(x, __shfl_xor(x, off, 64));
@@ -46,14 +53,65 @@ __device__ __forceinline__ float grp16_sum(float x) {
 
 #define NEG_INF (-1e30f)
 
+// ---- swizzled-tile helpers --------------------------------------------
+// A [ROWS][COLS] bf16 tile lives linearly in LDS; byte (row, colb) is
+// stored at row*COLS*2 + (colb ^ ((row&7)<<4)). Stage from a strided
+// global matrix; read 8-element fragments at the same swizzle.
+
+template <int ROWS, int COLS>
+__device__ __forceinline__ void stage_tile(const bf16* __restrict__ gbase,
+                                           int64_t row_stride, bf16* lds_tile,
+                                           int tid256) {
+  constexpr int UNITS = ROWS * COLS / 8;  // 16 B units
+  constexpr int UPR = COLS / 8;           // units per row
+  static_assert(UNITS % 256 == 0, "tile not divisible by 256 lanes");
+#pragma unroll
+  for (int i = 0; i < UNITS / 256; ++i) {
+    const int u = i * 256 + tid256;
+    const int row = u / UPR;
+    const int colb = ((u % UPR) * 16) ^ ((row & 7) << 4);  // pre-swizzle src
+    const bf16* src = gbase + (int64_t)row * row_stride + colb / 2;
+    // linear dest: this issue covers bytes [i*4096 + wid*1024 ...): the
+    // wave-uniform base is derived from the wave id inside tid256
+    const int wid = tid256 >> 6;
+    lds_void* dst = (lds_void*)((lds_char*)lds_tile + i * 4096 + wid * 1024);
+    __builtin_amdgcn_global_load_lds((g_void*)src, dst, 16, 0, 0);
+  }
+}
+
+template <int COLS>
+__device__ __forceinline__ short8 ld8_swz(const bf16* lds_tile, int row,
+                                          int colb) {
+  return *reinterpret_cast<const short8*>(
+      reinterpret_cast<const char*>(lds_tile) + row * (COLS * 2) +
+      (colb ^ ((row & 7) << 4)));
+}
+
+template <int COLS>
+__device__ __forceinline__ void st16_swz(bf16* lds_tile, int row, int col,
+                                         bf16 v) {
+  *reinterpret_cast<bf16*>(reinterpret_cast<char*>(lds_tile) + row * (COLS * 2) +
+                           ((col * 2) ^ ((row & 7) << 4))) = v;
+}
+
+// wave-local fence: P/dS LDS writes must land before same-wave ld8 reads
+__device__ __forceinline__ void wave_lds_fence() {
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_sched_barrier(0);
+}
+
 // ---------------------------------------------------------------- forward
+// Q[B,S,H,D] (strided), K[B,S,Hkv,D] (strided), Vt[B,Hkv,D,S] (contiguous)
+// -> O[B,S,H,D] (contiguous), lse[B,S,H] fp32.
 template <int D>
 __global__ __launch_bounds__(256) void flash_fwd_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ Vt, bf16* __restrict__ O, float* __restrict__ lse,
-    int B, int H, int Hkv, int S, float scale, int causal) {
-  constexpr int DS = D / 32;   // 32-wide d slices (MFMA K dim)
-  constexpr int DT = D / 16;   // 16-wide d subtiles (output cols)
+    int B, int H, int Hkv, int S, float scale, int causal,
+    int64_t sqb, int64_t sqs, int64_t sqh,
+    int64_t skb, int64_t sks, int64_t skh) {
+  constexpr int DS = D / 32;
+  constexpr int DT = D / 16;
   const int n_qt = S / 64;
   const int bh = blockIdx.x / n_qt;
   const int qt = blockIdx.x - bh * n_qt;
@@ -61,22 +119,22 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
   const int hkv = h / (H / Hkv);
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int lg = lane >> 4;   // 4-group index (0..3)
-  const int li = lane & 15;   // index within 16-lane group
+  const int lg = lane >> 4, li = lane & 15;
+  const int q0 = qt * 64 + wid * 16;
 
-  const int q0 = qt * 64 + wid * 16;  // this wave's first q row
-  const bf16* Qb = Q + ((int64_t)(b * H + h) * S) * D;
-  const bf16* Kb = K + ((int64_t)(b * Hkv + hkv) * S) * D;
+  const bf16* Qb = Q + b * sqb + h * sqh;
+  const bf16* Kb = K + b * skb + hkv * skh;
   const bf16* Vtb = Vt + ((int64_t)(b * Hkv + hkv) * D) * S;
 
-  __shared__ bf16 p_lds_all[4][16][64];
-  bf16(*p_lds)[64] = p_lds_all[wid];
+  __shared__ bf16 k_lds[64 * D];
+  __shared__ bf16 vt_lds[D * 64];
+  __shared__ bf16 p_lds_all[4][16 * 64];
+  bf16* p_lds = p_lds_all[wid];
 
-  // Q fragments, resident for the whole block
   short8 qf[DS];
 #pragma unroll
   for (int ds = 0; ds < DS; ++ds)
-    qf[ds] = ld8(Qb + (int64_t)(q0 + li) * D + ds * 32 + lg * 8);
+    qf[ds] = ld8(Qb + (int64_t)(q0 + li) * sqs + ds * 32 + lg * 8);
 
   float m[4], l[4];
   f32x4 o_acc[DT];
@@ -87,19 +145,23 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
 
   const int kv_end = causal ? (qt * 64 + 64) : S;
   for (int kv = 0; kv < kv_end; kv += 64) {
-    // ---- S = scale * Q K^T over four 16-key subtiles
+    __syncthreads();  // all waves done reading the previous tile
+    stage_tile<64, D>(Kb + (int64_t)kv * sks, sks, k_lds, threadIdx.x);
+    // Vt tile: rows d (stride S), cols k in [kv, kv+64)
+    stage_tile<D, 64>(Vtb + kv, S, vt_lds, threadIdx.x);
+    __syncthreads();  // staged (syncthreads drains vmcnt)
+
+    // ---- S = scale * Q K^T
     f32x4 s[4];
 #pragma unroll
     for (int sub = 0; sub < 4; ++sub) {
       f32x4 acc{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int ds = 0; ds < DS; ++ds) {
-        const short8 kf = ld8(Kb + (int64_t)(kv + sub * 16 + li) * D + ds * 32 + lg * 8);
-        acc = mfma16(qf[ds], kf, acc);
-      }
+      for (int ds = 0; ds < DS; ++ds)
+        acc = mfma16(qf[ds], ld8_swz<D>(k_lds, sub * 16 + li, ds * 64 + lg * 16), acc);
       s[sub] = acc;
     }
-    // ---- online softmax (per q-row = per reg r; row owned by 16-lane group)
+    // ---- online softmax per q-row (reg r), row owned by 16-lane group
     float alpha[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -129,45 +191,42 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
     for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
       for (int r = 0; r < 4; ++r) o_acc[dt][r] *= alpha[r];
-    // ---- P (C-layout) -> LDS (A-layout source)
-    __syncthreads();  // reads of previous iteration's P are done
+    // ---- P (C-layout) -> swizzled per-wave LDS (A-layout source)
 #pragma unroll
     for (int sub = 0; sub < 4; ++sub)
 #pragma unroll
       for (int r = 0; r < 4; ++r)
-        p_lds[lg * 4 + r][sub * 16 + li] = f2bf(s[sub][r]);
-    __syncthreads();
-    // ---- O += P V  (A from LDS, B from Vt: both contiguous 16B)
+        st16_swz<64>(p_lds, lg * 4 + r, sub * 16 + li, f2bf(s[sub][r]));
+    wave_lds_fence();
+    // ---- O += P V
 #pragma unroll
-    for (int dt = 0; dt < DT; ++dt) {
+    for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
-        const short8 pa = ld8(&p_lds[li][ks * 32 + lg * 8]);
-        const short8 vb = ld8(Vtb + (int64_t)(dt * 16 + li) * S + kv + ks * 32 + lg * 8);
-        o_acc[dt] = mfma16(pa, vb, o_acc[dt]);
-      }
-    }
+      for (int ks = 0; ks < 2; ++ks)
+        o_acc[dt] = mfma16(ld8_swz<64>(p_lds, li, ks * 64 + lg * 16),
+                           ld8_swz<64>(vt_lds, dt * 16 + li, ks * 64 + lg * 16),
+                           o_acc[dt]);
   }
-  // ---- epilogue: O /= l ; lse = m + log(l)
+  // ---- epilogue
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const float inv = (l[r] > 0.f) ? 1.f / l[r] : 0.f;
     const int qg = q0 + lg * 4 + r;
-    bf16* orow = O + (((int64_t)(b * H + h) * S) + qg) * D;
+    bf16* orow = O + (((int64_t)(b * S + qg)) * H + h) * D;
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt)
       orow[dt * 16 + li] = f2bf(o_acc[dt][r] * inv);
     if (li == 0)
-      lse[((int64_t)(b * H + h) * S) + qg] = m[r] + __logf(fmaxf(l[r], 1e-30f));
+      lse[((int64_t)(b * S + qg)) * H + h] = m[r] + __logf(fmaxf(l[r], 1e-30f));
   }
 }
 
 // ------------------------------------------------------- delta = rowsum(dO*O)
+// dO, O: contiguous [B,S,H,D] -> delta [B,S,H] (row-ordered, contiguous)
 __global__ void attn_delta_kernel(const bf16* __restrict__ dO,
                                   const bf16* __restrict__ O,
                                   float* __restrict__ delta, int64_t rows,
                                   int D) {
-  // one wave per row
   const int64_t row0 = (int64_t)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const int lane = threadIdx.x & 63;
   for (int64_t r = row0; r < rows; r += (int64_t)gridDim.x * (blockDim.x >> 6)) {
@@ -177,19 +236,23 @@ __global__ void attn_delta_kernel(const bf16* __restrict__ dO,
       acc += bf2f(dO[r * D + i + 1]) * bf2f(O[r * D + i + 1]);
     }
     acc = wave_reduce_sum(acc);
-    acc = __shfl(acc, 0, 64);
     if (lane == 0) delta[r] = acc;
   }
 }
 
 // ----------------------------------------------------------- backward dQ
+// Stages K[64][D], V[64][D] (B-operands for S and dP) and Kt[D][64]
+// (B-operand for dQ += dS*K). dO is contiguous [B,S,H,D].
 template <int D>
 __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
-    const bf16* __restrict__ Kt, const bf16* __restrict__ V,
+    const bf16* __restrict__ V, const bf16* __restrict__ Kt,
     const bf16* __restrict__ dO, const float* __restrict__ lse,
     const float* __restrict__ delta, bf16* __restrict__ dQ, int B, int H,
-    int Hkv, int S, float scale, int causal) {
+    int Hkv, int S, float scale, int causal,
+    int64_t sqb, int64_t sqs, int64_t sqh,
+    int64_t skb, int64_t sks, int64_t skh,
+    int64_t svb, int64_t svs, int64_t svh) {
   constexpr int DS = D / 32;
   constexpr int DT = D / 16;
   const int n_qt = S / 64;
@@ -202,28 +265,31 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
   const int lg = lane >> 4, li = lane & 15;
   const int q0 = qt * 64 + wid * 16;
 
-  const bf16* Qb = Q + ((int64_t)(b * H + h) * S) * D;
-  const bf16* dOb = dO + ((int64_t)(b * H + h) * S) * D;
-  const bf16* Kb = K + ((int64_t)(b * Hkv + hkv) * S) * D;
+  const bf16* Qb = Q + b * sqb + h * sqh;
+  const bf16* Kb = K + b * skb + hkv * skh;
+  const bf16* Vb = V + b * svb + hkv * svh;
   const bf16* Ktb = Kt + ((int64_t)(b * Hkv + hkv) * D) * S;
-  const bf16* Vb = V + ((int64_t)(b * Hkv + hkv) * S) * D;
-  const float* lse_b = lse + (int64_t)(b * H + h) * S;
-  const float* dl_b = delta + (int64_t)(b * H + h) * S;
+  const bf16* dOb = dO + ((int64_t)b * S * H + h) * D;  // row stride H*D
+  const float* lse_b = lse + (int64_t)b * S * H + h;    // stride H
+  const float* dl_b = delta + (int64_t)b * S * H + h;
 
-  __shared__ bf16 ds_lds_all[4][16][64];
-  bf16(*ds_lds)[64] = ds_lds_all[wid];
+  __shared__ bf16 k_lds[64 * D];
+  __shared__ bf16 v_lds[64 * D];
+  __shared__ bf16 kt_lds[D * 64];
+  __shared__ bf16 ds_lds_all[4][16 * 64];
+  bf16* ds_lds = ds_lds_all[wid];
 
   short8 qf[DS], dof[DS];
 #pragma unroll
   for (int ds = 0; ds < DS; ++ds) {
-    qf[ds] = ld8(Qb + (int64_t)(q0 + li) * D + ds * 32 + lg * 8);
-    dof[ds] = ld8(dOb + (int64_t)(q0 + li) * D + ds * 32 + lg * 8);
+    qf[ds] = ld8(Qb + (int64_t)(q0 + li) * sqs + ds * 32 + lg * 8);
+    dof[ds] = ld8(dOb + (int64_t)(q0 + li) * H * D + ds * 32 + lg * 8);
   }
   float lse_r[4], dl_r[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
-    lse_r[r] = lse_b[q0 + lg * 4 + r];
-    dl_r[r] = dl_b[q0 + lg * 4 + r];
+    lse_r[r] = lse_b[(int64_t)(q0 + lg * 4 + r) * H];
+    dl_r[r] = dl_b[(int64_t)(q0 + lg * 4 + r) * H];
   }
   f32x4 dq_acc[DT];
 #pragma unroll
@@ -231,16 +297,18 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
 
   const int kv_end = causal ? (qt * 64 + 64) : S;
   for (int kv = 0; kv < kv_end; kv += 64) {
-    __syncthreads();  // previous iteration's ds_lds reads done
+    __syncthreads();
+    stage_tile<64, D>(Kb + (int64_t)kv * sks, sks, k_lds, threadIdx.x);
+    stage_tile<64, D>(Vb + (int64_t)kv * svs, svs, v_lds, threadIdx.x);
+    stage_tile<D, 64>(Ktb + kv, S, kt_lds, threadIdx.x);
+    __syncthreads();
 #pragma unroll
     for (int sub = 0; sub < 4; ++sub) {
       f32x4 s_acc{0.f, 0.f, 0.f, 0.f}, dp_acc{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ds = 0; ds < DS; ++ds) {
-        const bf16* krow = Kb + (int64_t)(kv + sub * 16 + li) * D + ds * 32 + lg * 8;
-        const bf16* vrow = Vb + (int64_t)(kv + sub * 16 + li) * D + ds * 32 + lg * 8;
-        s_acc = mfma16(qf[ds], ld8(krow), s_acc);
-        dp_acc = mfma16(dof[ds], ld8(vrow), dp_acc);
+        s_acc = mfma16(qf[ds], ld8_swz<D>(k_lds, sub * 16 + li, ds * 64 + lg * 16), s_acc);
+        dp_acc = mfma16(dof[ds], ld8_swz<D>(v_lds, sub * 16 + li, ds * 64 + lg * 16), dp_acc);
       }
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -248,30 +316,30 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
         const int kg = kv + sub * 16 + li;
         float p = __expf(s_acc[r] * scale - lse_r[r]);
         if (causal && kg > qg) p = 0.f;
-        const float dsv = p * (dp_acc[r] - dl_r[r]) * scale;
-        ds_lds[lg * 4 + r][sub * 16 + li] = f2bf(dsv);
+        st16_swz<64>(ds_lds, lg * 4 + r, sub * 16 + li,
+                     f2bf(p * (dp_acc[r] - dl_r[r]) * scale));
       }
     }
-    __syncthreads();
+    wave_lds_fence();
 #pragma unroll
-    for (int dt = 0; dt < DT; ++dt) {
+    for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
-        const short8 a = ld8(&ds_lds[li][ks * 32 + lg * 8]);
-        const short8 bfr = ld8(Ktb + (int64_t)(dt * 16 + li) * S + kv + ks * 32 + lg * 8);
-        dq_acc[dt] = mfma16(a, bfr, dq_acc[dt]);
-      }
-    }
+      for (int ks = 0; ks < 2; ++ks)
+        dq_acc[dt] = mfma16(ld8_swz<64>(ds_lds, li, ks * 64 + lg * 16),
+                            ld8_swz<64>(kt_lds, dt * 16 + li, ks * 64 + lg * 16),
+                            dq_acc[dt]);
   }
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
-    bf16* row = dQ + (((int64_t)(b * H + h) * S) + q0 + lg * 4 + r) * D;
+    bf16* row = dQ + (((int64_t)(b * S + q0 + lg * 4 + r)) * H + h) * D;
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt) row[dt * 16 + li] = f2bf(dq_acc[dt][r]);
   }
 }
 
 // -------------------------------------------------------- backward dK, dV
+// Stages Q[64][D], dO[64][D] (B-operands for St and dPt), Qt[D][64] and
+// dOt[D][64] (B-operands for dK and dV accumulation).
 template <int D>
 __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ Qt,
@@ -279,7 +347,10 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
     const bf16* __restrict__ dO, const bf16* __restrict__ dOt,
     const float* __restrict__ lse, const float* __restrict__ delta,
     bf16* __restrict__ dK, bf16* __restrict__ dV, int B, int H, int Hkv,
-    int S, float scale, int causal) {
+    int S, float scale, int causal,
+    int64_t sqb, int64_t sqs, int64_t sqh,
+    int64_t skb, int64_t sks, int64_t skh,
+    int64_t svb, int64_t svs, int64_t svh) {
   constexpr int DS = D / 32;
   constexpr int DT = D / 16;
   const int n_kt = S / 64;
@@ -290,21 +361,25 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int lg = lane >> 4, li = lane & 15;
-  const int k0 = kt * 64 + wid * 16;  // this wave's first key row
+  const int k0 = kt * 64 + wid * 16;
 
-  const bf16* Kb = K + ((int64_t)(b * Hkv + hkv) * S) * D;
-  const bf16* Vb = V + ((int64_t)(b * Hkv + hkv) * S) * D;
+  const bf16* Kb = K + b * skb + hkv * skh;
+  const bf16* Vb = V + b * svb + hkv * svh;
 
-  __shared__ bf16 pt_lds_all[4][16][64];
-  __shared__ bf16 dst_lds_all[4][16][64];
-  bf16(*pt_lds)[64] = pt_lds_all[wid];
-  bf16(*dst_lds)[64] = dst_lds_all[wid];
+  __shared__ bf16 q_lds[64 * D];
+  __shared__ bf16 do_lds[64 * D];
+  __shared__ bf16 qt_lds[D * 64];
+  __shared__ bf16 dot_lds[D * 64];
+  __shared__ bf16 pt_lds_all[4][16 * 64];
+  __shared__ bf16 dst_lds_all[4][16 * 64];
+  bf16* pt_lds = pt_lds_all[wid];
+  bf16* dst_lds = dst_lds_all[wid];
 
   short8 kf[DS], vf[DS];
 #pragma unroll
   for (int ds = 0; ds < DS; ++ds) {
-    kf[ds] = ld8(Kb + (int64_t)(k0 + li) * D + ds * 32 + lg * 8);
-    vf[ds] = ld8(Vb + (int64_t)(k0 + li) * D + ds * 32 + lg * 8);
+    kf[ds] = ld8(Kb + (int64_t)(k0 + li) * sks + ds * 32 + lg * 8);
+    vf[ds] = ld8(Vb + (int64_t)(k0 + li) * svs + ds * 32 + lg * 8);
   }
   f32x4 dk_acc[DT], dv_acc[DT];
 #pragma unroll
@@ -315,58 +390,60 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
 
   for (int g = 0; g < group; ++g) {
     const int h = hkv * group + g;
-    const bf16* Qb = Q + ((int64_t)(b * H + h) * S) * D;
+    const bf16* Qb = Q + b * sqb + h * sqh;
     const bf16* Qtb = Qt + ((int64_t)(b * H + h) * D) * S;
-    const bf16* dOb = dO + ((int64_t)(b * H + h) * S) * D;
+    const bf16* dOb = dO + ((int64_t)b * S * H + h) * D;
     const bf16* dOtb = dOt + ((int64_t)(b * H + h) * D) * S;
-    const float* lse_b = lse + (int64_t)(b * H + h) * S;
-    const float* dl_b = delta + (int64_t)(b * H + h) * S;
+    const float* lse_b = lse + (int64_t)b * S * H + h;
+    const float* dl_b = delta + (int64_t)b * S * H + h;
     const int q_start = causal ? kt * 64 : 0;
     for (int q0g = q_start; q0g < S; q0g += 64) {
+      __syncthreads();
+      stage_tile<64, D>(Qb + (int64_t)q0g * sqs, sqs, q_lds, threadIdx.x);
+      stage_tile<64, D>(dOb + (int64_t)q0g * H * D, (int64_t)H * D, do_lds, threadIdx.x);
+      stage_tile<D, 64>(Qtb + q0g, S, qt_lds, threadIdx.x);
+      stage_tile<D, 64>(dOtb + q0g, S, dot_lds, threadIdx.x);
       __syncthreads();
 #pragma unroll
       for (int sub = 0; sub < 4; ++sub) {
         f32x4 st_acc{0.f, 0.f, 0.f, 0.f}, dpt_acc{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int ds = 0; ds < DS; ++ds) {
-          // B operands: column q = li, rows d -> contiguous in Q / dO rows
-          const short8 qb = ld8(Qb + (int64_t)(q0g + sub * 16 + li) * D + ds * 32 + lg * 8);
-          const short8 dob = ld8(dOb + (int64_t)(q0g + sub * 16 + li) * D + ds * 32 + lg * 8);
+          const short8 qb = ld8_swz<D>(q_lds, sub * 16 + li, ds * 64 + lg * 16);
+          const short8 dob = ld8_swz<D>(do_lds, sub * 16 + li, ds * 64 + lg * 16);
           st_acc = mfma16(kf[ds], qb, st_acc);
           dpt_acc = mfma16(vf[ds], dob, dpt_acc);
         }
         const int qg = q0g + sub * 16 + li;
-        const float lse_q = lse_b[qg];
-        const float dl_q = dl_b[qg];
+        const float lse_q = lse_b[(int64_t)qg * H];
+        const float dl_q = dl_b[(int64_t)qg * H];
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int kg = k0 + lg * 4 + r;
           float p = __expf(st_acc[r] * scale - lse_q);
           if (causal && kg > qg) p = 0.f;
-          pt_lds[lg * 4 + r][sub * 16 + li] = f2bf(p);
-          dst_lds[lg * 4 + r][sub * 16 + li] = f2bf(p * (dpt_acc[r] - dl_q) * scale);
+          st16_swz<64>(pt_lds, lg * 4 + r, sub * 16 + li, f2bf(p));
+          st16_swz<64>(dst_lds, lg * 4 + r, sub * 16 + li,
+                       f2bf(p * (dpt_acc[r] - dl_q) * scale));
         }
       }
-      __syncthreads();
+      wave_lds_fence();
 #pragma unroll
-      for (int dt = 0; dt < DT; ++dt) {
+      for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
-          const short8 pa = ld8(&pt_lds[li][ks * 32 + lg * 8]);
-          const short8 da = ld8(&dst_lds[li][ks * 32 + lg * 8]);
-          const short8 dob = ld8(dOtb + (int64_t)(dt * 16 + li) * S + q0g + ks * 32 + lg * 8);
-          const short8 qtb = ld8(Qtb + (int64_t)(dt * 16 + li) * S + q0g + ks * 32 + lg * 8);
-          dv_acc[dt] = mfma16(pa, dob, dv_acc[dt]);
-          dk_acc[dt] = mfma16(da, qtb, dk_acc[dt]);
+          const short8 pa = ld8_swz<64>(pt_lds, li, ks * 64 + lg * 16);
+          const short8 da = ld8_swz<64>(dst_lds, li, ks * 64 + lg * 16);
+          dv_acc[dt] = mfma16(pa, ld8_swz<64>(dot_lds, dt * 16 + li, ks * 64 + lg * 16), dv_acc[dt]);
+          dk_acc[dt] = mfma16(da, ld8_swz<64>(qt_lds, dt * 16 + li, ks * 64 + lg * 16), dk_acc[dt]);
         }
-      }
     }
   }
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int kg = k0 + lg * 4 + r;
-    bf16* krow = dK + (((int64_t)(b * Hkv + hkv) * S) + kg) * D;
-    bf16* vrow = dV + (((int64_t)(b * Hkv + hkv) * S) + kg) * D;
+    bf16* krow = dK + (((int64_t)(b * S + kg)) * Hkv + hkv) * D;
+    bf16* vrow = dV + (((int64_t)(b * S + kg)) * Hkv + hkv) * D;
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt) {
       krow[dt * 16 + li] = f2bf(dk_acc[dt][r]);
@@ -376,22 +453,27 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
 }
 
 // ------------------------------------------------------------------ host API
+#define DISPATCH_D(KER, ...)                                            \
+  do {                                                                  \
+    if (D == 128)                                                       \
+      hipLaunchKernelGGL(KER<128>, dim3(grid), dim3(256), 0, stream,    \
+                         __VA_ARGS__);                                  \
+    else                                                                \
+      hipLaunchKernelGGL(KER<64>, dim3(grid), dim3(256), 0, stream,     \
+                         __VA_ARGS__);                                  \
+  } while (0)
+
 PRIME_API int prime_flash_fwd(hipStream_t stream, const void* Q, const void* K,
                               const void* Vt, void* O, void* lse, int64_t B,
                               int64_t H, int64_t Hkv, int64_t S, int64_t D,
-                              double scale, int64_t causal) {
+                              double scale, int64_t causal,
+                              int64_t sqb, int64_t sqs, int64_t sqh,
+                              int64_t skb, int64_t sks, int64_t skh) {
   if (S % 64 != 0 || (D != 64 && D != 128)) return hipErrorInvalidValue;
   const int grid = (int)(B * H * (S / 64));
-  if (D == 128)
-    hipLaunchKernelGGL(flash_fwd_kernel<128>, dim3(grid), dim3(256), 0, stream,
-                       (const bf16*)Q, (const bf16*)K, (const bf16*)Vt,
-                       (bf16*)O, (float*)lse, (int)B, (int)H, (int)Hkv, (int)S,
-                       (float)scale, (int)causal);
-  else
-    hipLaunchKernelGGL(flash_fwd_kernel<64>, dim3(grid), dim3(256), 0, stream,
-                       (const bf16*)Q, (const bf16*)K, (const bf16*)Vt,
-                       (bf16*)O, (float*)lse, (int)B, (int)H, (int)Hkv, (int)S,
-                       (float)scale, (int)causal);
+  DISPATCH_D(flash_fwd_kernel, (const bf16*)Q, (const bf16*)K, (const bf16*)Vt,
+             (bf16*)O, (float*)lse, (int)B, (int)H, (int)Hkv, (int)S,
+             (float)scale, (int)causal, sqb, sqs, sqh, skb, sks, skh);
   return (int)hipGetLastError();
 }
 
@@ -406,25 +488,21 @@ PRIME_API int prime_attn_delta(hipStream_t stream, const void* dO,
 }
 
 PRIME_API int prime_flash_bwd_dq(hipStream_t stream, const void* Q,
-                                 const void* K, const void* Kt, const void* V,
+                                 const void* K, const void* V, const void* Kt,
                                  const void* dO, const void* lse,
                                  const void* delta, void* dQ, int64_t B,
                                  int64_t H, int64_t Hkv, int64_t S, int64_t D,
-                                 double scale, int64_t causal) {
+                                 double scale, int64_t causal,
+                                 int64_t sqb, int64_t sqs, int64_t sqh,
+                                 int64_t skb, int64_t sks, int64_t skh,
+                                 int64_t svb, int64_t svs, int64_t svh) {
   if (S % 64 != 0 || (D != 64 && D != 128)) return hipErrorInvalidValue;
   const int grid = (int)(B * H * (S / 64));
-  if (D == 128)
-    hipLaunchKernelGGL(flash_bwd_dq_kernel<128>, dim3(grid), dim3(256), 0,
-                       stream, (const bf16*)Q, (const bf16*)K, (const bf16*)Kt,
-                       (const bf16*)V, (const bf16*)dO, (const float*)lse,
-                       (const float*)delta, (bf16*)dQ, (int)B, (int)H,
-                       (int)Hkv, (int)S, (float)scale, (int)causal);
-  else
-    hipLaunchKernelGGL(flash_bwd_dq_kernel<64>, dim3(grid), dim3(256), 0,
-                       stream, (const bf16*)Q, (const bf16*)K, (const bf16*)Kt,
-                       (const bf16*)V, (const bf16*)dO, (const float*)lse,
-                       (const float*)delta, (bf16*)dQ, (int)B, (int)H,
-                       (int)Hkv, (int)S, (float)scale, (int)causal);
+  DISPATCH_D(flash_bwd_dq_kernel, (const bf16*)Q, (const bf16*)K,
+             (const bf16*)V, (const bf16*)Kt, (const bf16*)dO,
+             (const float*)lse, (const float*)delta, (bf16*)dQ, (int)B, (int)H,
+             (int)Hkv, (int)S, (float)scale, (int)causal, sqb, sqs, sqh, skb,
+             sks, skh, svb, svs, svh);
   return (int)hipGetLastError();
 }
 
@@ -434,22 +512,16 @@ PRIME_API int prime_flash_bwd_dkv(hipStream_t stream, const void* Q,
                                   const void* lse, const void* delta, void* dK,
                                   void* dV, int64_t B, int64_t H, int64_t Hkv,
                                   int64_t S, int64_t D, double scale,
-                                  int64_t causal) {
+                                  int64_t causal,
+                                  int64_t sqb, int64_t sqs, int64_t sqh,
+                                  int64_t skb, int64_t sks, int64_t skh,
+                                  int64_t svb, int64_t svs, int64_t svh) {
   if (S % 64 != 0 || (D != 64 && D != 128)) return hipErrorInvalidValue;
   const int grid = (int)(B * Hkv * (S / 64));
-  if (D == 128)
-    hipLaunchKernelGGL(flash_bwd_dkv_kernel<128>, dim3(grid), dim3(256), 0,
-                       stream, (const bf16*)Q, (const bf16*)Qt, (const bf16*)K,
-                       (const bf16*)V, (const bf16*)dO, (const bf16*)dOt,
-                       (const float*)lse, (const float*)delta, (bf16*)dK,
-                       (bf16*)dV, (int)B, (int)H, (int)Hkv, (int)S,
-                       (float)scale, (int)causal);
-  else
-    hipLaunchKernelGGL(flash_bwd_dkv_kernel<64>, dim3(grid), dim3(256), 0,
-                       stream, (const bf16*)Q, (const bf16*)Qt, (const bf16*)K,
-                       (const bf16*)V, (const bf16*)dO, (const bf16*)dOt,
-                       (const float*)lse, (const float*)delta, (bf16*)dK,
-                       (bf16*)dV, (int)B, (int)H, (int)Hkv, (int)S,
-                       (float)scale, (int)causal);
+  DISPATCH_D(flash_bwd_dkv_kernel, (const bf16*)Q, (const bf16*)Qt,
+             (const bf16*)K, (const bf16*)V, (const bf16*)dO, (const bf16*)dOt,
+             (const float*)lse, (const float*)delta, (bf16*)dK, (bf16*)dV,
+             (int)B, (int)H, (int)Hkv, (int)S, (float)scale, (int)causal, sqb,
+             sqs, sqh, skb, sks, skh, svb, svs, svh);
   return (int)hipGetLastError();
 }

@@ -152,21 +152,35 @@ def swiglu(gu: torch.Tensor) -> torch.Tensor:
 
 
 # ------------------------------------------------------- flash attention
+def _bshd_ok(t: torch.Tensor) -> bool:
+    """[B,S,H,D] view usable by the stride-aware kernels: last dim
+    contiguous, every stride a multiple of 8 elements (16 B alignment)."""
+    sb, ss, sh, sd = t.stride()
+    return sd == 1 and sb % 8 == 0 and ss % 8 == 0 and sh % 8 == 0 \
+        and t.storage_offset() % 8 == 0
+
+
 class _FlashAttention(torch.autograd.Function):
-    """Operates in [B,H,S,D] domain; the public wrapper transposes."""
+    """Stride-aware [B,S,H,D] domain: consumes q/k/v views of the packed
+    qkv GEMM output directly (no transposes / .contiguous() on the hot
+    path); only the genuinely-transposed operands (Vt/Kt/Qt/dOt, needed
+    for contiguous MFMA B-fragments) are materialized."""
 
     @staticmethod
     def forward(ctx, q, k, v, causal: bool, scale: float):
-        # q: [B,H,S,D] contiguous; k,v: [B,Hkv,S,D] contiguous
-        B, H, S, D = q.shape
-        Hkv = k.shape[1]
-        vt = v.transpose(-1, -2).contiguous()  # [B,Hkv,D,S]
-        o = torch.empty_like(q)
-        lse = torch.empty(B, H, S, device=q.device, dtype=torch.float32)
+        B, S, H, D = q.shape
+        Hkv = k.shape[2]
+        if not (_bshd_ok(q) and _bshd_ok(k) and _bshd_ok(v)):
+            q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        vt = v.permute(0, 2, 3, 1).contiguous()  # [B,Hkv,D,S]
+        o = torch.empty(B, S, H, D, device=q.device, dtype=q.dtype)
+        lse = torch.empty(B, S, H, device=q.device, dtype=torch.float32)
         check(
             lib().prime_flash_fwd(
                 stream_of(q), ptr(q), ptr(k), ptr(vt), ptr(o), ptr(lse),
                 B, H, Hkv, S, D, scale, int(causal),
+                q.stride(0), q.stride(1), q.stride(2),
+                k.stride(0), k.stride(1), k.stride(2),
             ),
             "flash_fwd",
         )
@@ -178,31 +192,39 @@ class _FlashAttention(torch.autograd.Function):
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
         causal, scale = ctx.meta
-        B, H, S, D = q.shape
-        Hkv = k.shape[1]
+        B, S, H, D = q.shape
+        Hkv = k.shape[2]
         do = do.contiguous()
-        delta = torch.empty(B * H * S, device=q.device, dtype=torch.float32)
+        delta = torch.empty(B * S * H, device=q.device, dtype=torch.float32)
         check(
-            lib().prime_attn_delta(stream_of(q), ptr(do), ptr(o), ptr(delta), B * H * S, D),
+            lib().prime_attn_delta(stream_of(q), ptr(do), ptr(o), ptr(delta), B * S * H, D),
             "attn_delta",
         )
-        kt = k.transpose(-1, -2).contiguous()
-        dq = torch.empty_like(q)
+        kt = k.permute(0, 2, 3, 1).contiguous()  # [B,Hkv,D,S]
+        dq = torch.empty(B, S, H, D, device=q.device, dtype=q.dtype)
         check(
             lib().prime_flash_bwd_dq(
-                stream_of(q), ptr(q), ptr(k), ptr(kt), ptr(v), ptr(do), ptr(lse), ptr(delta),
-                ptr(dq), B, H, Hkv, S, D, scale, int(causal),
+                stream_of(q), ptr(q), ptr(k), ptr(v), ptr(kt), ptr(do),
+                ptr(lse), ptr(delta), ptr(dq), B, H, Hkv, S, D, scale,
+                int(causal),
+                q.stride(0), q.stride(1), q.stride(2),
+                k.stride(0), k.stride(1), k.stride(2),
+                v.stride(0), v.stride(1), v.stride(2),
             ),
             "flash_bwd_dq",
         )
-        qt = q.transpose(-1, -2).contiguous()
-        dot = do.transpose(-1, -2).contiguous()
-        dk = torch.empty_like(k)
-        dv = torch.empty_like(v)
+        qt = q.permute(0, 2, 3, 1).contiguous()   # [B,H,D,S]
+        dot = do.permute(0, 2, 3, 1).contiguous()
+        dk = torch.empty(B, S, Hkv, D, device=q.device, dtype=q.dtype)
+        dv = torch.empty(B, S, Hkv, D, device=q.device, dtype=q.dtype)
         check(
             lib().prime_flash_bwd_dkv(
-                stream_of(q), ptr(q), ptr(qt), ptr(k), ptr(v), ptr(do), ptr(dot), ptr(lse),
-                ptr(delta), ptr(dk), ptr(dv), B, H, Hkv, S, D, scale, int(causal),
+                stream_of(q), ptr(q), ptr(qt), ptr(k), ptr(v), ptr(do),
+                ptr(dot), ptr(lse), ptr(delta), ptr(dk), ptr(dv),
+                B, H, Hkv, S, D, scale, int(causal),
+                q.stride(0), q.stride(1), q.stride(2),
+                k.stride(0), k.stride(1), k.stride(2),
+                v.stride(0), v.stride(1), v.stride(2),
             ),
             "flash_bwd_dkv",
         )
@@ -213,13 +235,7 @@ def flash_attention(q, k, v, causal: bool = True) -> torch.Tensor:
     """q: [B,S,H,D]; k,v: [B,S,Hkv,D] (GQA). Returns [B,S,H,D]."""
     if not _is_hip(q):
         return ref.attention(q, k, v, causal)
-    D = q.shape[-1]
-    scale = D**-0.5
-    qc = q.transpose(1, 2).contiguous()
-    kc = k.transpose(1, 2).contiguous()
-    vc = v.transpose(1, 2).contiguous()
-    o = _FlashAttention.apply(qc, kc, vc, causal, scale)
-    return o.transpose(1, 2)
+    return _FlashAttention.apply(q, k, v, causal, q.shape[-1] ** -0.5)
 
 
 # --------------------------------------------------------- cross entropy
