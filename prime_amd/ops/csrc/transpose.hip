@@ -1,0 +1,61 @@
+// Batched bf16 transpose [B,S,H,D] (strided) -> [B,H,D,S] (contiguous)
+// for the attention operands that genuinely need a transposed copy
+// (Vt/Kt/Qt/dOt: contiguous MFMA B-fragments). PyTorch's
+// permute().contiguous() runs this pattern at ~80 GB/s on ROCm; this
+// LDS-tiled version (64x64 tiles, 16 B loads AND stores, padded LDS rows)
+// is HBM-bound (~5 TB/s class).
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) short short8;
+
+#define TP 72  // padded LDS row: 64 + 8 elems (16 B) keeps both phases conflict-light
+
+__global__ __launch_bounds__(256) void transpose_bshd_kernel(
+    const bf16* __restrict__ src, bf16* __restrict__ dst, int B, int S, int H,
+    int D, int64_t sb, int64_t ss, int64_t sh) {
+  __shared__ bf16 tile[64][TP];
+  const int n_st = S / 64, n_dt = D / 64;
+  int idx = blockIdx.x;
+  const int dt = idx % n_dt; idx /= n_dt;
+  const int st = idx % n_st; idx /= n_st;
+  const int h = idx % H; idx /= H;
+  const int b = idx;
+  const bf16* sbase = src + b * sb + h * sh + (int64_t)(st * 64) * ss + dt * 64;
+  bf16* dbase = dst + (((int64_t)(b * H + h) * D) + dt * 64) * S + st * 64;
+
+  const int t = threadIdx.x;
+  // load 64 rows x 64 cols: 512 x 8-elem units, 2 per thread
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int u = i * 256 + t;
+    const int r = u >> 3, c8 = (u & 7) * 8;
+    *reinterpret_cast<short8*>(&tile[r][c8]) =
+        *reinterpret_cast<const short8*>(sbase + (int64_t)r * ss + c8);
+  }
+  __syncthreads();
+  // store transposed: dst row = d (64), cols = s (64)
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int u = i * 256 + t;
+    const int d = u >> 3, s8 = (u & 7) * 8;
+    short8 v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      reinterpret_cast<short*>(&v)[j] =
+          reinterpret_cast<const short*>(&tile[s8 + j][d])[0];
+    *reinterpret_cast<short8*>(dbase + (int64_t)d * S + s8) = v;
+  }
+}
+
+PRIME_API int prime_transpose_bshd(hipStream_t stream, const void* src,
+                                   void* dst, int64_t B, int64_t S, int64_t H,
+                                   int64_t D, int64_t sb, int64_t ss,
+                                   int64_t sh) {
+  if (S % 64 != 0 || D % 64 != 0) return hipErrorInvalidValue;
+  const int64_t grid = B * H * (S / 64) * (D / 64);
+  if (grid > 0x7fffffff) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(transpose_bshd_kernel, dim3((int)grid), dim3(256), 0,
+                     stream, (const bf16*)src, (bf16*)dst, (int)B, (int)S,
+                     (int)H, (int)D, sb, ss, sh);
+  return (int)hipGetLastError();
+}

@@ -213,3 +213,18 @@ def test_pseudograd_gpu():
     delta = torch.empty(1024, device=DEV)
     ops.pseudograd(outer, master, delta)
     torch.testing.assert_close(delta, outer - master)
+
+
+def test_transpose_bshd():
+    from prime_amd.ops.functional import transpose_bshd
+
+    B, S, H, D = 2, 128, 3, 128
+    x = torch.randn(B, S, H, D, device=DEV).bfloat16()
+    got = transpose_bshd(x)
+    want = x.permute(0, 2, 3, 1).contiguous()
+    assert got.shape == want.shape
+    torch.testing.assert_close(got, want)
+    # strided view input (like a qkv split)
+    big = torch.randn(B, S, H * 2, D, device=DEV).bfloat16()
+    view = big[:, :, :H, :]
+    torch.testing.assert_close(transpose_bshd(view), view.permute(0, 2, 3, 1).contiguous())
