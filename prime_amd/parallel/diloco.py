@@ -38,6 +38,8 @@ class DilocoOptimizer:
         H: int = 100,
         outer_device: str = "auto",
         chunk_elems: int = 256 * 1024 * 1024,  # 1 GB fp32 per streamed chunk
+        elastic=None,  # ElasticWorker (leader rank only)
+        elastic_mode: bool = False,  # True on ALL ranks of an elastic worker
     ):
         self.flat = flat
         self.mesh = mesh
@@ -45,6 +47,12 @@ class DilocoOptimizer:
         self.outer_lr = outer_lr
         self.outer_momentum = outer_momentum
         self.H = H
+        self.elastic = elastic
+        self.elastic_mode = elastic_mode or elastic is not None
+        self._view = None
+        import threading
+
+        self._outer_lock = threading.Lock()
         self.inner_step_count = 0
         self.outer_step_count = 0
 
@@ -94,11 +102,42 @@ class DilocoOptimizer:
         return True
 
     def outer_step(self) -> None:
-        if self.outer_device == "gpu":
-            self._outer_step_resident()
-        else:
-            self._outer_step_streamed()
+        if self.elastic is not None:
+            # boundary: agree on membership + rebuild the cross-worker group
+            self._view = self.elastic.sync()
+        with self._outer_lock:
+            if self.outer_device == "gpu":
+                self._outer_step_resident()
+            else:
+                self._outer_step_streamed()
         self.outer_step_count += 1
+
+    def _allreduce(self, delta_chunk: torch.Tensor) -> None:
+        """Average a (padded) delta chunk across workers.
+
+        Elastic mode: the worker's leader runs the int8 gloo ring over the
+        current epoch's cross-worker group, then the result fans out to the
+        worker's other ranks over the local (RCCL) group. Static mode: the
+        mesh's pre-built outer group (int8 ring over RCCL P2P)."""
+        if self.elastic_mode:
+            if self.elastic is not None:
+                from .elastic import ring_allreduce_int8_pg
+
+                v = self._view
+                if v is not None and v.world > 1:
+                    if delta_chunk.is_cuda:
+                        host = delta_chunk.to("cpu")
+                        ring_allreduce_int8_pg(host, v.pg, v.my_index, v.world)
+                        delta_chunk.copy_(host, non_blocking=True)
+                    else:
+                        ring_allreduce_int8_pg(delta_chunk, v.pg, v.my_index, v.world)
+            if self.mesh.local_group is not None:
+                import torch.distributed as dist
+
+                src = self.mesh.worker_id * self.mesh.cfg.worker_size
+                dist.broadcast(delta_chunk, src=src, group=self.mesh.local_group)
+        else:
+            self.mesh.outer_allreduce_avg(delta_chunk)
 
     def _outer_step_resident(self) -> None:
         f = self.flat
@@ -108,7 +147,7 @@ class DilocoOptimizer:
             ops.pseudograd(theta[:n], f.master32, delta[:n])
             if n < delta.numel():
                 delta[n:].zero_()
-            self.mesh.outer_allreduce_avg(delta)
+            self._allreduce(delta)
             ops.nesterov_outer(
                 theta[:n], f.master32, f.flat_w, self.outer_buf[:n], delta[:n],
                 lr=self.outer_lr, mu=self.outer_momentum,
@@ -117,7 +156,7 @@ class DilocoOptimizer:
             delta[:n] = theta[:n] - f.master32
             if n < delta.numel():
                 delta[n:].zero_()
-            self.mesh.outer_allreduce_avg(delta)
+            self._allreduce(delta)
             buf = self.outer_buf[:n]
             buf.mul_(self.outer_momentum).add_(delta[:n])
             theta[:n].add_(delta[:n] + self.outer_momentum * buf, alpha=-self.outer_lr)
@@ -138,7 +177,7 @@ class DilocoOptimizer:
                 ops.pseudograd(gt[:live], f.master32[c0:c0 + live], gd[:live])
             if live < k:
                 gd[live:k].zero_()
-            self.mesh.outer_allreduce_avg(gd[:k])
+            self._allreduce(gd[:k])
             if live > 0:
                 ops.nesterov_outer(
                     gt[:live], f.master32[c0:c0 + live],
@@ -150,6 +189,31 @@ class DilocoOptimizer:
         torch.cuda.synchronize()
 
     # ------------------------------------------------------------ ckpt
+    def live_state(self) -> dict:
+        """Consistent CPU snapshot for live peer recovery (taken under the
+        outer-step lock so a joiner never sees a half-applied outer step)."""
+        with self._outer_lock:
+            return {
+                "theta_outer": self.theta_outer.detach().to("cpu").clone(),
+                "outer_buf": self.outer_buf.detach().to("cpu").clone(),
+                "inner_step": self.inner_step_count,
+                "outer_step": self.outer_step_count,
+            }
+
+    def load_bootstrap(self, payload: dict) -> None:
+        """Adopt a peer's live state (joining worker)."""
+        n = self.flat.numel_padded
+        theta = payload["theta_outer"]
+        with self._outer_lock:
+            k = min(theta.numel(), self.theta_outer.numel())
+            self.theta_outer[:k].copy_(theta[:k].to(self.theta_outer.device))
+            self.outer_buf[:k].copy_(payload["outer_buf"][:k].to(self.outer_buf.device))
+            self.inner_step_count = int(payload["inner_step"])
+            self.outer_step_count = int(payload["outer_step"])
+            th = self.theta_outer[:n]
+            self.flat.load_flat_(th.to(self.flat.master32.device)
+                                 if th.device != self.flat.master32.device else th)
+
     def state_dict(self) -> dict:
         return {
             "theta_outer": self.theta_outer,

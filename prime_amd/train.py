@@ -40,8 +40,11 @@ class Trainer:
         self.run_dir = Path(run_dir) if run_dir else Path("runs") / cfg.run_name
         self.run_dir.mkdir(parents=True, exist_ok=True)
 
+        world = int(os.environ.get("WORLD_SIZE", 1))
         self.mesh = ElasticDeviceMesh(MeshConfig(
-            worker_size=cfg.parallel.worker_size,
+            # elastic mode: this torchrun job IS one worker; the
+            # cross-worker layer is the dynamic TCPStore fabric
+            worker_size=world if cfg.parallel.elastic else cfg.parallel.worker_size,
             backend=cfg.parallel.backend,
             quant_outer=cfg.diloco.quant_int8,
         ))
@@ -79,13 +82,33 @@ class Trainer:
             self.flat, lr=cfg.optim.lr, betas=tuple(cfg.optim.betas),
             eps=cfg.optim.eps, weight_decay=cfg.optim.weight_decay,
         )
+        self.elastic = None
+        if cfg.parallel.elastic and self.mesh.is_leader:
+            from .parallel.elastic import ElasticWorker
+
+            self.elastic = ElasticWorker(
+                worker_name=cfg.run_name,
+                host_store=os.environ.get("PRIME_GLOBAL_HOST", "0") == "1",
+                heartbeat_interval=cfg.parallel.heartbeat_interval,
+                heartbeat_timeout=cfg.parallel.heartbeat_timeout,
+                ckpt_provider=lambda: self.diloco.live_state(),
+            )
         self.diloco = DilocoOptimizer(
             self.flat, self.mesh, self.inner,
             outer_lr=cfg.diloco.outer_lr,
             outer_momentum=cfg.diloco.outer_momentum,
             H=cfg.diloco.H if cfg.diloco.enabled else 10**9,
             outer_device=cfg.diloco.outer_device,
+            elastic=self.elastic,
+            elastic_mode=cfg.parallel.elastic,
         )
+        if self.elastic is not None:
+            payload = self.elastic.bootstrap_from_peer()
+            if payload is not None:
+                self.diloco.load_bootstrap(payload)
+                self.log.info(
+                    f"live-recovered from peer at outer step {payload['outer_step']}"
+                )
 
         data_cfg = DataConfig(
             kind=cfg.data.kind, path=cfg.data.path, seq_len=cfg.model.seq_len,
@@ -239,6 +262,8 @@ class Trainer:
         return result
 
     def close(self) -> None:
+        if self.elastic is not None:
+            self.elastic.close(leaving=True)
         if self.ckpt:
             self.ckpt.wait()
         self.metrics.close()
