@@ -1,5 +1,6 @@
 from .diloco import DilocoOptimizer
 from .flat import FlatParamSpace, FusedAdamW
+from .fsdp import ShardedParamSpace
 from .mesh import ElasticDeviceMesh, MeshConfig
 from .ring import allreduce_fp32, ring_allreduce_int8
 
@@ -7,6 +8,7 @@ __all__ = [
     "DilocoOptimizer",
     "FlatParamSpace",
     "FusedAdamW",
+    "ShardedParamSpace",
     "ElasticDeviceMesh",
     "MeshConfig",
     "allreduce_fp32",

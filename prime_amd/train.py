@@ -71,12 +71,30 @@ class Trainer:
             self.model = self.model.to(dtype=torch.bfloat16)
         self.model.reset_rope(self.device)  # tables stay fp32
 
-        # broadcast initial params so every rank starts identical
-        self.flat = FlatParamSpace(self.model)
-        if self.mesh.initialized:
-            import torch.distributed as dist
-            dist.broadcast(self.flat.flat_w, src=0)
-            self.flat.master32.copy_(self.flat.flat_w.float())
+        self.fsdp = cfg.parallel.fsdp and cfg.parallel.worker_size > 1
+        if self.fsdp:
+            from .parallel.fsdp import ShardedParamSpace
+
+            if not cfg.model.activation_checkpointing:
+                self.model.activation_checkpointing = True
+            self.flat = ShardedParamSpace(self.model, self.mesh)
+            # shard-aligned broadcast across workers (rank i of each worker
+            # holds the same shard); same-seed init already matches, this
+            # guards against nondeterministic init kernels
+            if self.mesh.outer_group is not None:
+                import torch.distributed as dist
+
+                dist.broadcast(self.flat.flat_w, src=self.mesh.worker_rank,
+                               group=self.mesh.outer_group)
+                self.flat.master32.copy_(self.flat.flat_w.float())
+        else:
+            # broadcast initial params so every rank starts identical
+            self.flat = FlatParamSpace(self.model)
+            if self.mesh.initialized:
+                import torch.distributed as dist
+
+                dist.broadcast(self.flat.flat_w, src=0)
+                self.flat.master32.copy_(self.flat.flat_w.float())
 
         self.inner = FusedAdamW(
             self.flat, lr=cfg.optim.lr, betas=tuple(cfg.optim.betas),
@@ -147,7 +165,10 @@ class Trainer:
             loss = self.model.loss(x, y) / cfg.data.grad_accum
             loss.backward()
             loss_acc = loss.detach() if loss_acc is None else loss_acc + loss.detach()
-        self.mesh.local_allreduce_grad(self.flat.flat_grad)
+        if self.fsdp:
+            self.flat.finalize_grads()  # sharded units were reduce-scattered in bwd
+        else:
+            self.mesh.local_allreduce_grad(self.flat.flat_grad)
         if cfg.optim.grad_clip > 0:
             self.flat.clip_grad_norm_(cfg.optim.grad_clip)
         did_outer = self.diloco.step()

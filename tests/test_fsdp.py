@@ -1,0 +1,100 @@
+"""FSDP sharding tests on CPU/gloo: 2-rank sharded training must match
+2-rank data-parallel training (same seed, same data shards)."""
+import torch
+
+from tests.conftest import run_distributed
+
+
+def _train_worker(rank, world, fsdp, steps):
+    from prime_amd.utils.config import (
+        DilocoConfig, MetricsConfig, ModelConfig, ParallelConfig, TrainConfig,
+    )
+    from prime_amd.train import Trainer
+
+    cfg = TrainConfig(
+        run_name=f"fsdp_{fsdp}",
+        steps=steps,
+        model=ModelConfig(name="llama_test", seq_len=64,
+                          activation_checkpointing=True),
+        diloco=DilocoConfig(H=10**6),  # pure inner loop for this test
+        parallel=ParallelConfig(worker_size=world, fsdp=fsdp),
+        metrics=MetricsConfig(log_interval=100),
+    )
+    cfg.data.micro_batch_size = 2
+    tr = Trainer(cfg, run_dir=f"/tmp/prime_amd_test/fsdp{fsdp}_r{rank}")
+    losses = []
+    for _ in range(steps):
+        losses.append(float(tr.train_step()))
+    # materialize full params for comparison
+    if fsdp:
+        full = {}
+        for u in tr.flat.units:
+            tr.flat.gather_unit(u)
+            for n, p in u.params:
+                full[n] = p.detach().clone()
+            tr.flat.release_unit(u)
+        for n, p in tr.flat.repl_params:
+            full[n] = p.detach().clone()
+    else:
+        full = {n: p.detach().clone() for n, p in tr.model.named_parameters()}
+    tr.close()
+    return {"losses": losses, "w": {k: v.sum().item() for k, v in full.items()},
+            "head": full["layers.0.attn.wqkv.weight"].flatten()[:8].tolist()}
+
+
+def test_fsdp_matches_dp_two_ranks():
+    dp = run_distributed(_train_worker, 2, args=(False, 3), timeout=300)
+    sh = run_distributed(_train_worker, 2, args=(True, 3), timeout=300)
+    # same data, same seed => same loss trajectory (bf16-free CPU fp32 math)
+    for a, b in zip(dp[0]["losses"], sh[0]["losses"]):
+        assert abs(a - b) < 1e-3, (dp[0]["losses"], sh[0]["losses"])
+    # parameters after training match between the two modes
+    for k in dp[0]["w"]:
+        assert abs(dp[0]["w"][k] - sh[0]["w"][k]) < 2e-2, k
+    assert dp[0]["head"] == dp[1]["head"]  # DP replicas in sync
+    assert sh[0]["head"] == sh[1]["head"]  # gathered shards identical
+
+
+def _fsdp_diloco_worker(rank, world):
+    from prime_amd.utils.config import (
+        DilocoConfig, MetricsConfig, ModelConfig, ParallelConfig, TrainConfig,
+    )
+    from prime_amd.train import Trainer
+
+    cfg = TrainConfig(
+        run_name="fsdp_diloco",
+        steps=4,
+        model=ModelConfig(name="llama_test", seq_len=64,
+                          activation_checkpointing=True),
+        diloco=DilocoConfig(H=2),
+        parallel=ParallelConfig(worker_size=world, fsdp=True),
+        metrics=MetricsConfig(log_interval=100),
+    )
+    cfg.data.micro_batch_size = 2
+    tr = Trainer(cfg, run_dir=f"/tmp/prime_amd_test/fsdp_diloco_r{rank}")
+    res = tr.run()
+    out = (res["outer_steps"], float(tr.flat.flat_w.float().sum()))
+    tr.close()
+    return out
+
+
+def test_fsdp_with_diloco_outer():
+    """FSDP inner sharding + DiLoCo outer step run together (single worker
+    of 2 ranks: outer step is identity-average but exercises the path)."""
+    outs = run_distributed(_fsdp_diloco_worker, 2, timeout=300)
+    assert outs[0][0] == 2
+    assert all(torch.isfinite(torch.tensor([o[1] for o in outs])))
+
+
+def test_fsdp_requires_checkpointing():
+    import pytest
+
+    from prime_amd.models import build_model
+    from prime_amd.parallel.fsdp import ShardedParamSpace
+
+    class FakeMesh:
+        local_group = object()
+
+    m = build_model("llama_test")  # no activation checkpointing
+    with pytest.raises(ValueError, match="activation_checkpointing"):
+        ShardedParamSpace(m, FakeMesh())
