@@ -1,0 +1,352 @@
+"""prime-amd CLI — run lifecycle UX modeled on the reference's `prime
+train` command family (reference: prime_cli commands/rl.py — TOML config
+as default argument, strict schema with friendly errors, logs/metrics/
+checkpoints/list/stop/restart verbs, JSON/table dual output).
+
+Unlike the reference (a thin HTTPS client for a hosted service), runs here
+execute locally: the launcher spawns `torch.distributed.run` with one rank
+per GPU and tracks them in a run registry directory.
+"""
+from __future__ import annotations
+
+import json
+import os
+import signal
+import socket
+import subprocess
+import sys
+import time
+import uuid
+from pathlib import Path
+from typing import Optional
+
+import typer
+from typer.core import TyperGroup
+
+from ..utils.config import ConfigError, default_config_toml, load_config
+from ..utils.logging import render_log_line
+
+
+class DefaultRunGroup(TyperGroup):
+    """Makes `prime-amd train cfg.toml` work without the `run` verb
+    (reference: DefaultGroup in prime_cli commands/rl.py:1197-1244)."""
+
+    def resolve_command(self, ctx, args):
+        if args and args[0] not in self.commands and not args[0].startswith("-"):
+            args = ["run", *args]
+        return super().resolve_command(ctx, args)
+
+
+app = typer.Typer(help="MI355X-native DiLoCo training engine", no_args_is_help=True)
+train_app = typer.Typer(help="Launch and manage training runs",
+                        no_args_is_help=True, cls=DefaultRunGroup)
+config_app = typer.Typer(help="CLI configuration", no_args_is_help=True)
+app.add_typer(train_app, name="train")
+app.add_typer(config_app, name="config")
+
+
+def runs_root() -> Path:
+    root = Path(os.environ.get("PRIME_AMD_RUNS_DIR", "~/.prime_amd/runs")).expanduser()
+    root.mkdir(parents=True, exist_ok=True)
+    return root
+
+
+def _find_run(ref: str) -> Path:
+    root = runs_root()
+    d = root / ref
+    if d.exists():
+        return d
+    matches = [p for p in root.iterdir() if p.is_dir() and p.name.startswith(ref)]
+    if len(matches) == 1:
+        return matches[0]
+    if not matches:
+        typer.secho(f"no run matching '{ref}'", fg="red")
+        raise typer.Exit(1)
+    typer.secho(f"ambiguous run '{ref}': {[m.name for m in matches]}", fg="red")
+    raise typer.Exit(1)
+
+
+def _status(run_dir: Path) -> dict:
+    f = run_dir / "status.json"
+    if not f.exists():
+        return {"status": "UNKNOWN"}
+    st = json.loads(f.read_text())
+    if st.get("status") == "RUNNING":
+        pid = st.get("pid")
+        if pid and not _pid_alive(pid):
+            st["status"] = "DIED"
+    return st
+
+
+def _pid_alive(pid: int) -> bool:
+    try:
+        os.kill(pid, 0)
+        return True
+    except OSError:
+        return False
+
+
+def _free_port() -> int:
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+# ------------------------------------------------------------------ train
+@train_app.command("run")
+def run(
+    config: str = typer.Argument(..., help="TOML run config"),
+    nproc: int = typer.Option(0),
+    detach: bool = typer.Option(False, "--detach", "-d"),
+):
+    try:
+        cfg = load_config(config)
+    except ConfigError as e:
+        typer.secho(str(e), fg="red")
+        raise typer.Exit(2)
+    run_id = f"{cfg.run_name}-{uuid.uuid4().hex[:6]}"
+    run_dir = runs_root() / run_id
+    run_dir.mkdir(parents=True)
+    (run_dir / "config.toml").write_text(Path(config).read_text())
+
+    if nproc <= 0:
+        nproc = cfg.parallel.worker_size
+        try:
+            import torch
+
+            if torch.cuda.is_available():
+                nproc = max(nproc, 1)
+        except ImportError:
+            pass
+    cmd = [sys.executable, "-m", "prime_amd.cli.runner", str(run_dir / "config.toml"), str(run_dir)]
+    if nproc > 1:
+        cmd = [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", f"--nproc-per-node={nproc}",
+            "--master-addr", "127.0.0.1", "--master-port", str(_free_port()),
+            "-m", "prime_amd.cli.runner", str(run_dir / "config.toml"), str(run_dir),
+        ]
+    typer.secho(f"run {run_id}: {' '.join(cmd)}", fg="cyan")
+    log = open(run_dir / "launcher.log", "w")
+    proc = subprocess.Popen(cmd, stdout=log, stderr=subprocess.STDOUT,
+                            start_new_session=True)
+    (run_dir / "launcher.pid").write_text(str(proc.pid))
+    (run_dir / "status.json").write_text(json.dumps(
+        {"status": "STARTING", "pid": proc.pid, "started": time.time()}))
+    if detach:
+        typer.echo(f"started (pid {proc.pid}); follow with: prime-amd train logs {run_id} -f")
+        return
+    try:
+        rc = proc.wait()
+    except KeyboardInterrupt:
+        typer.secho("interrupt: stopping run", fg="yellow")
+        os.killpg(proc.pid, signal.SIGTERM)
+        rc = proc.wait()
+    st = _status(run_dir)
+    color = "green" if st.get("status") == "COMPLETED" else "red"
+    typer.secho(f"run {run_id}: {st.get('status')} (rc={rc})", fg=color)
+    if st.get("status") == "COMPLETED":
+        typer.echo(json.dumps(st.get("result", {}), indent=2))
+
+
+@train_app.command("init")
+def init(
+    path: str = typer.Argument("train.toml"),
+    model: str = typer.Option("llama_150m", help="model preset"),
+):
+    """Write a config template (reference: config template generator)."""
+    p = Path(path)
+    if p.exists():
+        typer.secho(f"{p} exists; not overwriting", fg="red")
+        raise typer.Exit(1)
+    p.write_text(default_config_toml(model))
+    typer.secho(f"wrote {p}", fg="green")
+
+
+@train_app.command("list")
+def list_runs(json_out: bool = typer.Option(False, "--json")):
+    rows = []
+    for d in sorted(runs_root().iterdir()):
+        if not d.is_dir():
+            continue
+        st = _status(d)
+        last = _last_metrics(d)
+        rows.append({
+            "run": d.name, "status": st.get("status", "?"),
+            "step": last.get("step", "-"), "loss": last.get("loss", "-"),
+            "tok/s": last.get("tokens_per_sec", "-"),
+        })
+    if json_out:
+        typer.echo(json.dumps(rows, indent=2))
+        return
+    if not rows:
+        typer.echo("no runs")
+        return
+    fmt = "{run:40s} {status:10s} {step:>8} {loss:>10} {tok_s:>12}"
+    typer.echo(fmt.format(run="RUN", status="STATUS", step="STEP", loss="LOSS", tok_s="TOK/S"))
+    for r in rows:
+        loss = f"{r['loss']:.4f}" if isinstance(r["loss"], float) else r["loss"]
+        tps = f"{r['tok/s']:,.0f}" if isinstance(r["tok/s"], float) else r["tok/s"]
+        typer.echo(fmt.format(run=r["run"], status=r["status"], step=str(r["step"]),
+                              loss=loss, tok_s=tps))
+
+
+def _last_metrics(run_dir: Path) -> dict:
+    f = run_dir / "metrics.jsonl"
+    if not f.exists():
+        return {}
+    lines = f.read_text().splitlines()
+    return json.loads(lines[-1]) if lines else {}
+
+
+@train_app.command("logs")
+def logs(
+    run: str = typer.Argument(...),
+    follow: bool = typer.Option(False, "--follow", "-f"),
+    rank: int = typer.Option(0),
+    raw: bool = typer.Option(False, help="raw JSON lines"),
+):
+    d = _find_run(run)
+    f = d / f"rank{rank}.log"
+    if not f.exists():
+        f = d / "launcher.log"
+    if not f.exists():
+        typer.secho("no logs yet", fg="yellow")
+        raise typer.Exit(1)
+    with open(f) as fh:
+        while True:
+            line = fh.readline()
+            if line:
+                if raw:
+                    typer.echo(line.rstrip())
+                else:
+                    rendered = render_log_line(line)
+                    typer.echo(rendered if rendered else line.rstrip())
+            elif follow and _status(d).get("status") in ("RUNNING", "STARTING"):
+                time.sleep(0.5)
+            else:
+                break
+
+
+@train_app.command("metrics")
+def metrics(run: str = typer.Argument(...), last: int = typer.Option(10),
+            json_out: bool = typer.Option(False, "--json")):
+    d = _find_run(run)
+    from ..utils.metrics import read_metrics
+
+    rows = read_metrics(d / "metrics.jsonl")[-last:]
+    if json_out:
+        typer.echo(json.dumps(rows, indent=2))
+        return
+    if not rows:
+        typer.echo("no metrics yet")
+        return
+    hdr = f"{'STEP':>8} {'LOSS':>10} {'TOK/S':>12} {'MFU':>7} {'MS/STEP':>9} {'OUTER':>6}"
+    typer.echo(hdr)
+    for r in rows:
+        typer.echo(f"{r.get('step', 0):>8} {r.get('loss', float('nan')):>10.4f} "
+                   f"{r.get('tokens_per_sec', 0):>12,.0f} {r.get('mfu', 0):>7.3f} "
+                   f"{r.get('ms_per_step', 0):>9.1f} {r.get('outer_steps', 0):>6}")
+
+
+@train_app.command("checkpoints")
+def checkpoints(run: str = typer.Argument(...)):
+    d = _find_run(run)
+    ck = d / "ckpt"
+    if not ck.exists():
+        typer.echo("no checkpoints")
+        return
+    for tag in sorted(ck.iterdir()):
+        if tag.is_dir() and not tag.is_symlink():
+            files = list(tag.glob("*.pt"))
+            size = sum(f.stat().st_size for f in files)
+            typer.echo(f"{tag.name:16s} {len(files)} file(s) {size/1e9:8.2f} GB")
+
+
+@train_app.command("stop")
+def stop(run: str = typer.Argument(...)):
+    d = _find_run(run)
+    st = _status(d)
+    pid = st.get("pid")
+    if not pid or not _pid_alive(pid):
+        typer.secho("not running", fg="yellow")
+        return
+    # exact recorded pid (process group) — never pattern-based
+    try:
+        os.killpg(pid, signal.SIGTERM)
+    except OSError:
+        os.kill(pid, signal.SIGTERM)
+    (d / "status.json").write_text(json.dumps({"status": "STOPPED", "pid": pid,
+                                               "ended": time.time()}))
+    typer.secho(f"stopped {d.name}", fg="green")
+
+
+@train_app.command("restart")
+def restart(run: str = typer.Argument(...), detach: bool = typer.Option(False, "-d")):
+    """Restart a run from its latest checkpoint."""
+    d = _find_run(run)
+    cfg_path = d / "config.toml"
+    if not cfg_path.exists():
+        typer.secho("run has no config.toml", fg="red")
+        raise typer.Exit(1)
+    text = cfg_path.read_text()
+    if "[checkpoint]" in text and "resume" not in text:
+        text = text.replace("[checkpoint]", '[checkpoint]\nresume = "latest"')
+    elif "[checkpoint]" not in text:
+        text += '\n[checkpoint]\nresume = "latest"\n'
+    new_cfg = d / "config.restart.toml"
+    new_cfg.write_text(text)
+    run_cmd(str(new_cfg), detach)
+
+
+def run_cmd(config: str, detach: bool) -> None:
+    run(config, nproc=0, detach=detach)
+
+
+@train_app.command("models")
+def models():
+    """List model presets (reference: `prime train models`)."""
+    from ..models import CONFIGS
+
+    for name, c in CONFIGS.items():
+        typer.echo(f"{name:16s} {c.n_params()/1e9:7.2f}B params  dim={c.dim} "
+                   f"layers={c.n_layers} heads={c.n_heads}/{c.n_kv_heads} "
+                   f"vocab={c.vocab_size}")
+
+
+# ----------------------------------------------------------------- config
+CFG_PATH = Path("~/.prime_amd/config.json").expanduser()
+
+
+@config_app.command("view")
+def config_view():
+    cfg = json.loads(CFG_PATH.read_text()) if CFG_PATH.exists() else {}
+    cfg["runs_dir (effective)"] = str(runs_root())
+    typer.echo(json.dumps(cfg, indent=2))
+
+
+@config_app.command("set")
+def config_set(key: str, value: str):
+    CFG_PATH.parent.mkdir(parents=True, exist_ok=True)
+    cfg = json.loads(CFG_PATH.read_text()) if CFG_PATH.exists() else {}
+    cfg[key] = value
+    CFG_PATH.write_text(json.dumps(cfg, indent=2))
+    typer.secho(f"{key} = {value}", fg="green")
+
+
+# ------------------------------------------------------------------ bench
+@app.command("bench")
+def bench(steps: int = 10, warmup: int = 3, model: str = "intellect_10b"):
+    """Run the flagship benchmark (see bench.py for the driver contract)."""
+    import bench as bench_mod  # noqa: F401 — repo-root bench
+    subprocess.run([sys.executable, str(Path(bench_mod.__file__)),
+                    "--steps", str(steps), "--warmup", str(warmup),
+                    "--model", model], check=True)
+
+
+def main() -> None:
+    app()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,90 @@
+"""CLI tests with typer CliRunner invoking the real app (reference test
+strategy: CLI-level golden output tests, no network/GPU)."""
+import json
+import time
+
+import pytest
+from typer.testing import CliRunner
+
+from prime_amd.cli.main import app
+
+runner = CliRunner()
+
+
+@pytest.fixture()
+def runs_dir(tmp_path, monkeypatch):
+    monkeypatch.setenv("PRIME_AMD_RUNS_DIR", str(tmp_path / "runs"))
+    return tmp_path / "runs"
+
+
+def test_help():
+    r = runner.invoke(app, ["--help"])
+    assert r.exit_code == 0
+    assert "train" in r.output
+
+
+def test_train_init_and_validate(tmp_path, runs_dir):
+    cfg = tmp_path / "t.toml"
+    r = runner.invoke(app, ["train", "init", str(cfg), "--model", "llama_test"])
+    assert r.exit_code == 0 and cfg.exists()
+    # template must be loadable
+    from prime_amd.utils.config import load_config
+
+    assert load_config(cfg).model.name == "llama_test"
+    # refuses overwrite
+    r2 = runner.invoke(app, ["train", "init", str(cfg)])
+    assert r2.exit_code == 1
+
+
+def test_train_rejects_bad_config(tmp_path, runs_dir):
+    bad = tmp_path / "bad.toml"
+    bad.write_text("[model]\nnam = 'x'\n")
+    r = runner.invoke(app, ["train", "run", str(bad)])
+    assert r.exit_code == 2
+    assert "model.nam" in r.output
+
+
+def test_models_listing():
+    r = runner.invoke(app, ["train", "models"])
+    assert r.exit_code == 0
+    assert "intellect_10b" in r.output
+    assert "llama_70b" in r.output
+
+
+def test_full_run_lifecycle(tmp_path, runs_dir):
+    cfg = tmp_path / "run.toml"
+    cfg.write_text(
+        'run_name = "cli_e2e"\nsteps = 3\n'
+        '[model]\nname = "llama_test"\nseq_len = 64\n'
+        '[data]\nmicro_batch_size = 2\n'
+        '[diloco]\nH = 2\n[metrics]\nlog_interval = 1\n'
+    )
+    r = runner.invoke(app, ["train", "run", str(cfg)])
+    assert r.exit_code == 0, r.output
+    assert "COMPLETED" in r.output
+
+    r = runner.invoke(app, ["train", "list"])
+    assert "cli_e2e" in r.output and "COMPLETED" in r.output
+
+    r = runner.invoke(app, ["train", "list", "--json"])
+    rows = json.loads(r.output)
+    run_id = rows[0]["run"]
+
+    r = runner.invoke(app, ["train", "metrics", run_id])
+    assert r.exit_code == 0 and "LOSS" in r.output
+
+    r = runner.invoke(app, ["train", "logs", run_id])
+    assert r.exit_code == 0 and "starting run" in r.output
+
+    r = runner.invoke(app, ["train", "checkpoints", run_id])
+    assert r.exit_code == 0
+
+
+def test_config_view_set(tmp_path, monkeypatch, runs_dir):
+    import prime_amd.cli.main as cli_main  # module (not the shadowing func)
+
+    monkeypatch.setattr(cli_main, "CFG_PATH", tmp_path / "cfg.json")
+    r = runner.invoke(app, ["config", "set", "default_model", "llama_1b"])
+    assert r.exit_code == 0
+    r = runner.invoke(app, ["config", "view"])
+    assert "llama_1b" in r.output
