@@ -81,8 +81,9 @@ def test_full_run_lifecycle(tmp_path, runs_dir):
 
 
 def test_config_view_set(tmp_path, monkeypatch, runs_dir):
-    import prime_amd.cli.main as cli_main  # module (not the shadowing func)
+    import importlib
 
+    cli_main = importlib.import_module("prime_amd.cli.main")
     monkeypatch.setattr(cli_main, "CFG_PATH", tmp_path / "cfg.json")
     r = runner.invoke(app, ["config", "set", "default_model", "llama_1b"])
     assert r.exit_code == 0
