@@ -27,6 +27,9 @@ def main() -> None:
     ap.add_argument("--micro-batch", type=int, default=8)
     ap.add_argument("--h", type=int, default=100, help="DiLoCo inner steps per outer sync")
     ap.add_argument("--no-outer-warmup", action="store_true")
+    ap.add_argument("--ckpt", choices=["on", "off"], default="off",
+                    help="activation checkpointing (off: 288 GB HBM fits the "
+                         "10B config's activations; recompute costs ~25%%)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", 1))
@@ -47,7 +50,7 @@ def main() -> None:
         steps=args.steps,
         model=ModelConfig(
             name=args.model, seq_len=args.seq_len,
-            activation_checkpointing=True,
+            activation_checkpointing=(args.ckpt == "on"),
         ),
         data=DataSection(kind="synthetic", micro_batch_size=args.micro_batch),
         diloco=DilocoConfig(H=args.h, quant_int8=True, outer_device="auto"),
