@@ -42,3 +42,14 @@ def test_defaults():
     assert cfg.parallel.worker_size == 1
     assert cfg.optim.lr == pytest.approx(3e-4)
     assert cfg.checkpoint.interval == 0
+
+
+def test_shipped_configs_load():
+    from pathlib import Path
+
+    cfg_dir = Path(__file__).resolve().parent.parent / "configs"
+    files = sorted(cfg_dir.glob("*.toml"))
+    assert len(files) >= 5
+    for f in files:
+        cfg = load_config(f)
+        assert cfg.steps > 0, f
