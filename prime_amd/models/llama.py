@@ -68,6 +68,11 @@ class MLP(nn.Module):
 
 
 class Block(nn.Module):
+    """Pre-norm block threaded through the fused add+RMSNorm op: carries
+    (delta, residual) instead of a single stream, so every residual add is
+    fused into the following norm's pass (one fewer hidden-stream
+    read+write per block side)."""
+
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
         self.attn_norm = RMSNorm(cfg.dim, cfg.norm_eps)
@@ -75,10 +80,13 @@ class Block(nn.Module):
         self.mlp_norm = RMSNorm(cfg.dim, cfg.norm_eps)
         self.mlp = MLP(cfg)
 
-    def forward(self, x, cos, sin):
-        x = x + self.attn(self.attn_norm(x), cos, sin)
-        x = x + self.mlp(self.mlp_norm(x))
-        return x
+    def forward(self, x, res, cos, sin):
+        y, s = ops.fused_add_rmsnorm(x, res, self.attn_norm.weight,
+                                     self.attn_norm.eps)
+        a = self.attn(y, cos, sin)
+        y2, s2 = ops.fused_add_rmsnorm(a, s, self.mlp_norm.weight,
+                                       self.mlp_norm.eps)
+        return self.mlp(y2), s2
 
 
 class Llama(nn.Module):
@@ -121,13 +129,15 @@ class Llama(nn.Module):
     def forward(self, tokens: torch.Tensor) -> torch.Tensor:
         """tokens [B,S] -> hidden states [B,S,dim] (pre-lm_head)."""
         x = self.tok_embeddings(tokens)
+        res = None
         cos, sin = self.rope_cos, self.rope_sin
         for blk in self.layers:
             if self.activation_checkpointing and self.training:
-                x = checkpoint(blk, x, cos, sin, use_reentrant=False)
+                x, res = checkpoint(blk, x, res, cos, sin, use_reentrant=False)
             else:
-                x = blk(x, cos, sin)
-        return self.norm(x)
+                x, res = blk(x, res, cos, sin)
+        y, _ = ops.fused_add_rmsnorm(x, res, self.norm.weight, self.norm.eps)
+        return y
 
     def loss(self, tokens: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:
         """Fused lm_head + cross-entropy. tokens/targets: [B,S]."""

@@ -21,6 +21,8 @@ _SIGS = {
     # name -> argtypes (all return c_int hipError_t)
     "prime_rmsnorm_fwd": [ctypes.c_void_p] * 5 + [ctypes.c_int64] * 2 + [ctypes.c_double],
     "prime_rmsnorm_bwd": [ctypes.c_void_p] * 7 + [ctypes.c_int64] * 2 + [ctypes.c_double],
+    "prime_add_rmsnorm_fwd": [ctypes.c_void_p] * 6 + [ctypes.c_int64] * 2 + [ctypes.c_double],
+    "prime_add_rmsnorm_bwd": [ctypes.c_void_p] * 7 + [ctypes.c_int64] * 2 + [ctypes.c_double],
     "prime_rope": [ctypes.c_void_p] * 5 + [ctypes.c_int64] * 4 + [ctypes.c_int, ctypes.c_int64],
     "prime_swiglu_fwd": [ctypes.c_void_p] * 3 + [ctypes.c_int64] * 2,
     "prime_swiglu_bwd": [ctypes.c_void_p] * 4 + [ctypes.c_int64] * 2,
@@ -40,10 +42,14 @@ _SIGS = {
 
 
 def _load() -> ctypes.CDLL:
-    if not LIB_PATH.exists():
+    import os
+
+    override = os.environ.get("PRIME_AMD_LIB_PATH")  # A/B testing of builds
+    path = override or str(LIB_PATH)
+    if override is None and not LIB_PATH.exists():
         # last resort: try building (hipcc cross-compiles without a GPU)
         build(verbose=True)
-    lib = ctypes.CDLL(str(LIB_PATH))
+    lib = ctypes.CDLL(path)
     for name, argtypes in _SIGS.items():
         fn = getattr(lib, name)
         fn.argtypes = argtypes

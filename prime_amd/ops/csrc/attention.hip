@@ -26,6 +26,18 @@
 // (dK/dV); tile = 64 x 64. S % 64 == 0 (checked host-side), D in {64,128}.
 #include "common.h"
 
+// SETPRIO(x): s_setprio around MFMA clusters — compile-time toggled for
+// A/B (guide: structure-conditional; +4-7% on some attn shapes, negative
+// on lockstep structures). Build with -DPRIME_SETPRIO=0 to disable.
+#ifndef PRIME_SETPRIO
+#define PRIME_SETPRIO 1
+#endif
+#if PRIME_SETPRIO
+#define SETPRIO(x) __builtin_amdgcn_s_setprio(x)
+#else
+#define SETPRIO(x)
+#endif
+
 typedef __attribute__((ext_vector_type(8))) short short8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((address_space(3))) char lds_char;
@@ -160,7 +172,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
 
     // ---- S = scale * Q K^T
     f32x4 s[4];
-    __builtin_amdgcn_s_setprio(1);
+    SETPRIO(1);
 #pragma unroll
     for (int sub = 0; sub < 4; ++sub) {
       f32x4 acc{0.f, 0.f, 0.f, 0.f};
@@ -169,7 +181,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
         acc = mfma16(qf[ds], ld8_swz<D>(k_lds[idx], sub * 16 + li, ds * 64 + lg * 16), acc);
       s[sub] = acc;
     }
-    __builtin_amdgcn_s_setprio(0);
+    SETPRIO(0);
     // ---- online softmax per q-row (reg r), row owned by 16-lane group
     float alpha[4];
 #pragma unroll
@@ -208,7 +220,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
         st16_swz<64>(p_lds, lg * 4 + r, sub * 16 + li, f2bf(s[sub][r]));
     wave_lds_fence();
     // ---- O += P V
-    __builtin_amdgcn_s_setprio(1);
+    SETPRIO(1);
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
@@ -216,7 +228,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
         o_acc[dt] = mfma16(ld8_swz<64>(p_lds, li, ks * 64 + lg * 16),
                            ld8_swz<64>(vt_lds[idx], dt * 16 + li, ks * 64 + lg * 16),
                            o_acc[dt]);
-    __builtin_amdgcn_s_setprio(0);
+    SETPRIO(0);
   }
   // ---- epilogue
 #pragma unroll
@@ -315,7 +327,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
       stage_tile<64, D>(Kb + (int64_t)(kv + 64) * sks, sks, k_lds[idx ^ 1], threadIdx.x);
       stage_tile<64, D>(Vb + (int64_t)(kv + 64) * svs, svs, v_lds[idx ^ 1], threadIdx.x);
     }
-    __builtin_amdgcn_s_setprio(1);
+    SETPRIO(1);
 #pragma unroll
     for (int sub = 0; sub < 4; ++sub) {
       f32x4 s_acc{0.f, 0.f, 0.f, 0.f}, dp_acc{0.f, 0.f, 0.f, 0.f};
@@ -324,7 +336,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
         s_acc = mfma16(qf[ds], ld8_swz<D>(k_lds[idx], sub * 16 + li, ds * 64 + lg * 16), s_acc);
         dp_acc = mfma16(dof[ds], ld8_swz<D>(v_lds[idx], sub * 16 + li, ds * 64 + lg * 16), dp_acc);
       }
-      __builtin_amdgcn_s_setprio(0);
+      SETPRIO(0);
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int qg = q0 + lg * 4 + r;
@@ -334,11 +346,11 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
         st16_swz<64>(ds_lds, lg * 4 + r, sub * 16 + li,
                      f2bf(p * (dp_acc[r] - dl_r[r]) * scale));
       }
-      __builtin_amdgcn_s_setprio(1);
+      SETPRIO(1);
     }
-    __builtin_amdgcn_s_setprio(0);
+    SETPRIO(0);
     wave_lds_fence();
-    __builtin_amdgcn_s_setprio(1);
+    SETPRIO(1);
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
@@ -346,7 +358,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
         dq_acc[dt] = mfma16(ld8_swz<64>(ds_lds, li, ks * 64 + lg * 16),
                             ld8(Ktb + (int64_t)(dt * 16 + li) * S + kv + ks * 32 + lg * 8),
                             dq_acc[dt]);
-    __builtin_amdgcn_s_setprio(0);
+    SETPRIO(0);
   }
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -423,7 +435,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
         stage_tile<64, D>(Qb + (int64_t)(q0g + 64) * sqs, sqs, q_lds[idx ^ 1], threadIdx.x);
         stage_tile<64, D>(dOb + (int64_t)(q0g + 64) * H * D, (int64_t)H * D, do_lds[idx ^ 1], threadIdx.x);
       }
-      __builtin_amdgcn_s_setprio(1);
+      SETPRIO(1);
 #pragma unroll
       for (int sub = 0; sub < 4; ++sub) {
         f32x4 st_acc{0.f, 0.f, 0.f, 0.f}, dpt_acc{0.f, 0.f, 0.f, 0.f};
@@ -434,7 +446,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
           st_acc = mfma16(kf[ds], qb, st_acc);
           dpt_acc = mfma16(vf[ds], dob, dpt_acc);
         }
-        __builtin_amdgcn_s_setprio(0);
+        SETPRIO(0);
         const int qg = q0g + sub * 16 + li;
         const float lse_q = lse_b[(int64_t)qg * H];
         const float dl_q = dl_b[(int64_t)qg * H];
@@ -447,11 +459,11 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
           st16_swz<64>(dst_lds, lg * 4 + r, sub * 16 + li,
                        f2bf(p * (dpt_acc[r] - dl_q) * scale));
         }
-        __builtin_amdgcn_s_setprio(1);
+        SETPRIO(1);
       }
-      __builtin_amdgcn_s_setprio(0);
+      SETPRIO(0);
       wave_lds_fence();
-      __builtin_amdgcn_s_setprio(1);
+      SETPRIO(1);
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
@@ -461,7 +473,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
           dv_acc[dt] = mfma16(pa, ld8(dOtb + (int64_t)(dt * 16 + li) * S + q0g + ks * 32 + lg * 8), dv_acc[dt]);
           dk_acc[dt] = mfma16(da, ld8(Qtb + (int64_t)(dt * 16 + li) * S + q0g + ks * 32 + lg * 8), dk_acc[dt]);
         }
-      __builtin_amdgcn_s_setprio(0);
+      SETPRIO(0);
     }
   }
 #pragma unroll

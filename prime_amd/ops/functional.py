@@ -348,3 +348,74 @@ def mfma_probe(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
     C = torch.empty(16, 16, device=A.device, dtype=torch.float32)
     check(lib().prime_mfma_probe(stream_of(A), ptr(A), ptr(B), ptr(C)), "mfma_probe")
     return C
+
+
+# ----------------------------------------------- fused residual + RMSNorm
+class _AddRMSNorm(torch.autograd.Function):
+    """(x, res) -> (y, s): s = x (+ res) is the new residual stream,
+    y = rmsnorm(s) * w. One fused pass instead of {add; norm} (and the
+    backward folds the residual-branch grad into the norm dx pass)."""
+
+    @staticmethod
+    def forward(ctx, x, res, w, eps: float):
+        if not _is_hip(x):
+            s = x if res is None else x + res
+            y = ref.rmsnorm(s, w, eps)
+            ctx.save_for_backward(s, w)
+            ctx.eps = eps
+            ctx.cpu = True
+            return y, s
+        x = x.contiguous()
+        res = res.contiguous() if res is not None else None
+        D = x.shape[-1]
+        R = x.numel() // D
+        s = torch.empty_like(x)
+        y = torch.empty_like(x)
+        rstd = torch.empty(R, device=x.device, dtype=torch.float32)
+        check(
+            lib().prime_add_rmsnorm_fwd(
+                stream_of(x), ptr(x), ptr(res), ptr(w), ptr(s), ptr(y),
+                ptr(rstd), R, D, eps,
+            ),
+            "add_rmsnorm_fwd",
+        )
+        ctx.save_for_backward(s, w, rstd)
+        ctx.eps = eps
+        ctx.cpu = False
+        ctx.has_res = res is not None
+        return y, s
+
+    @staticmethod
+    def backward(ctx, dy, ds_res):
+        saved = ctx.saved_tensors
+        if ctx.cpu:
+            s, w = saved
+            s2 = s.detach().float().requires_grad_(True)
+            w2 = w.detach().float().requires_grad_(True)
+            with torch.enable_grad():
+                y = ref.rmsnorm(s2, w2, ctx.eps)
+            gs, gw = torch.autograd.grad(y, [s2, w2], dy.float())
+            if ds_res is not None:
+                gs = gs + ds_res.float()
+            gs = gs.to(dy.dtype)
+            return gs, (gs if ctx.needs_input_grad[1] else None), gw.to(w.dtype), None
+        s, w, rstd = saved
+        dy = dy.contiguous()
+        D = s.shape[-1]
+        R = s.numel() // D
+        ds = torch.empty_like(s)
+        dw = torch.zeros(D, device=s.device, dtype=torch.float32)
+        dres_in = ds_res.contiguous() if ds_res is not None else None
+        check(
+            lib().prime_add_rmsnorm_bwd(
+                stream_of(s), ptr(dy), ptr(dres_in), ptr(s), ptr(w), ptr(rstd),
+                ptr(ds), ptr(dw), R, D, ctx.eps,
+            ),
+            "add_rmsnorm_bwd",
+        )
+        return ds, (ds if ctx.needs_input_grad[1] else None), dw.to(w.dtype), None
+
+
+def fused_add_rmsnorm(x, res, w, eps: float = 1e-5):
+    """Returns (y, s): y = rmsnorm(x + res) * w, s = x + res."""
+    return _AddRMSNorm.apply(x, res, w, eps)

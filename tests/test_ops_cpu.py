@@ -97,3 +97,23 @@ def test_adamw_reference_decreases_simple_loss():
         ref.adamw_step(p32, g, m, v, lr=0.05, beta1=0.9, beta2=0.95,
                        eps=1e-8, wd=0.0, step=step)
     assert p32.abs().max() < 0.5
+
+
+def test_fused_add_rmsnorm_cpu():
+    x = torch.randn(4, 64, requires_grad=True)
+    res = torch.randn(4, 64, requires_grad=True)
+    w = torch.randn(64, requires_grad=True)
+    y, s = ops.fused_add_rmsnorm(x, res, w)
+    torch.testing.assert_close(s, x + res)
+    torch.testing.assert_close(y, ops.rmsnorm((x + res).detach(), w.detach()))
+    (y.sum() + 2 * s.sum()).backward()
+    # analytic check vs separate ops
+    x2 = x.detach().clone().requires_grad_(True)
+    r2 = res.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    s2 = x2 + r2
+    y2 = ops.rmsnorm(s2, w2)
+    (y2.sum() + 2 * s2.sum()).backward()
+    torch.testing.assert_close(x.grad, x2.grad, atol=1e-5, rtol=1e-4)
+    torch.testing.assert_close(res.grad, r2.grad, atol=1e-5, rtol=1e-4)
+    torch.testing.assert_close(w.grad, w2.grad, atol=1e-5, rtol=1e-4)

@@ -228,3 +228,27 @@ def test_transpose_bshd():
     big = torch.randn(B, S, H * 2, D, device=DEV).bfloat16()
     view = big[:, :, :H, :]
     torch.testing.assert_close(transpose_bshd(view), view.permute(0, 2, 3, 1).contiguous())
+
+
+def test_fused_add_rmsnorm_gpu():
+    from prime_amd import ops
+
+    x = _bf(torch.randn(33, 1024)).requires_grad_(True)
+    res = _bf(torch.randn(33, 1024)).requires_grad_(True)
+    w = _bf(torch.randn(1024)).requires_grad_(True)
+    y, s = ops.fused_add_rmsnorm(x, res, w)
+    dy = _bf(torch.randn_like(y.detach()))
+    dsr = _bf(torch.randn_like(s.detach()))
+    torch.autograd.backward([y, s], [dy, dsr])
+
+    xr = x.detach().float().cpu().requires_grad_(True)
+    rr = res.detach().float().cpu().requires_grad_(True)
+    wr = w.detach().float().cpu().requires_grad_(True)
+    sr = xr + rr
+    yr = ops.reference.rmsnorm(sr, wr)
+    torch.autograd.backward([yr, sr], [dy.float().cpu(), dsr.float().cpu()])
+    torch.testing.assert_close(y.detach().float().cpu(), yr.detach(), atol=3e-2, rtol=3e-2)
+    torch.testing.assert_close(s.detach().float().cpu(), sr.detach(), atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(x.grad.float().cpu(), xr.grad, atol=5e-2, rtol=5e-2)
+    torch.testing.assert_close(res.grad.float().cpu(), rr.grad, atol=5e-2, rtol=5e-2)
+    torch.testing.assert_close(w.grad.float().cpu(), wr.grad, atol=0.15, rtol=5e-2)

@@ -5,6 +5,9 @@ Prints achieved TF/s for fwd / bwd-dq / bwd-dkv with causal FLOP counting.
 """
 import sys
 import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 import torch
 
