@@ -21,8 +21,8 @@ _SIGS = {
     # name -> argtypes (all return c_int hipError_t)
     "prime_rmsnorm_fwd": [ctypes.c_void_p] * 5 + [ctypes.c_int64] * 2 + [ctypes.c_double],
     "prime_rmsnorm_bwd": [ctypes.c_void_p] * 7 + [ctypes.c_int64] * 2 + [ctypes.c_double],
-    "prime_add_rmsnorm_fwd": [ctypes.c_void_p] * 6 + [ctypes.c_int64] * 2 + [ctypes.c_double],
-    "prime_add_rmsnorm_bwd": [ctypes.c_void_p] * 7 + [ctypes.c_int64] * 2 + [ctypes.c_double],
+    "prime_add_rmsnorm_fwd": [ctypes.c_void_p] * 7 + [ctypes.c_int64] * 2 + [ctypes.c_double],
+    "prime_add_rmsnorm_bwd": [ctypes.c_void_p] * 8 + [ctypes.c_int64] * 2 + [ctypes.c_double],
     "prime_rope": [ctypes.c_void_p] * 5 + [ctypes.c_int64] * 4 + [ctypes.c_int, ctypes.c_int64],
     "prime_swiglu_fwd": [ctypes.c_void_p] * 3 + [ctypes.c_int64] * 2,
     "prime_swiglu_bwd": [ctypes.c_void_p] * 4 + [ctypes.c_int64] * 2,
@@ -51,7 +51,10 @@ def _load() -> ctypes.CDLL:
         build(verbose=True)
     lib = ctypes.CDLL(path)
     for name, argtypes in _SIGS.items():
-        fn = getattr(lib, name)
+        try:
+            fn = getattr(lib, name)
+        except AttributeError:
+            continue  # older A/B variant builds may lack newest entry points
         fn.argtypes = argtypes
         fn.restype = ctypes.c_int
     return lib
