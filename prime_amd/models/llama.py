@@ -40,7 +40,7 @@ class Attention(nn.Module):
         self.wqkv = nn.Linear(cfg.dim, (cfg.n_heads + 2 * cfg.n_kv_heads) * hd, bias=False)
         self.wo = nn.Linear(cfg.n_heads * hd, cfg.dim, bias=False)
 
-    def forward(self, x, cos, sin):
+    def forward(self, x, cos, sin, cache=None, pos: int = 0):
         B, S, _ = x.shape
         cfg = self.cfg
         hd = cfg.head_dim
@@ -51,8 +51,18 @@ class Attention(nn.Module):
         q = q.view(B, S, cfg.n_heads, hd)
         k = k.view(B, S, cfg.n_kv_heads, hd)
         v = v.view(B, S, cfg.n_kv_heads, hd)
-        q = ops.apply_rope(q, cos, sin)
-        k = ops.apply_rope(k, cos, sin)
+        q = ops.apply_rope(q, cos, sin, pos_offset=pos)
+        k = ops.apply_rope(k, cos, sin, pos_offset=pos)
+        if cache is not None:
+            kc, vc = cache  # [B, Smax, Hkv, hd]
+            kc[:, pos : pos + S] = k
+            vc[:, pos : pos + S] = v
+            if S == 1:  # decode: memory-bound cache-streaming kernel
+                o = ops.attn_decode(q, kc, vc, length=pos + 1)
+                return self.wo(o.reshape(B, 1, cfg.n_heads * hd))
+            # prefill (pos == 0): causal flash over the prompt
+            o = ops.flash_attention(q, k, v, causal=True)
+            return self.wo(o.reshape(B, S, cfg.n_heads * hd))
         o = ops.flash_attention(q, k, v, causal=True)
         return self.wo(o.reshape(B, S, cfg.n_heads * hd))
 
@@ -80,10 +90,10 @@ class Block(nn.Module):
         self.mlp_norm = RMSNorm(cfg.dim, cfg.norm_eps)
         self.mlp = MLP(cfg)
 
-    def forward(self, x, res, cos, sin):
+    def forward(self, x, res, cos, sin, cache=None, pos: int = 0):
         y, s = ops.fused_add_rmsnorm(x, res, self.attn_norm.weight,
                                      self.attn_norm.eps)
-        a = self.attn(y, cos, sin)
+        a = self.attn(y, cos, sin, cache=cache, pos=pos)
         y2, s2 = ops.fused_add_rmsnorm(a, s, self.mlp_norm.weight,
                                        self.mlp_norm.eps)
         return self.mlp(y2), s2
@@ -126,16 +136,18 @@ class Llama(nn.Module):
         self.rope_cos = cos
         self.rope_sin = sin
 
-    def forward(self, tokens: torch.Tensor) -> torch.Tensor:
-        """tokens [B,S] -> hidden states [B,S,dim] (pre-lm_head)."""
+    def forward(self, tokens: torch.Tensor, caches=None, pos: int = 0) -> torch.Tensor:
+        """tokens [B,S] -> hidden states [B,S,dim] (pre-lm_head).
+        caches: optional per-layer (k,v) KV caches for inference."""
         x = self.tok_embeddings(tokens)
         res = None
         cos, sin = self.rope_cos, self.rope_sin
-        for blk in self.layers:
+        for i, blk in enumerate(self.layers):
             if self.activation_checkpointing and self.training:
                 x, res = checkpoint(blk, x, res, cos, sin, use_reentrant=False)
             else:
-                x, res = blk(x, res, cos, sin)
+                x, res = blk(x, res, cos, sin,
+                             cache=caches[i] if caches else None, pos=pos)
         y, _ = ops.fused_add_rmsnorm(x, res, self.norm.weight, self.norm.eps)
         return y
 

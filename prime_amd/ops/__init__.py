@@ -7,6 +7,7 @@ torch.nn.functional.linear (library GEMMs; everything fused is ours).
 """
 from .functional import (
     apply_rope,
+    attn_decode,
     fused_add_rmsnorm,
     cross_entropy,
     dequant_int8,
@@ -26,6 +27,7 @@ from .build import build, LIB_PATH
 
 __all__ = [
     "apply_rope",
+    "attn_decode",
     "fused_add_rmsnorm",
     "cross_entropy",
     "dequant_int8",

@@ -419,3 +419,27 @@ class _AddRMSNorm(torch.autograd.Function):
 def fused_add_rmsnorm(x, res, w, eps: float = 1e-5):
     """Returns (y, s): y = rmsnorm(x + res) * w, s = x + res."""
     return _AddRMSNorm.apply(x, res, w, eps)
+
+
+# ------------------------------------------------------- decode attention
+def attn_decode(q, k_cache, v_cache, length: int) -> torch.Tensor:
+    """Single-token decode: q [B,H,D] (or [B,1,H,D]) against the first
+    `length` rows of k/v caches [B,Smax,Hkv,D]. Returns [B,H,D]."""
+    if q.dim() == 4:
+        q = q.squeeze(1)
+    B, H, D = q.shape
+    _, Smax, Hkv, _ = k_cache.shape
+    if not _is_hip(q):
+        qs = q.unsqueeze(1)  # [B,1,H,D]
+        o = ref.attention(qs, k_cache[:, :length], v_cache[:, :length], causal=False)
+        return o.squeeze(1)
+    q = q.contiguous()
+    o = torch.empty_like(q)
+    check(
+        lib().prime_attn_decode(
+            stream_of(q), ptr(q), ptr(k_cache), ptr(v_cache), ptr(o),
+            B, H, Hkv, Smax, length, D, D**-0.5,
+        ),
+        "attn_decode",
+    )
+    return o

@@ -252,3 +252,33 @@ def test_fused_add_rmsnorm_gpu():
     torch.testing.assert_close(x.grad.float().cpu(), xr.grad, atol=5e-2, rtol=5e-2)
     torch.testing.assert_close(res.grad.float().cpu(), rr.grad, atol=5e-2, rtol=5e-2)
     torch.testing.assert_close(w.grad.float().cpu(), wr.grad, atol=0.15, rtol=5e-2)
+
+
+def test_attn_decode_vs_reference():
+    from prime_amd import ops
+
+    B, H, Hkv, D, Smax, L = 2, 8, 2, 128, 512, 300
+    q = _bf(torch.randn(B, H, D))
+    kc = _bf(torch.randn(B, Smax, Hkv, D))
+    vc = _bf(torch.randn(B, Smax, Hkv, D))
+    got = ops.attn_decode(q, kc, vc, length=L)
+    want = ops.reference.attention(
+        q.float().cpu().unsqueeze(1), kc[:, :L].float().cpu(),
+        vc[:, :L].float().cpu(), causal=False,
+    ).squeeze(1)
+    torch.testing.assert_close(got.float().cpu(), want, atol=3e-2, rtol=3e-2)
+
+
+def test_attn_decode_head_dim_64():
+    from prime_amd import ops
+
+    B, H, Hkv, D, Smax, L = 1, 4, 4, 64, 256, 129
+    q = _bf(torch.randn(B, H, D))
+    kc = _bf(torch.randn(B, Smax, Hkv, D))
+    vc = _bf(torch.randn(B, Smax, Hkv, D))
+    got = ops.attn_decode(q, kc, vc, length=L)
+    want = ops.reference.attention(
+        q.float().cpu().unsqueeze(1), kc[:, :L].float().cpu(),
+        vc[:, :L].float().cpu(), causal=False,
+    ).squeeze(1)
+    torch.testing.assert_close(got.float().cpu(), want, atol=3e-2, rtol=3e-2)

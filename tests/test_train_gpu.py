@@ -77,3 +77,18 @@ def test_gpu_checkpoint_async_roundtrip(tmp_path):
     torch.testing.assert_close(tr2.flat.master32[:1024].cpu(), w)
     assert tr2.diloco.outer_step_count == 1
     tr2.close()
+
+
+def test_generate_on_gpu(tmp_path):
+    import torch
+
+    from prime_amd.models import build_model
+    from prime_amd.models.generate import generate
+
+    torch.manual_seed(0)
+    m = build_model("llama_150m").to("cuda", dtype=torch.bfloat16)
+    m.reset_rope("cuda")
+    prompt = torch.randint(0, m.cfg.vocab_size, (2, 33), device="cuda")
+    out = generate(m, prompt, max_new_tokens=8, temperature=0.0)
+    assert out.shape == (2, 41)
+    assert (out[:, :33] == prompt).all()
