@@ -26,18 +26,6 @@
 // (dK/dV); tile = 64 x 64. S % 64 == 0 (checked host-side), D in {64,128}.
 #include "common.h"
 
-// SETPRIO(x): s_setprio around MFMA clusters — compile-time toggled for
-// A/B (guide: structure-conditional; +4-7% on some attn shapes, negative
-// on lockstep structures). Build with -DPRIME_SETPRIO=0 to disable.
-#ifndef PRIME_SETPRIO
-#define PRIME_SETPRIO 1
-#endif
-#if PRIME_SETPRIO
-#define SETPRIO(x) __builtin_amdgcn_s_setprio(x)
-#else
-#define SETPRIO(x)
-#endif
-
 typedef __attribute__((ext_vector_type(8))) short short8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((address_space(3))) char lds_char;
@@ -138,8 +126,8 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
   const bf16* Kb = K + b * skb + hkv * skh;
   const bf16* Vtb = Vt + ((int64_t)(b * Hkv + hkv) * D) * S;
 
-  __shared__ bf16 k_lds[2][64 * D];
-  __shared__ bf16 vt_lds[2][D * 64];
+  __shared__ bf16 k_lds[64 * D];
+  __shared__ bf16 vt_lds[D * 64];
   __shared__ bf16 p_lds_all[4][16 * 64];
   bf16* p_lds = p_lds_all[wid];
 
@@ -156,32 +144,23 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
   for (int dt = 0; dt < DT; ++dt) o_acc[dt] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int kv_end = causal ? (qt * 64 + 64) : S;
-  // prologue: stage tile 0 into buffer 0; each iteration then prefetches
-  // tile t+1 into the other buffer right after the barrier, so the
-  // barrier's vmcnt drain only waits on loads that had a whole tile's
-  // compute to land (guide §5.5 T3-lite: never a cold drain in the loop)
-  stage_tile<64, D>(Kb, sks, k_lds[0], threadIdx.x);
-  stage_tile<D, 64>(Vtb, S, vt_lds[0], threadIdx.x);
-  int idx = 0;
-  for (int kv = 0; kv < kv_end; kv += 64, idx ^= 1) {
-    __syncthreads();  // tile[idx] staged; everyone done with tile[idx^1]
-    if (kv + 64 < kv_end) {
-      stage_tile<64, D>(Kb + (int64_t)(kv + 64) * sks, sks, k_lds[idx ^ 1], threadIdx.x);
-      stage_tile<D, 64>(Vtb + kv + 64, S, vt_lds[idx ^ 1], threadIdx.x);
-    }
+  for (int kv = 0; kv < kv_end; kv += 64) {
+    __syncthreads();  // all waves done reading the previous tile
+    stage_tile<64, D>(Kb + (int64_t)kv * sks, sks, k_lds, threadIdx.x);
+    // Vt tile: rows d (stride S), cols k in [kv, kv+64)
+    stage_tile<D, 64>(Vtb + kv, S, vt_lds, threadIdx.x);
+    __syncthreads();  // staged (syncthreads drains vmcnt)
 
     // ---- S = scale * Q K^T
     f32x4 s[4];
-    SETPRIO(1);
 #pragma unroll
     for (int sub = 0; sub < 4; ++sub) {
       f32x4 acc{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ds = 0; ds < DS; ++ds)
-        acc = mfma16(qf[ds], ld8_swz<D>(k_lds[idx], sub * 16 + li, ds * 64 + lg * 16), acc);
+        acc = mfma16(qf[ds], ld8_swz<D>(k_lds, sub * 16 + li, ds * 64 + lg * 16), acc);
       s[sub] = acc;
     }
-    SETPRIO(0);
     // ---- online softmax per q-row (reg r), row owned by 16-lane group
     float alpha[4];
 #pragma unroll
@@ -220,15 +199,13 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
         st16_swz<64>(p_lds, lg * 4 + r, sub * 16 + li, f2bf(s[sub][r]));
     wave_lds_fence();
     // ---- O += P V
-    SETPRIO(1);
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks)
         o_acc[dt] = mfma16(ld8_swz<64>(p_lds, li, ks * 64 + lg * 16),
-                           ld8_swz<64>(vt_lds[idx], dt * 16 + li, ks * 64 + lg * 16),
+                           ld8_swz<64>(vt_lds, dt * 16 + li, ks * 64 + lg * 16),
                            o_acc[dt]);
-    SETPRIO(0);
   }
   // ---- epilogue
 #pragma unroll
@@ -296,8 +273,9 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
   const float* lse_b = lse + (int64_t)b * S * H + h;    // stride H
   const float* dl_b = delta + (int64_t)b * S * H + h;
 
-  __shared__ bf16 k_lds[2][64 * D];
-  __shared__ bf16 v_lds[2][64 * D];
+  __shared__ bf16 k_lds[64 * D];
+  __shared__ bf16 v_lds[64 * D];
+  __shared__ bf16 kt_lds[D * 64];
   __shared__ bf16 ds_lds_all[4][16 * 64];
   bf16* ds_lds = ds_lds_all[wid];
 
@@ -318,25 +296,20 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
   for (int dt = 0; dt < DT; ++dt) dq_acc[dt] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int kv_end = causal ? (qt * 64 + 64) : S;
-  stage_tile<64, D>(Kb, sks, k_lds[0], threadIdx.x);
-  stage_tile<64, D>(Vb, svs, v_lds[0], threadIdx.x);
-  int idx = 0;
-  for (int kv = 0; kv < kv_end; kv += 64, idx ^= 1) {
+  for (int kv = 0; kv < kv_end; kv += 64) {
     __syncthreads();
-    if (kv + 64 < kv_end) {
-      stage_tile<64, D>(Kb + (int64_t)(kv + 64) * sks, sks, k_lds[idx ^ 1], threadIdx.x);
-      stage_tile<64, D>(Vb + (int64_t)(kv + 64) * svs, svs, v_lds[idx ^ 1], threadIdx.x);
-    }
-    SETPRIO(1);
+    stage_tile<64, D>(Kb + (int64_t)kv * sks, sks, k_lds, threadIdx.x);
+    stage_tile<64, D>(Vb + (int64_t)kv * svs, svs, v_lds, threadIdx.x);
+    stage_tile<D, 64>(Ktb + kv, S, kt_lds, threadIdx.x);
+    __syncthreads();
 #pragma unroll
     for (int sub = 0; sub < 4; ++sub) {
       f32x4 s_acc{0.f, 0.f, 0.f, 0.f}, dp_acc{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ds = 0; ds < DS; ++ds) {
-        s_acc = mfma16(qf[ds], ld8_swz<D>(k_lds[idx], sub * 16 + li, ds * 64 + lg * 16), s_acc);
-        dp_acc = mfma16(dof[ds], ld8_swz<D>(v_lds[idx], sub * 16 + li, ds * 64 + lg * 16), dp_acc);
+        s_acc = mfma16(qf[ds], ld8_swz<D>(k_lds, sub * 16 + li, ds * 64 + lg * 16), s_acc);
+        dp_acc = mfma16(dof[ds], ld8_swz<D>(v_lds, sub * 16 + li, ds * 64 + lg * 16), dp_acc);
       }
-      SETPRIO(0);
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int qg = q0 + lg * 4 + r;
@@ -346,19 +319,15 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
         st16_swz<64>(ds_lds, lg * 4 + r, sub * 16 + li,
                      f2bf(p * (dp_acc[r] - dl_r[r]) * scale));
       }
-      SETPRIO(1);
     }
-    SETPRIO(0);
     wave_lds_fence();
-    SETPRIO(1);
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks)
         dq_acc[dt] = mfma16(ld8_swz<64>(ds_lds, li, ks * 64 + lg * 16),
-                            ld8(Ktb + (int64_t)(dt * 16 + li) * S + kv + ks * 32 + lg * 8),
+                            ld8_swz<64>(kt_lds, dt * 16 + li, ks * 64 + lg * 16),
                             dq_acc[dt]);
-    SETPRIO(0);
   }
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -397,8 +366,10 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
   const bf16* Kb = K + b * skb + hkv * skh;
   const bf16* Vb = V + b * svb + hkv * svh;
 
-  __shared__ bf16 q_lds[2][64 * D];
-  __shared__ bf16 do_lds[2][64 * D];
+  __shared__ bf16 q_lds[64 * D];
+  __shared__ bf16 do_lds[64 * D];
+  __shared__ bf16 qt_lds[D * 64];
+  __shared__ bf16 dot_lds[D * 64];
   __shared__ bf16 pt_lds_all[4][16 * 64];
   __shared__ bf16 dst_lds_all[4][16 * 64];
   bf16* pt_lds = pt_lds_all[wid];
@@ -426,27 +397,23 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
     const float* lse_b = lse + (int64_t)b * S * H + h;
     const float* dl_b = delta + (int64_t)b * S * H + h;
     const int q_start = causal ? kt * 64 : 0;
-    stage_tile<64, D>(Qb + (int64_t)q_start * sqs, sqs, q_lds[0], threadIdx.x);
-    stage_tile<64, D>(dOb + (int64_t)q_start * H * D, (int64_t)H * D, do_lds[0], threadIdx.x);
-    int idx = 0;
-    for (int q0g = q_start; q0g < S; q0g += 64, idx ^= 1) {
+    for (int q0g = q_start; q0g < S; q0g += 64) {
       __syncthreads();
-      if (q0g + 64 < S) {
-        stage_tile<64, D>(Qb + (int64_t)(q0g + 64) * sqs, sqs, q_lds[idx ^ 1], threadIdx.x);
-        stage_tile<64, D>(dOb + (int64_t)(q0g + 64) * H * D, (int64_t)H * D, do_lds[idx ^ 1], threadIdx.x);
-      }
-      SETPRIO(1);
+      stage_tile<64, D>(Qb + (int64_t)q0g * sqs, sqs, q_lds, threadIdx.x);
+      stage_tile<64, D>(dOb + (int64_t)q0g * H * D, (int64_t)H * D, do_lds, threadIdx.x);
+      stage_tile<D, 64>(Qtb + q0g, S, qt_lds, threadIdx.x);
+      stage_tile<D, 64>(dOtb + q0g, S, dot_lds, threadIdx.x);
+      __syncthreads();
 #pragma unroll
       for (int sub = 0; sub < 4; ++sub) {
         f32x4 st_acc{0.f, 0.f, 0.f, 0.f}, dpt_acc{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int ds = 0; ds < DS; ++ds) {
-          const short8 qb = ld8_swz<D>(q_lds[idx], sub * 16 + li, ds * 64 + lg * 16);
-          const short8 dob = ld8_swz<D>(do_lds[idx], sub * 16 + li, ds * 64 + lg * 16);
+          const short8 qb = ld8_swz<D>(q_lds, sub * 16 + li, ds * 64 + lg * 16);
+          const short8 dob = ld8_swz<D>(do_lds, sub * 16 + li, ds * 64 + lg * 16);
           st_acc = mfma16(kf[ds], qb, st_acc);
           dpt_acc = mfma16(vf[ds], dob, dpt_acc);
         }
-        SETPRIO(0);
         const int qg = q0g + sub * 16 + li;
         const float lse_q = lse_b[(int64_t)qg * H];
         const float dl_q = dl_b[(int64_t)qg * H];
@@ -459,21 +426,17 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
           st16_swz<64>(dst_lds, lg * 4 + r, sub * 16 + li,
                        f2bf(p * (dpt_acc[r] - dl_q) * scale));
         }
-        SETPRIO(1);
       }
-      SETPRIO(0);
       wave_lds_fence();
-      SETPRIO(1);
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
           const short8 pa = ld8_swz<64>(pt_lds, li, ks * 64 + lg * 16);
           const short8 da = ld8_swz<64>(dst_lds, li, ks * 64 + lg * 16);
-          dv_acc[dt] = mfma16(pa, ld8(dOtb + (int64_t)(dt * 16 + li) * S + q0g + ks * 32 + lg * 8), dv_acc[dt]);
-          dk_acc[dt] = mfma16(da, ld8(Qtb + (int64_t)(dt * 16 + li) * S + q0g + ks * 32 + lg * 8), dk_acc[dt]);
+          dv_acc[dt] = mfma16(pa, ld8_swz<64>(dot_lds, dt * 16 + li, ks * 64 + lg * 16), dv_acc[dt]);
+          dk_acc[dt] = mfma16(da, ld8_swz<64>(qt_lds, dt * 16 + li, ks * 64 + lg * 16), dk_acc[dt]);
         }
-      SETPRIO(0);
     }
   }
 #pragma unroll
