@@ -334,6 +334,63 @@ def config_set(key: str, value: str):
     typer.secho(f"{key} = {value}", fg="green")
 
 
+# --------------------------------------------------------------- generate
+@app.command("generate")
+def generate_cmd(
+    model: str = typer.Option("llama_150m", help="model preset"),
+    prompt_tokens: str = typer.Option("1,2,3,4", help="comma-separated token ids"),
+    max_new: int = typer.Option(32),
+    temperature: float = typer.Option(0.0),
+    top_k: int = typer.Option(0),
+    seed: int = typer.Option(0),
+    checkpoint: Optional[str] = typer.Option(None, help="checkpoint dir (latest tag)"),
+):
+    """Generate tokens with the KV-cache decode path (greedy by default)."""
+    import torch
+
+    from ..models import build_model
+    from ..models.generate import generate
+
+    torch.manual_seed(seed)
+    dev = "cuda" if torch.cuda.is_available() else "cpu"
+    m = build_model(model)
+    if dev == "cuda":
+        m = m.to(dev, dtype=torch.bfloat16)
+    m.reset_rope(dev)
+    if checkpoint:
+        from ..ckpt import CheckpointManager
+        from ..parallel.flat import FlatParamSpace
+
+        flat = FlatParamSpace(m)
+        payload = CheckpointManager(checkpoint).load(map_location=dev)
+        if payload is None:
+            typer.secho("no checkpoint found", fg="red")
+            raise typer.Exit(1)
+        flat.load_flat_(payload["tensors"]["master32"].to(dev))
+    toks = torch.tensor([[int(t) for t in prompt_tokens.split(",")]], device=dev)
+    out = generate(m, toks, max_new, temperature=temperature, top_k=top_k,
+                   seed=seed)
+    typer.echo(",".join(str(int(t)) for t in out[0]))
+
+
+# ------------------------------------------------------------------ store
+@app.command("store")
+def store_cmd(port: int = typer.Option(29777), addr: str = typer.Option("0.0.0.0")):
+    """Host a standalone TCPStore for the elastic cross-worker fabric
+    (workers point PRIME_GLOBAL_ADDR/PORT here; survives worker churn)."""
+    import time as _time
+
+    from torch.distributed import TCPStore
+
+    TCPStore(addr, port, is_master=True, wait_for_workers=False)
+    typer.secho(f"elastic store listening on {addr}:{port} (ctrl-c to stop)", fg="green")
+    try:
+        while True:
+            _time.sleep(3600)
+    except KeyboardInterrupt:
+        pass
+
+
 # ------------------------------------------------------------------ bench
 @app.command("bench")
 def bench(steps: int = 10, warmup: int = 3, model: str = "intellect_10b"):

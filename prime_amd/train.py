@@ -56,6 +56,12 @@ class Trainer:
         self.metrics = MetricsWriter(
             self.run_dir / "metrics.jsonl" if self.mesh.rank == 0 else None
         )
+        from .utils.metrics import WandbWriter
+
+        self.wandb = WandbWriter(
+            cfg.metrics.wandb and self.mesh.rank == 0, cfg.run_name,
+            {"model": cfg.model.name, "H": cfg.diloco.H},
+        )
 
         torch.manual_seed(1234)
         # construct directly on the target device: CPU-side init of a 10B
@@ -253,6 +259,10 @@ class Trainer:
                 tps = self.tokens_per_step * window_steps / dt
                 tps_gpu = tps / max(1, self.mesh.world_size)
                 last_loss = float(loss)
+                self.wandb.write(
+                    self.step_count, loss=last_loss, tokens_per_sec=tps,
+                    mfu=mfu(tps_gpu, self.flops_per_token),
+                )
                 self.metrics.write(
                     self.step_count, loss=last_loss, tokens_per_sec=tps,
                     tokens_per_sec_per_gpu=tps_gpu,
@@ -283,6 +293,7 @@ class Trainer:
         return result
 
     def close(self) -> None:
+        self.wandb.close()
         if self.elastic is not None:
             self.elastic.close(leaving=True)
         if self.ckpt:

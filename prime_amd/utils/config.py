@@ -74,6 +74,7 @@ class CheckpointConfig(_Strict):
 class MetricsConfig(_Strict):
     log_interval: int = 10
     jsonl: bool = True
+    wandb: bool = False
 
 
 class TrainConfig(_Strict):

@@ -51,3 +51,29 @@ def read_metrics(path: str | Path) -> list[dict]:
         except json.JSONDecodeError:
             continue
     return out
+
+
+class WandbWriter:
+    """Optional Weights & Biases sink (reference forwards W&B config to its
+    trainer; here it is a local opt-in). No-op unless wandb is importable
+    AND enabled in the run config."""
+
+    def __init__(self, enabled: bool, run_name: str, config: dict | None = None):
+        self._run = None
+        if not enabled:
+            return
+        try:
+            import wandb  # noqa: F401
+
+            self._run = wandb.init(project="prime-amd", name=run_name,
+                                   config=config or {})
+        except Exception:  # noqa: BLE001 — wandb absent/offline: stay silent
+            self._run = None
+
+    def write(self, step: int, **metrics) -> None:
+        if self._run is not None:
+            self._run.log(metrics, step=step)
+
+    def close(self) -> None:
+        if self._run is not None:
+            self._run.finish()
