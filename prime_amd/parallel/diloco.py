@@ -27,6 +27,16 @@ _RING_ALIGN = 64 * QBLK
 _HOST_THRESHOLD = 4_000_000_000  # params; above this "auto" offloads to host
 
 
+def _mem_available_bytes() -> int | None:
+    try:
+        for line in open("/proc/meminfo"):
+            if line.startswith("MemAvailable:"):
+                return int(line.split()[1]) * 1024
+    except OSError:
+        pass
+    return None
+
+
 class DilocoOptimizer:
     def __init__(
         self,
@@ -76,13 +86,29 @@ class DilocoOptimizer:
             self._n_comm = (n + self.chunk - 1) // self.chunk * self.chunk
 
             def _host_buf(k: int) -> torch.Tensor:
-                try:
-                    return torch.zeros(k, dtype=torch.float32, pin_memory=True)
-                except RuntimeError:  # pinned pool exhausted: pageable fallback
-                    return torch.zeros(k, dtype=torch.float32)
+                # pinning is a throughput optimization (PCIe at full rate);
+                # skip it when the node can't afford it: 8 ranks x 84 GB of
+                # pinned pages would exhaust a typical host. Never risk the
+                # OOM-killer for an every-H-steps transfer.
+                import os
+
+                world = int(os.environ.get("WORLD_SIZE", 1))
+                want_pin = os.environ.get("PRIME_AMD_OUTER_PIN", "auto")
+                if want_pin != "0":
+                    need = 2 * k * 4 * world  # theta+buf across all local ranks
+                    avail = _mem_available_bytes()
+                    if want_pin == "1" or (avail is not None and need < 0.5 * avail):
+                        try:
+                            return torch.zeros(k, dtype=torch.float32, pin_memory=True)
+                        except RuntimeError:
+                            pass
+                return torch.zeros(k, dtype=torch.float32)
 
             self.theta_outer = _host_buf(self._n_comm)
-            self.theta_outer[:n].copy_(flat.master32.to("cpu"))
+            # chunked D2H init (no 40 GB host temp from master32.to("cpu"))
+            for c0 in range(0, n, self.chunk):
+                c1 = min(c0 + self.chunk, n)
+                self.theta_outer[c0:c1].copy_(flat.master32[c0:c1])
             self.outer_buf = _host_buf(self._n_comm)
             dev = flat.device
             self._g_theta = torch.zeros(self.chunk, device=dev, dtype=torch.float32)
