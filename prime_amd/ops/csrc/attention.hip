@@ -346,16 +346,19 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
     const bf16* __restrict__ K, const bf16* __restrict__ V,
     const bf16* __restrict__ dO, const bf16* __restrict__ dOt,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    bf16* __restrict__ dK, bf16* __restrict__ dV, int B, int H, int Hkv,
-    int S, float scale, int causal,
+    float* __restrict__ wsK, float* __restrict__ wsV, int B, int H, int Hkv,
+    int S, float scale, int causal, int splits,
     int64_t sqb, int64_t sqs, int64_t sqh,
     int64_t skb, int64_t sks, int64_t skh,
     int64_t svb, int64_t svs, int64_t svh) {
   constexpr int DS = D / 32;
   constexpr int DT = D / 16;
   const int n_kt = S / 64;
-  const int bh = blockIdx.x / n_kt;
-  const int kt = blockIdx.x - bh * n_kt;
+  const int per_split = (B * Hkv) * n_kt;
+  const int split = blockIdx.x / per_split;
+  const int rem = blockIdx.x - split * per_split;
+  const int bh = rem / n_kt;
+  const int kt = rem - bh * n_kt;
   const int b = bh / Hkv, hkv = bh - b * Hkv;
   const int group = H / Hkv;
   const int wid = threadIdx.x >> 6;
@@ -396,8 +399,8 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
     const bf16* dOtb = dOt + ((int64_t)(b * H + h) * D) * S;
     const float* lse_b = lse + (int64_t)b * S * H + h;
     const float* dl_b = delta + (int64_t)b * S * H + h;
-    const int q_start = causal ? kt * 64 : 0;
-    for (int q0g = q_start; q0g < S; q0g += 64) {
+    const int q_start = (causal ? kt * 64 : 0) + split * 64;
+    for (int q0g = q_start; q0g < S; q0g += 64 * splits) {
       __syncthreads();
       stage_tile<64, D>(Qb + (int64_t)q0g * sqs, sqs, q_lds, threadIdx.x);
       stage_tile<64, D>(dOb + (int64_t)q0g * H * D, (int64_t)H * D, do_lds, threadIdx.x);
@@ -439,16 +442,42 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
         }
     }
   }
+  // fp32 partials: ws[split][b][hkv][kg][d]
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int kg = k0 + lg * 4 + r;
-    bf16* krow = dK + (((int64_t)(b * S + kg)) * Hkv + hkv) * D;
-    bf16* vrow = dV + (((int64_t)(b * S + kg)) * Hkv + hkv) * D;
+    const int64_t base = ((((int64_t)split * B + b) * Hkv + hkv) * S + kg) * D;
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt) {
-      krow[dt * 16 + li] = f2bf(dk_acc[dt][r]);
-      vrow[dt * 16 + li] = f2bf(dv_acc[dt][r]);
+      wsK[base + dt * 16 + li] = dk_acc[dt][r];
+      wsV[base + dt * 16 + li] = dv_acc[dt][r];
     }
+  }
+}
+
+// reduce fp32 split-partials -> bf16 dK/dV in [B,S,Hkv,D]
+__global__ void dkv_reduce_kernel(const float* __restrict__ wsK,
+                                  const float* __restrict__ wsV,
+                                  bf16* __restrict__ dK, bf16* __restrict__ dV,
+                                  int B, int Hkv, int S, int D, int splits) {
+  const int64_t n = (int64_t)B * Hkv * S * D;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float k = 0.f, v = 0.f;
+    for (int sp = 0; sp < splits; ++sp) {
+      k += wsK[(int64_t)sp * n + i];
+      v += wsV[(int64_t)sp * n + i];
+    }
+    // i decodes as [b][hkv][s][d] -> output [b][s][hkv][d]
+    const int d = (int)(i % D);
+    int64_t r = i / D;
+    const int sq = (int)(r % S);
+    r /= S;
+    const int hk = (int)(r % Hkv);
+    const int bb = (int)(r / Hkv);
+    const int64_t o = (((int64_t)(bb * S + sq)) * Hkv + hk) * D + d;
+    dK[o] = f2bf(k);
+    dV[o] = f2bf(v);
   }
 }
 
@@ -510,18 +539,25 @@ PRIME_API int prime_flash_bwd_dkv(hipStream_t stream, const void* Q,
                                   const void* Qt, const void* K, const void* V,
                                   const void* dO, const void* dOt,
                                   const void* lse, const void* delta, void* dK,
-                                  void* dV, int64_t B, int64_t H, int64_t Hkv,
-                                  int64_t S, int64_t D, double scale,
-                                  int64_t causal,
+                                  void* dV, void* wsK, void* wsV,
+                                  int64_t splits, int64_t B, int64_t H,
+                                  int64_t Hkv, int64_t S, int64_t D,
+                                  double scale, int64_t causal,
                                   int64_t sqb, int64_t sqs, int64_t sqh,
                                   int64_t skb, int64_t sks, int64_t skh,
                                   int64_t svb, int64_t svs, int64_t svh) {
   if (S % 64 != 0 || (D != 64 && D != 128)) return hipErrorInvalidValue;
-  const int grid = (int)(B * Hkv * (S / 64));
+  const int grid = (int)(B * Hkv * (S / 64) * splits);
   DISPATCH_D(flash_bwd_dkv_kernel, (const bf16*)Q, (const bf16*)Qt,
              (const bf16*)K, (const bf16*)V, (const bf16*)dO, (const bf16*)dOt,
-             (const float*)lse, (const float*)delta, (bf16*)dK, (bf16*)dV,
-             (int)B, (int)H, (int)Hkv, (int)S, (float)scale, (int)causal, sqb,
-             sqs, sqh, skb, sks, skh, svb, svs, svh);
+             (const float*)lse, (const float*)delta, (float*)wsK, (float*)wsV,
+             (int)B, (int)H, (int)Hkv, (int)S, (float)scale, (int)causal,
+             (int)splits, sqb, sqs, sqh, skb, sks, skh, svb, svs, svh);
+  int err = hipGetLastError();
+  if (err) return err;
+  int rgrid = prime_grid(B * Hkv * S * D, 256);
+  hipLaunchKernelGGL(dkv_reduce_kernel, dim3(rgrid), dim3(256), 0, stream,
+                     (const float*)wsK, (const float*)wsV, (bf16*)dK,
+                     (bf16*)dV, (int)B, (int)Hkv, (int)S, (int)D, (int)splits);
   return (int)hipGetLastError();
 }

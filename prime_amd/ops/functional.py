@@ -235,10 +235,17 @@ class _FlashAttention(torch.autograd.Function):
         dot = transpose_bshd(do)
         dk = torch.empty(B, S, Hkv, D, device=q.device, dtype=q.dtype)
         dv = torch.empty(B, S, Hkv, D, device=q.device, dtype=q.dtype)
+        # split the causal q loop across the grid: the unsplit grid is only
+        # B*Hkv*S/64 blocks with 1..S/64 trip-count imbalance
+        base_blocks = B * Hkv * (S // 64)
+        splits = max(1, min(8, 6144 // max(1, base_blocks), S // 64))
+        ws = torch.empty(2, splits, B, Hkv, S, D, device=q.device,
+                         dtype=torch.float32)
         check(
             lib().prime_flash_bwd_dkv(
                 stream_of(q), ptr(q), ptr(qt), ptr(k), ptr(v), ptr(do),
                 ptr(dot), ptr(lse), ptr(delta), ptr(dk), ptr(dv),
+                ptr(ws[0]), ptr(ws[1]), splits,
                 B, H, Hkv, S, D, scale, int(causal),
                 q.stride(0), q.stride(1), q.stride(2),
                 k.stride(0), k.stride(1), k.stride(2),
