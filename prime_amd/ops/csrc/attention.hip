@@ -116,7 +116,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
   const int bh = blockIdx.x / n_qt;
   // longest-trip q-tiles first: causal trip count is qt+1, so schedule
   // large qt early to keep the tail of the wavefront busy
-  const int qt = n_qt - 1 - (blockIdx.x - bh * n_qt);
+  const int qt = n_qt - 1 - (blockIdx.x - bh * n_qt);  // longest trips first
   const int b = bh / H, h = bh - b * H;
   const int hkv = h / (H / Hkv);
   const int wid = threadIdx.x >> 6;
