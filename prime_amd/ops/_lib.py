@@ -37,6 +37,7 @@ _SIGS = {
     "prime_flash_bwd_dq": [ctypes.c_void_p] * 9 + [ctypes.c_int64] * 5 + [ctypes.c_double] + [ctypes.c_int64] * 10,
     "prime_flash_bwd_dkv": [ctypes.c_void_p] * 13 + [ctypes.c_int64] * 6 + [ctypes.c_double] + [ctypes.c_int64] * 10,
     "prime_mfma_probe": [ctypes.c_void_p] * 4,
+    "prime_mfma_probe32": [ctypes.c_void_p] * 4,
     "prime_transpose_bshd": [ctypes.c_void_p] * 3 + [ctypes.c_int64] * 7,
     "prime_attn_decode": [ctypes.c_void_p] * 5 + [ctypes.c_int64] * 6 + [ctypes.c_double],
 }

@@ -450,3 +450,10 @@ def attn_decode(q, k_cache, v_cache, length: int) -> torch.Tensor:
         "attn_decode",
     )
     return o
+
+
+def mfma_probe32(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
+    """32x32x16 bf16 MFMA single-tile matmul (layout verification)."""
+    C = torch.empty(32, 32, device=A.device, dtype=torch.float32)
+    check(lib().prime_mfma_probe32(stream_of(A), ptr(A), ptr(B), ptr(C)), "mfma_probe32")
+    return C

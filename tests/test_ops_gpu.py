@@ -282,3 +282,14 @@ def test_attn_decode_head_dim_64():
         vc[:, :L].float().cpu(), causal=False,
     ).squeeze(1)
     torch.testing.assert_close(got.float().cpu(), want, atol=3e-2, rtol=3e-2)
+
+
+def test_mfma32_layout_vs_matmul():
+    from prime_amd.ops.functional import mfma_probe32
+
+    A = torch.randn(32, 16).clamp(-2, 2)
+    B = torch.randn(16, 32).clamp(-2, 2)
+    got = mfma_probe32(_bf(A).contiguous(), _bf(B).contiguous())
+    torch.cuda.synchronize()
+    want = _bf(A).float() @ _bf(B).float()
+    torch.testing.assert_close(got.cpu(), want.cpu(), atol=1e-3, rtol=1e-3)
