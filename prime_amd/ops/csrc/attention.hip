@@ -25,6 +25,7 @@
 // Workgroup = 4 waves; each wave owns 16 q-rows (fwd/dQ) or 16 k-rows
 // (dK/dV); tile = 64 x 64. S % 64 == 0 (checked host-side), D in {64,128}.
 #include "common.h"
+#include <stdlib.h>
 
 typedef __attribute__((ext_vector_type(8))) short short8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
@@ -221,6 +222,197 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
     if (li == 0)
       lse[((int64_t)(b * S + qg)) * H + h] = m[r] + __logf(fmaxf(l[r], 1e-30f));
   }
+}
+
+
+// ------------------------------------------------- forward v4 (32x32 MFMA)
+// Swapped-operand structure (guide §B attn, 8-warp ladder): St = K·Q^T with
+// v_mfma_f32_32x32x16_bf16 puts each q-row's scores LANE-LOCAL (lane pair
+// (l, l^32) splits the row), so the online softmax is in-register (no
+// cross-16-lane shuffle chains) and P feeds the PV A-fragment directly
+// after ONE lane-pair exchange — the per-tile P LDS bounce of the 16x16
+// kernel disappears. 4 waves x 32 q-rows = 128-row q tiles.
+//
+// 32x32x16 fragment maps (HW-verified, test_mfma32_layout_vs_matmul):
+//   A[32][16]: lane holds A[lane&31][(lane>>5)*8 + j]
+//   B[16][32]: lane holds B[(lane>>5)*8 + j][lane&31]
+//   C[32][32]: lane holds C[(r&3) + 8*(r>>2) + 4*(lane>>5)][lane&31]
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+__device__ __forceinline__ f32x16 mfma32(short8 a, short8 b, f32x16 c) {
+  return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+}
+
+template <int D>  // D == 128 only
+__global__ __launch_bounds__(256) void flash_fwd32_kernel(
+    const bf16* __restrict__ Q, const bf16* __restrict__ K,
+    const bf16* __restrict__ Vt, bf16* __restrict__ O, float* __restrict__ lse,
+    int B, int H, int Hkv, int S, float scale, int causal,
+    int64_t sqb, int64_t sqs, int64_t sqh,
+    int64_t skb, int64_t sks, int64_t skh) {
+  constexpr int DSL = D / 16;  // 16-wide d slices for the K dim (8)
+  const int n_qt = S / 128;
+  const int bh = blockIdx.x / n_qt;
+  const int qt = n_qt - 1 - (blockIdx.x - bh * n_qt);  // longest trips first
+  const int b = bh / H, h = bh - b * H;
+  const int hkv = h / (H / Hkv);
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lo = lane & 31, hi = lane >> 5;
+  const int q0w = qt * 128 + wid * 32;  // this wave's first q row
+
+  const bf16* Qb = Q + b * sqb + h * sqh;
+  const bf16* Kb = K + b * skb + hkv * skh;
+  const bf16* Vtb = Vt + ((int64_t)(b * Hkv + hkv) * D) * S;
+
+  __shared__ bf16 k_lds[64 * D];
+  __shared__ bf16 vt_lds[D * 64];
+  __shared__ float bcast_all[4][32];  // per-wave alpha / inv-l broadcast
+  float* bcast = bcast_all[wid];
+
+  // Q^T B-fragments, pre-scaled by 1/sqrt(D): lane holds Q[q0w+lo][ds*16+hi*8+j]
+  short8 qf[DSL];
+#pragma unroll
+  for (int ds = 0; ds < DSL; ++ds) {
+    const bf16* src = Qb + (int64_t)(q0w + lo) * sqs + ds * 16 + hi * 8;
+    short8 raw = ld8(src);
+    short8 sc;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      bf16 v = reinterpret_cast<const bf16*>(&raw)[j];
+      reinterpret_cast<bf16*>(&sc)[j] = f2bf(bf2f(v) * scale);
+    }
+    qf[ds] = sc;
+  }
+
+  float m_run = NEG_INF, l_run = 0.f;
+  f32x16 o_acc[4];  // O[32q x 128d]: 4 col-tiles of 32 d
+#pragma unroll
+  for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) o_acc[dt][r] = 0.f;
+
+  const int kv_end = causal ? (qt * 128 + 128) : S;
+  for (int kv = 0; kv < kv_end; kv += 64) {
+    __syncthreads();
+    stage_tile<64, D>(Kb + (int64_t)kv * sks, sks, k_lds, threadIdx.x);
+    stage_tile<D, 64>(Vtb + kv, S, vt_lds, threadIdx.x);
+    __syncthreads();
+
+    // ---- St = (K q^T): two 32x32 C tiles over the 64-key block.
+    // A-frag rows = k (lane&31), cols = d slice; read from swizzled K LDS.
+    f32x16 st[2];
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) st[t][r] = 0.f;
+#pragma unroll
+      for (int ds = 0; ds < DSL; ++ds) {
+        const short8 kf = ld8_swz<D>(k_lds, t * 32 + lo, ds * 32 + hi * 16);
+        st[t] = mfma32(kf, qf[ds], st[t]);
+      }
+    }
+    // ---- online softmax: lane pair (lo, hi) and (lo, hi^1) split the
+    // 64 scores of q row q0w+lo; st[t][r] is k = kv + t*32 + crow(r,hi)
+    float tmax = NEG_INF;
+#pragma unroll
+    for (int t = 0; t < 2; ++t)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kg = kv + t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        if (causal && kg > q0w + lo) st[t][r] = NEG_INF;
+        tmax = fmaxf(tmax, st[t][r]);
+      }
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+    const float mnew = fmaxf(m_run, tmax);
+    const float alpha = __expf(m_run - mnew);
+    float psum = 0.f;
+#pragma unroll
+    for (int t = 0; t < 2; ++t)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const float pv = (st[t][r] <= NEG_INF) ? 0.f : __expf(st[t][r] - mnew);
+        st[t][r] = pv;
+        psum += pv;
+      }
+    psum += __shfl_xor(psum, 32, 64);
+    l_run = l_run * alpha + psum;
+    m_run = mnew;
+
+    // ---- O *= alpha, broadcast per C row (rows are crow(r,hi), but alpha
+    // lives on lane q=lo): per-wave LDS broadcast
+    if (hi == 0) bcast[lo] = alpha;
+    wave_lds_fence();
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r)
+        o_acc[dt][r] *= bcast[(r & 3) + 8 * (r >> 2) + 4 * hi];
+
+    // ---- P -> PV A-fragments: pack to bf16 and exchange the partner
+    // half-chunks (4 values = 2 dwords) per 16-k slice
+    short8 pa[4];
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+#pragma unroll
+      for (int kst = 0; kst < 2; ++kst) {
+        // my kept chunk: r group g_keep = 2*kst + hi; my sent chunk g_send
+        // = 2*kst + (1-hi); both packed as 2 dwords of 2 bf16
+        const int g_keep = 2 * kst + hi;
+        const int g_send = 2 * kst + 1 - hi;
+        unsigned keep0, keep1, send0, send1;
+        {
+          const bf16 a0 = f2bf(st[t][4 * g_keep + 0]), a1 = f2bf(st[t][4 * g_keep + 1]);
+          const bf16 a2 = f2bf(st[t][4 * g_keep + 2]), a3 = f2bf(st[t][4 * g_keep + 3]);
+          keep0 = (unsigned)reinterpret_cast<const unsigned short&>(a0) |
+                  ((unsigned)reinterpret_cast<const unsigned short&>(a1) << 16);
+          keep1 = (unsigned)reinterpret_cast<const unsigned short&>(a2) |
+                  ((unsigned)reinterpret_cast<const unsigned short&>(a3) << 16);
+          const bf16 b0 = f2bf(st[t][4 * g_send + 0]), b1 = f2bf(st[t][4 * g_send + 1]);
+          const bf16 b2 = f2bf(st[t][4 * g_send + 2]), b3 = f2bf(st[t][4 * g_send + 3]);
+          send0 = (unsigned)reinterpret_cast<const unsigned short&>(b0) |
+                  ((unsigned)reinterpret_cast<const unsigned short&>(b1) << 16);
+          send1 = (unsigned)reinterpret_cast<const unsigned short&>(b2) |
+                  ((unsigned)reinterpret_cast<const unsigned short&>(b3) << 16);
+        }
+        const unsigned got0 = __shfl_xor((int)send0, 32, 64);
+        const unsigned got1 = __shfl_xor((int)send1, 32, 64);
+        // A-frag element order j0..j7 = k 16*kst + 8*hi + j; first 4 come
+        // from the hi'=0 owner, last 4 from the hi'=1 owner
+        unsigned w0, w1, w2, w3;
+        if (hi == 0) { w0 = keep0; w1 = keep1; w2 = got0; w3 = got1; }
+        else         { w0 = got0;  w1 = got1;  w2 = keep0; w3 = keep1; }
+        short8 frag;
+        reinterpret_cast<unsigned*>(&frag)[0] = w0;
+        reinterpret_cast<unsigned*>(&frag)[1] = w1;
+        reinterpret_cast<unsigned*>(&frag)[2] = w2;
+        reinterpret_cast<unsigned*>(&frag)[3] = w3;
+        pa[t * 2 + kst] = frag;
+      }
+    }
+    // ---- O += P V  (B-frags from swizzled Vt LDS)
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks)
+        o_acc[dt] = mfma32(pa[ks],
+                           ld8_swz<64>(vt_lds, dt * 32 + lo, ks * 32 + hi * 16),
+                           o_acc[dt]);
+  }
+  // ---- epilogue: O /= l (per-row broadcast), write + lse
+  if (hi == 0) bcast[lo] = (l_run > 0.f) ? 1.f / l_run : 0.f;
+  wave_lds_fence();
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int qr = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    const float inv = bcast[qr];
+    bf16* orow = O + (((int64_t)(b * S + q0w + qr)) * H + h) * D;
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt)
+      orow[dt * 32 + lo] = f2bf(o_acc[dt][r] * inv);
+  }
+  if (hi == 0)
+    lse[((int64_t)(b * S + q0w + lo)) * H + h] = m_run + __logf(fmaxf(l_run, 1e-30f));
 }
 
 // ------------------------------------------------------- delta = rowsum(dO*O)
@@ -501,6 +693,16 @@ PRIME_API int prime_flash_fwd(hipStream_t stream, const void* Q, const void* K,
                               int64_t sqb, int64_t sqs, int64_t sqh,
                               int64_t skb, int64_t sks, int64_t skh) {
   if (S % 64 != 0 || (D != 64 && D != 128)) return hipErrorInvalidValue;
+  static const char* v4env = getenv("PRIME_ATTN_V4");
+  const bool use_v4 = (D == 128) && (S % 128 == 0) && !(v4env && v4env[0] == '0');
+  if (use_v4) {
+    const int grid = (int)(B * H * (S / 128));
+    hipLaunchKernelGGL(flash_fwd32_kernel<128>, dim3(grid), dim3(256), 0,
+                       stream, (const bf16*)Q, (const bf16*)K, (const bf16*)Vt,
+                       (bf16*)O, (float*)lse, (int)B, (int)H, (int)Hkv, (int)S,
+                       (float)scale, (int)causal, sqb, sqs, sqh, skb, sks, skh);
+    return (int)hipGetLastError();
+  }
   const int grid = (int)(B * H * (S / 64));
   DISPATCH_D(flash_fwd_kernel, (const bf16*)Q, (const bf16*)K, (const bf16*)Vt,
              (bf16*)O, (float*)lse, (int)B, (int)H, (int)Hkv, (int)S,
