@@ -46,7 +46,9 @@ app.add_typer(config_app, name="config")
 
 
 def runs_root() -> Path:
-    root = Path(os.environ.get("PRIME_AMD_RUNS_DIR", "~/.prime_amd/runs")).expanduser()
+    from ..utils.contexts import Contexts
+
+    root = Path(str(Contexts().current()["runs_dir"])).expanduser()
     root.mkdir(parents=True, exist_ok=True)
     return root
 
@@ -315,23 +317,59 @@ def models():
 
 
 # ----------------------------------------------------------------- config
-CFG_PATH = Path("~/.prime_amd/config.json").expanduser()
+# named contexts with env-var precedence (reference: core/config.py)
+from ..utils.contexts import Contexts  # noqa: E402
 
 
 @config_app.command("view")
 def config_view():
-    cfg = json.loads(CFG_PATH.read_text()) if CFG_PATH.exists() else {}
-    cfg["runs_dir (effective)"] = str(runs_root())
-    typer.echo(json.dumps(cfg, indent=2))
+    """Effective settings (defaults < context < PRIME_AMD_* env vars)."""
+    typer.echo(json.dumps(Contexts().current(), indent=2))
 
 
 @config_app.command("set")
 def config_set(key: str, value: str):
-    CFG_PATH.parent.mkdir(parents=True, exist_ok=True)
-    cfg = json.loads(CFG_PATH.read_text()) if CFG_PATH.exists() else {}
-    cfg[key] = value
-    CFG_PATH.write_text(json.dumps(cfg, indent=2))
+    Contexts().set(key, value)
     typer.secho(f"{key} = {value}", fg="green")
+
+
+@config_app.command("save")
+def config_save(name: str):
+    """Save the current config as a named context."""
+    try:
+        p = Contexts().save(name)
+    except ValueError as e:
+        typer.secho(str(e), fg="red")
+        raise typer.Exit(1)
+    typer.secho(f"saved context '{name}' -> {p}", fg="green")
+
+
+@config_app.command("use")
+def config_use(name: str):
+    """Switch to a saved context."""
+    try:
+        Contexts().use(name)
+    except (ValueError, FileNotFoundError) as e:
+        typer.secho(str(e), fg="red")
+        raise typer.Exit(1)
+    typer.secho(f"now using context '{name}'", fg="green")
+
+
+@config_app.command("envs")
+def config_envs():
+    """List saved contexts."""
+    for name in Contexts().list():
+        typer.echo(name)
+
+
+@config_app.command("delete")
+def config_delete(name: str):
+    try:
+        Contexts().delete(name)
+    except ValueError as e:
+        typer.secho(str(e), fg="red")
+        raise typer.Exit(1)
+    typer.secho(f"deleted context '{name}'", fg="green")
 
 
 # --------------------------------------------------------------- generate

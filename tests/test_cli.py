@@ -81,11 +81,32 @@ def test_full_run_lifecycle(tmp_path, runs_dir):
 
 
 def test_config_view_set(tmp_path, monkeypatch, runs_dir):
-    import importlib
-
-    cli_main = importlib.import_module("prime_amd.cli.main")
-    monkeypatch.setattr(cli_main, "CFG_PATH", tmp_path / "cfg.json")
+    monkeypatch.setenv("PRIME_AMD_HOME", str(tmp_path / "home2"))
     r = runner.invoke(app, ["config", "set", "default_model", "llama_1b"])
     assert r.exit_code == 0
     r = runner.invoke(app, ["config", "view"])
     assert "llama_1b" in r.output
+
+
+def test_config_contexts(tmp_path, monkeypatch, runs_dir):
+    monkeypatch.setenv("PRIME_AMD_HOME", str(tmp_path / "home"))
+    r = runner.invoke(app, ["config", "set", "default_model", "llama_8b"])
+    assert r.exit_code == 0
+    r = runner.invoke(app, ["config", "save", "prod"])
+    assert r.exit_code == 0
+    runner.invoke(app, ["config", "set", "default_model", "llama_test"])
+    r = runner.invoke(app, ["config", "view"])
+    assert "llama_test" in r.output
+    r = runner.invoke(app, ["config", "use", "prod"])
+    assert r.exit_code == 0
+    r = runner.invoke(app, ["config", "view"])
+    assert "llama_8b" in r.output
+    r = runner.invoke(app, ["config", "envs"])
+    assert "prod" in r.output
+    # env-var precedence
+    monkeypatch.setenv("PRIME_AMD_DEFAULT_MODEL", "llama_70b")
+    r = runner.invoke(app, ["config", "view"])
+    assert "llama_70b" in r.output
+    # path traversal rejected
+    r = runner.invoke(app, ["config", "save", "../evil"])
+    assert r.exit_code == 1
