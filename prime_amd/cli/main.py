@@ -372,6 +372,50 @@ def config_delete(name: str):
     typer.secho(f"deleted context '{name}'", fg="green")
 
 
+# ------------------------------------------------------------------- eval
+@app.command("eval")
+def eval_cmd(
+    model: str = typer.Option("llama_150m"),
+    data: Optional[str] = typer.Option(None, help="token file (.bin); default synthetic"),
+    seq_len: int = typer.Option(1024),
+    batches: int = typer.Option(10),
+    micro_batch: int = typer.Option(4),
+    checkpoint: Optional[str] = typer.Option(None, help="checkpoint dir"),
+    json_out: bool = typer.Option(False, "--json"),
+):
+    """Held-out loss / perplexity (reference: prime eval surface)."""
+    import torch
+
+    from ..data import DataConfig
+    from ..models import build_model
+    from ..models.evaluate import evaluate_perplexity
+
+    dev = "cuda" if torch.cuda.is_available() else "cpu"
+    torch.manual_seed(0)
+    m = build_model(model)
+    if dev == "cuda":
+        m = m.to(dev, dtype=torch.bfloat16)
+        m.reset_rope(dev)
+    if checkpoint:
+        from ..ckpt import CheckpointManager
+        from ..parallel.flat import FlatParamSpace
+
+        flat = FlatParamSpace(m)
+        payload = CheckpointManager(checkpoint).load(map_location=dev)
+        if payload is None:
+            typer.secho("no checkpoint found", fg="red")
+            raise typer.Exit(1)
+        flat.load_flat_(payload["tensors"]["master32"].to(dev))
+    dc = DataConfig(kind="token_file" if data else "synthetic", path=data,
+                    seq_len=seq_len, micro_batch_size=micro_batch)
+    res = evaluate_perplexity(m, dc, n_batches=batches, device=dev)
+    if json_out:
+        typer.echo(json.dumps(res))
+    else:
+        typer.echo(f"loss {res['loss']:.4f}  ppl {res['perplexity']:.2f}  "
+                   f"({res['tokens']:,} tokens)")
+
+
 # --------------------------------------------------------------- generate
 @app.command("generate")
 def generate_cmd(

@@ -110,3 +110,12 @@ def test_config_contexts(tmp_path, monkeypatch, runs_dir):
     # path traversal rejected
     r = runner.invoke(app, ["config", "save", "../evil"])
     assert r.exit_code == 1
+
+
+def test_eval_command(runs_dir):
+    r = runner.invoke(app, ["eval", "--model", "llama_test", "--seq-len", "64",
+                            "--batches", "2", "--micro-batch", "2", "--json"])
+    assert r.exit_code == 0, r.output
+    res = json.loads(r.output)
+    assert res["tokens"] == 2 * 2 * 64
+    assert res["perplexity"] > 1
