@@ -94,14 +94,21 @@ class ElasticDeviceMesh:
         flat_grad.div_(self.cfg.worker_size)
 
     def outer_allreduce_avg(self, delta32: torch.Tensor) -> None:
-        """Cross-worker pseudo-gradient average (int8 ring by default)."""
+        """Cross-worker pseudo-gradient average (int8 ring by default;
+        multi-ring over distinct xGMI neighbor offsets when the worker
+        count and buffer size allow — see ring.ring_allreduce_int8_multi)."""
         if self.outer_group is None and self.n_workers == 1:
             return
         group = self.outer_group
-        if self.cfg.quant_outer:
-            ring.ring_allreduce_int8(delta32, group=group, average=True)
-        else:
+        if not self.cfg.quant_outer:
             ring.allreduce_fp32(delta32, group=group, average=True)
+            return
+        W = self.n_workers
+        R = len([o for o in range(1, W) if ring._gcd(o, W) == 1])
+        if W > 2 and R > 1 and delta32.numel() % (R * W * ring.QBLK) == 0:
+            ring.ring_allreduce_int8_multi(delta32, group=group, average=True)
+        else:
+            ring.ring_allreduce_int8(delta32, group=group, average=True)
 
     def barrier(self) -> None:
         if self.initialized:

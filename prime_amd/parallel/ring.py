@@ -105,3 +105,130 @@ def allreduce_fp32(delta: torch.Tensor, group=None, average: bool = True) -> Non
     dist.all_reduce(delta, group=group)
     if average:
         delta.div_(dist.get_world_size(group))
+
+
+def ring_allreduce_int8_multi(
+    delta: torch.Tensor,
+    group: dist.ProcessGroup | None = None,
+    average: bool = True,
+    n_rings: int | None = None,
+) -> None:
+    """Multi-ring int8 all-reduce: the buffer is split into `n_rings`
+    sub-buffers, each reduced over a DIFFERENT ring orientation (ring r
+    steps by r+1 ranks), so every rank drives up to n_rings xGMI
+    point-to-point links concurrently instead of serializing on one
+    (xGMI is 7 p2p links/GPU at ~153 GB/s each — a single ring is
+    per-link bound). All rings' sends/recvs for a hop are
+    issued in ONE batch_isend_irecv so RCCL can overlap them.
+
+    Ring r's step offset is o = r+1; that ring's "next" is (rank+o) % W.
+    gcd(o, W) != 1 would split the ring into cycles, so only coprime
+    offsets are used (W=8: o in {1,3,5,7} -> up to 4 concurrent rings).
+    """
+    W = dist.get_world_size(group)
+    if W == 1:
+        return
+    rank = dist.get_rank(group)
+    offsets = [o for o in range(1, W) if _gcd(o, W) == 1]
+    if n_rings is not None:
+        offsets = offsets[:n_rings]
+    R = len(offsets)
+    if R <= 1:
+        return ring_allreduce_int8(delta, group, average)
+    n = delta.numel()
+    assert n % (R * W * QBLK) == 0, (
+        f"delta size {n} not divisible by R*W*QBLK={R * W * QBLK}"
+    )
+    sub = n // R
+    subs = [delta[r * sub : (r + 1) * sub] for r in range(R)]
+    part = sub // W
+    g = lambda r: dist.get_global_rank(group, r) if group is not None else r
+
+    recv_q = [torch.empty(part, dtype=torch.int8, device=delta.device) for _ in range(R)]
+    recv_s = [torch.empty(part // QBLK, dtype=torch.float32, device=delta.device) for _ in range(R)]
+
+    # ---- reduce-scatter: W-1 hops; each hop issues all rings together
+    for step in range(W - 1):
+        p2p, metas = [], []
+        for ri, o in enumerate(offsets):
+            # ring ri in the basis of offset o: logical position of this
+            # rank is rank * inv(o) ... simpler: walk indices directly:
+            # at step s, send partition owned (start - s) in ring order.
+            nxt, prv = (rank + o) % W, (rank - o) % W
+            send_idx = _ring_pos(rank, o, W, -step)
+            recv_idx = _ring_pos(rank, o, W, -step - 1)
+            q, sc = _quant(subs[ri][send_idx * part : (send_idx + 1) * part])
+            q, sc = q.contiguous(), sc.contiguous()
+            p2p += [
+                dist.P2POp(dist.isend, q, g(nxt), group),
+                dist.P2POp(dist.isend, sc, g(nxt), group),
+                dist.P2POp(dist.irecv, recv_q[ri], g(prv), group),
+                dist.P2POp(dist.irecv, recv_s[ri], g(prv), group),
+            ]
+            metas.append((ri, recv_idx))
+        for w in dist.batch_isend_irecv(p2p):
+            w.wait()
+        for ri, recv_idx in metas:
+            _dequant_add(recv_q[ri], recv_s[ri],
+                         subs[ri][recv_idx * part : (recv_idx + 1) * part],
+                         accumulate=True)
+    # ---- all-gather: W-1 hops
+    send_q, send_s = [], []
+    for ri, o in enumerate(offsets):
+        own = _ring_pos(rank, o, W, 1)
+        seg = subs[ri][own * part : (own + 1) * part]
+        if average:
+            seg.div_(W)
+        q, sc = _quant(seg)
+        q, sc = q.contiguous(), sc.contiguous()
+        _dequant_add(q, sc, seg, accumulate=False)
+        send_q.append(q)
+        send_s.append(sc)
+    for step in range(W - 1):
+        p2p, metas = [], []
+        for ri, o in enumerate(offsets):
+            nxt, prv = (rank + o) % W, (rank - o) % W
+            recv_idx = _ring_pos(rank, o, W, -step)
+            p2p += [
+                dist.P2POp(dist.isend, send_q[ri], g(nxt), group),
+                dist.P2POp(dist.isend, send_s[ri], g(nxt), group),
+                dist.P2POp(dist.irecv, recv_q[ri], g(prv), group),
+                dist.P2POp(dist.irecv, recv_s[ri], g(prv), group),
+            ]
+            metas.append((ri, recv_idx))
+        for w in dist.batch_isend_irecv(p2p):
+            w.wait()
+        for ri, recv_idx in metas:
+            _dequant_add(recv_q[ri], recv_s[ri],
+                         subs[ri][recv_idx * part : (recv_idx + 1) * part],
+                         accumulate=False)
+            send_q[ri], recv_q[ri] = recv_q[ri].clone(), send_q[ri]
+            send_s[ri], recv_s[ri] = recv_s[ri].clone(), send_s[ri]
+
+
+def _gcd(a: int, b: int) -> int:
+    while b:
+        a, b = b, a % b
+    return a
+
+
+def _ring_pos(rank: int, offset: int, W: int, hop: int) -> int:
+    """Partition index held by `rank` after `hop` logical ring steps in the
+    ring with step `offset`. In ring coordinates u = position of rank
+    (rank = u*offset mod W), the standard schedule assigns partition
+    ((u + hop) mod W) mapped back to partition ids = rank-space ids."""
+    # position u of this rank in ring order: u * offset ≡ rank (mod W)
+    u = (rank * _modinv(offset, W)) % W
+    v = (u + hop) % W
+    return (v * offset) % W
+
+
+def _modinv(a: int, m: int) -> int:
+    # a coprime to m
+    x0, x1 = 0, 1
+    mm, aa = m, a
+    while aa > 1:
+        q = aa // mm
+        aa, mm = mm, aa - q * mm
+        x1, x0 = x0, x1 - q * x0
+    return x1 % m

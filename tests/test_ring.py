@@ -48,3 +48,39 @@ def test_ring_fp32_exact():
     outs = run_distributed(_ring_worker, 2, args=(False,))
     for err, _, _ in outs:
         assert err < 1e-6
+
+
+def _multi_ring_worker(rank, world):
+    import torch.distributed as dist
+
+    from prime_amd.parallel import ring
+    from prime_amd.ops import QBLK
+
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    offsets = [o for o in range(1, world) if ring._gcd(o, world) == 1]
+    R = len(offsets)
+    torch.manual_seed(7)
+    n = R * world * QBLK * 2
+    full = torch.randn(world, n)
+    mine = full[rank].clone()
+    want = full.mean(0)
+    ring.ring_allreduce_int8_multi(mine, average=True)
+    err = (mine - want).abs().max().item()
+    scale = full.abs().max().item()
+    dist.barrier()
+    dist.destroy_process_group()
+    return err, scale, mine[:8].tolist()
+
+
+def test_multi_ring_int8_four_ranks():
+    outs = run_distributed(_multi_ring_worker, 4)
+    for err, scale, _ in outs:
+        assert err < scale * 8 / 127 + 1e-6, err
+    assert outs[0][2] == outs[1][2] == outs[3][2]
+
+
+def test_multi_ring_int8_three_ranks():
+    outs = run_distributed(_multi_ring_worker, 3)
+    for err, scale, _ in outs:
+        assert err < scale * 6 / 127 + 1e-6, err
+    assert outs[0][2] == outs[2][2]
