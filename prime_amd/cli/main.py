@@ -173,8 +173,12 @@ def list_runs(json_out: bool = typer.Option(False, "--json")):
             continue
         st = _status(d)
         last = _last_metrics(d)
+        status = st.get("status", "?")
+        fa = st.get("failure_analysis")
+        if fa:
+            status = f"{status}:{fa['category']}"
         rows.append({
-            "run": d.name, "status": st.get("status", "?"),
+            "run": d.name, "status": status,
             "step": last.get("step", "-"), "loss": last.get("loss", "-"),
             "tok/s": last.get("tokens_per_sec", "-"),
         })
@@ -184,7 +188,7 @@ def list_runs(json_out: bool = typer.Option(False, "--json")):
     if not rows:
         typer.echo("no runs")
         return
-    fmt = "{run:40s} {status:10s} {step:>8} {loss:>10} {tok_s:>12}"
+    fmt = "{run:40s} {status:22s} {step:>8} {loss:>10} {tok_s:>12}"
     typer.echo(fmt.format(run="RUN", status="STATUS", step="STEP", loss="LOSS", tok_s="TOK/S"))
     for r in rows:
         loss = f"{r['loss']:.4f}" if isinstance(r["loss"], float) else r["loss"]
@@ -273,14 +277,14 @@ def stop(run: str = typer.Argument(...)):
     if not pid or not _pid_alive(pid):
         typer.secho("not running", fg="yellow")
         return
-    # exact recorded pid (process group) — never pattern-based
+    # exact recorded pid (process group) — never pattern-based. The runner
+    # handles SIGTERM gracefully: checkpoint at the next step boundary,
+    # then it writes the final STOPPED status itself.
     try:
         os.killpg(pid, signal.SIGTERM)
     except OSError:
         os.kill(pid, signal.SIGTERM)
-    (d / "status.json").write_text(json.dumps({"status": "STOPPED", "pid": pid,
-                                               "ended": time.time()}))
-    typer.secho(f"stopped {d.name}", fg="green")
+    typer.secho(f"stop requested for {d.name} (graceful: checkpoints first)", fg="green")
 
 
 @train_app.command("restart")

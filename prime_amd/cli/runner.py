@@ -5,6 +5,7 @@ from __future__ import annotations
 
 import json
 import os
+import signal
 import sys
 import time
 from pathlib import Path
@@ -13,24 +14,35 @@ from pathlib import Path
 def main() -> int:
     cfg_path, run_dir = sys.argv[1], Path(sys.argv[2])
     rank = int(os.environ.get("RANK", 0))
-    from prime_amd.train import train_from_config
+    from prime_amd.train import Trainer
     from prime_amd.utils.config import load_config
+    from prime_amd.utils.failures import classify_failure
 
     cfg = load_config(cfg_path)
     if rank == 0:
         (run_dir / "status.json").write_text(json.dumps(
             {"status": "RUNNING", "pid": os.getpid(), "started": time.time(),
              "steps_total": cfg.steps}))
+    trainer = Trainer(cfg, run_dir)
+
+    def _on_term(signum, frame):  # graceful: checkpoint at next step boundary
+        trainer.stop_requested = True
+
+    signal.signal(signal.SIGTERM, _on_term)
     try:
-        result = train_from_config(cfg, run_dir)
+        result = trainer.run()
     except BaseException as e:  # noqa: BLE001
         if rank == 0:
             (run_dir / "status.json").write_text(json.dumps(
-                {"status": "FAILED", "error": repr(e), "ended": time.time()}))
+                {"status": "FAILED", "ended": time.time(),
+                 "failure_analysis": classify_failure(e)}))
         raise
+    finally:
+        trainer.close()
     if rank == 0:
+        status = "STOPPED" if trainer.stop_requested else "COMPLETED"
         (run_dir / "status.json").write_text(json.dumps(
-            {"status": "COMPLETED", "ended": time.time(), "result": result}))
+            {"status": status, "ended": time.time(), "result": result}))
     return 0
 
 

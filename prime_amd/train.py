@@ -37,6 +37,7 @@ def lr_at(step: int, cfg) -> float:
 class Trainer:
     def __init__(self, cfg: TrainConfig, run_dir: str | Path | None = None):
         self.cfg = cfg
+        self.stop_requested = False
         self.run_dir = Path(run_dir) if run_dir else Path("runs") / cfg.run_name
         self.run_dir.mkdir(parents=True, exist_ok=True)
 
@@ -249,6 +250,11 @@ class Trainer:
         last_loss = float("nan")
         loss = None
         for _ in range(cfg.steps):
+            if self.stop_requested:
+                self.log.warning("stop requested: saving checkpoint and exiting")
+                if self.ckpt:
+                    self.save_checkpoint()
+                break
             loss = self.train_step()
             window_steps += 1
             if self.step_count % cfg.metrics.log_interval == 0:
