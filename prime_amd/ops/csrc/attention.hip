@@ -265,8 +265,8 @@ __global__ __launch_bounds__(256) void flash_fwd32_kernel(
   const bf16* Kb = K + b * skb + hkv * skh;
   const bf16* Vtb = Vt + ((int64_t)(b * Hkv + hkv) * D) * S;
 
-  __shared__ bf16 k_lds[64 * D];
-  __shared__ bf16 vt_lds[D * 64];
+  __shared__ bf16 k_lds[2][64 * D];
+  __shared__ bf16 vt_lds[2][D * 64];
   __shared__ float bcast_all[4][32];  // per-wave alpha / inv-l broadcast
   float* bcast = bcast_all[wid];
 
@@ -293,11 +293,18 @@ __global__ __launch_bounds__(256) void flash_fwd32_kernel(
     for (int r = 0; r < 16; ++r) o_acc[dt][r] = 0.f;
 
   const int kv_end = causal ? (qt * 128 + 128) : S;
-  for (int kv = 0; kv < kv_end; kv += 64) {
+  // prefetch-behind-barrier: at 2 waves/SIMD (VGPR-bound) there is no
+  // other-wave overlap to hide a synchronous stage; give the next tile's
+  // loads the whole current tile's compute to land
+  stage_tile<64, D>(Kb, sks, k_lds[0], threadIdx.x);
+  stage_tile<D, 64>(Vtb, S, vt_lds[0], threadIdx.x);
+  int idx = 0;
+  for (int kv = 0; kv < kv_end; kv += 64, idx ^= 1) {
     __syncthreads();
-    stage_tile<64, D>(Kb + (int64_t)kv * sks, sks, k_lds, threadIdx.x);
-    stage_tile<D, 64>(Vtb + kv, S, vt_lds, threadIdx.x);
-    __syncthreads();
+    if (kv + 64 < kv_end) {
+      stage_tile<64, D>(Kb + (int64_t)(kv + 64) * sks, sks, k_lds[idx ^ 1], threadIdx.x);
+      stage_tile<D, 64>(Vtb + kv + 64, S, vt_lds[idx ^ 1], threadIdx.x);
+    }
 
     // ---- St = (K q^T): two 32x32 C tiles over the 64-key block.
     // A-frag rows = k (lane&31), cols = d slice; read from swizzled K LDS.
@@ -308,7 +315,7 @@ __global__ __launch_bounds__(256) void flash_fwd32_kernel(
       for (int r = 0; r < 16; ++r) st[t][r] = 0.f;
 #pragma unroll
       for (int ds = 0; ds < DSL; ++ds) {
-        const short8 kf = ld8_swz<D>(k_lds, t * 32 + lo, ds * 32 + hi * 16);
+        const short8 kf = ld8_swz<D>(k_lds[idx], t * 32 + lo, ds * 32 + hi * 16);
         st[t] = mfma32(kf, qf[ds], st[t]);
       }
     }
@@ -396,7 +403,7 @@ __global__ __launch_bounds__(256) void flash_fwd32_kernel(
 #pragma unroll
       for (int ks = 0; ks < 4; ++ks)
         o_acc[dt] = mfma32(pa[ks],
-                           ld8_swz<64>(vt_lds, dt * 32 + lo, ks * 32 + hi * 16),
+                           ld8_swz<64>(vt_lds[idx], dt * 32 + lo, ks * 32 + hi * 16),
                            o_acc[dt]);
   }
   // ---- epilogue: O /= l (per-row broadcast), write + lse
