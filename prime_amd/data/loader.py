@@ -20,6 +20,7 @@ class DataConfig:
     seq_len: int = 2048
     micro_batch_size: int = 4
     seed: int = 1234
+    shuffle: bool = True
 
 
 class SyntheticTokens:
@@ -50,7 +51,10 @@ class SyntheticTokens:
 
 
 class TokenFileDataset:
-    """Memory-mapped flat token file; strided contiguous windows per shard."""
+    """Memory-mapped flat token file (.bin of uint16/uint32, e.g. a
+    pre-tokenized corpus). Windows of seq_len+1 tokens are sampled via a
+    seeded per-epoch permutation (shuffle=True) or sequentially; sharded
+    by (shard, n_shards); resumable from {epoch, batch_idx}."""
 
     def __init__(self, cfg: DataConfig, vocab_size: int, shard: int, n_shards: int):
         if not cfg.path:
@@ -62,24 +66,42 @@ class TokenFileDataset:
         self.shard = shard
         self.n_shards = n_shards
         self.batch_idx = 0
+        self.epoch = 0
         self.windows = (len(self.tokens) - 1) // cfg.seq_len
+        if self.windows < 1:
+            raise ValueError(f"{p} too small for seq_len={cfg.seq_len}")
+        self._perm: np.ndarray | None = None
+        self._perm_epoch = -1
 
     def state_dict(self) -> dict:
-        return {"batch_idx": self.batch_idx}
+        return {"batch_idx": self.batch_idx, "epoch": self.epoch}
 
     def load_state_dict(self, sd: dict) -> None:
         self.batch_idx = int(sd["batch_idx"])
+        self.epoch = int(sd.get("epoch", 0))
+
+    def _window(self, i: int) -> int:
+        if not self.cfg.shuffle:
+            return i % self.windows
+        epoch = i // self.windows
+        if epoch != self._perm_epoch:
+            rng = np.random.default_rng(self.cfg.seed + epoch)
+            self._perm = rng.permutation(self.windows)
+            self._perm_epoch = epoch
+        return int(self._perm[i % self.windows])
 
     def next_batch(self, device: torch.device) -> tuple[torch.Tensor, torch.Tensor]:
         b, s = self.cfg.micro_batch_size, self.cfg.seq_len
         xs, ys = [], []
         for i in range(b):
-            w = (self.batch_idx * self.n_shards * b + self.shard * b + i) % self.windows
+            gi = self.batch_idx * self.n_shards * b + self.shard * b + i
+            w = self._window(gi)
             start = w * s
             chunk = torch.from_numpy(self.tokens[start : start + s + 1].astype(np.int64))
             xs.append(chunk[:-1])
             ys.append(chunk[1:])
         self.batch_idx += 1
+        self.epoch = (self.batch_idx * self.n_shards * b) // self.windows
         x = torch.stack(xs).to(device, non_blocking=True)
         y = torch.stack(ys).to(device, non_blocking=True)
         return x, y

@@ -138,6 +138,7 @@ class Trainer:
         data_cfg = DataConfig(
             kind=cfg.data.kind, path=cfg.data.path, seq_len=cfg.model.seq_len,
             micro_batch_size=cfg.data.micro_batch_size, seed=cfg.data.seed,
+            shuffle=cfg.data.shuffle,
         )
         self.data = build_dataloader(
             data_cfg, self.model_cfg.vocab_size,

@@ -32,6 +32,7 @@ class DataSection(_Strict):
     micro_batch_size: int = 4
     grad_accum: int = 1
     seed: int = 1234
+    shuffle: bool = True
 
 
 class OptimConfig(_Strict):

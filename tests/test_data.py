@@ -1,0 +1,75 @@
+"""Data pipeline tests: synthetic determinism, token-file sharding/shuffle/
+resume."""
+import numpy as np
+import torch
+
+from prime_amd.data import DataConfig, build_dataloader
+
+
+def _token_file(tmp_path, n=10000, vocab=1000):
+    toks = np.random.default_rng(0).integers(0, vocab, n).astype(np.uint16)
+    p = tmp_path / "toks.bin"
+    toks.tofile(p)
+    return str(p), vocab
+
+
+def test_synthetic_deterministic_and_resumable():
+    cfg = DataConfig(seq_len=32, micro_batch_size=2, seed=5)
+    a = build_dataloader(cfg, 100, 0, 1)
+    b = build_dataloader(cfg, 100, 0, 1)
+    xa, _ = a.next_batch(torch.device("cpu"))
+    xb, _ = b.next_batch(torch.device("cpu"))
+    assert torch.equal(xa, xb)
+    # resume: loader c skips to batch 1 and matches a's second batch
+    c = build_dataloader(cfg, 100, 0, 1)
+    c.load_state_dict(a.state_dict())
+    xa2, _ = a.next_batch(torch.device("cpu"))
+    xc, _ = c.next_batch(torch.device("cpu"))
+    assert torch.equal(xa2, xc)
+
+
+def test_synthetic_shards_disjoint():
+    cfg = DataConfig(seq_len=32, micro_batch_size=2, seed=5)
+    a = build_dataloader(cfg, 100, 0, 2)
+    b = build_dataloader(cfg, 100, 1, 2)
+    xa, _ = a.next_batch(torch.device("cpu"))
+    xb, _ = b.next_batch(torch.device("cpu"))
+    assert not torch.equal(xa, xb)
+
+
+def test_token_file_targets_shifted(tmp_path):
+    path, vocab = _token_file(tmp_path)
+    cfg = DataConfig(kind="token_file", path=path, seq_len=64,
+                     micro_batch_size=2, shuffle=False)
+    dl = build_dataloader(cfg, vocab, 0, 1)
+    x, y = dl.next_batch(torch.device("cpu"))
+    assert torch.equal(x[:, 1:], y[:, :-1])
+
+
+def test_token_file_shuffle_covers_epoch(tmp_path):
+    path, vocab = _token_file(tmp_path, n=64 * 20 + 1)
+    cfg = DataConfig(kind="token_file", path=path, seq_len=64,
+                     micro_batch_size=1, shuffle=True, seed=3)
+    dl = build_dataloader(cfg, vocab, 0, 1)
+    seen = []
+    for _ in range(dl.windows):
+        x, _ = dl.next_batch(torch.device("cpu"))
+        seen.append(int(x[0, 0]))
+    # one epoch visits every window exactly once (permutation)
+    starts = {int(t) for t in np.asarray(dl.tokens[: 64 * 20 : 64])}
+    assert len(seen) == dl.windows
+    assert set(seen) == starts or len(set(seen)) == len(seen)
+
+
+def test_token_file_resume(tmp_path):
+    path, vocab = _token_file(tmp_path)
+    cfg = DataConfig(kind="token_file", path=path, seq_len=64,
+                     micro_batch_size=2, shuffle=True)
+    a = build_dataloader(cfg, vocab, 0, 1)
+    a.next_batch(torch.device("cpu"))
+    st = a.state_dict()
+    xa, _ = a.next_batch(torch.device("cpu"))
+    b = build_dataloader(cfg, vocab, 0, 1)
+    b.load_state_dict(st)
+    xb, _ = b.next_batch(torch.device("cpu"))
+    assert torch.equal(xa, xb)
