@@ -13,6 +13,7 @@ from .functional import (
     dequant_int8,
     flash_attention,
     fused_adamw,
+    grad_sqnorm,
     mfma_probe,
     nesterov_outer,
     pseudograd,
@@ -33,6 +34,7 @@ __all__ = [
     "dequant_int8",
     "flash_attention",
     "fused_adamw",
+    "grad_sqnorm",
     "mfma_probe",
     "nesterov_outer",
     "pseudograd",

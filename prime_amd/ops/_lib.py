@@ -26,7 +26,8 @@ _SIGS = {
     "prime_rope": [ctypes.c_void_p] * 5 + [ctypes.c_int64] * 4 + [ctypes.c_int, ctypes.c_int64],
     "prime_swiglu_fwd": [ctypes.c_void_p] * 3 + [ctypes.c_int64] * 2,
     "prime_swiglu_bwd": [ctypes.c_void_p] * 4 + [ctypes.c_int64] * 2,
-    "prime_adamw": [ctypes.c_void_p] * 6 + [ctypes.c_int64] + [ctypes.c_double] * 5 + [ctypes.c_int64],
+    "prime_adamw": [ctypes.c_void_p] * 7 + [ctypes.c_int64] + [ctypes.c_double] * 5 + [ctypes.c_int64],
+    "prime_grad_sqnorm": [ctypes.c_void_p] * 3 + [ctypes.c_int64],
     "prime_pseudograd": [ctypes.c_void_p] * 4 + [ctypes.c_int64],
     "prime_quant_int8": [ctypes.c_void_p] * 4 + [ctypes.c_int64],
     "prime_dequant_int8": [ctypes.c_void_p] * 4 + [ctypes.c_int64] * 2,

@@ -9,9 +9,14 @@
 
 __global__ void adamw_kernel(float* __restrict__ p32, bf16* __restrict__ p16,
                              const bf16* __restrict__ g, float* __restrict__ m,
-                             float* __restrict__ v, int64_t N, float lr,
-                             float beta1, float beta2, float eps, float wd,
-                             float c1, float c2) {
+                             float* __restrict__ v,
+                             const float* __restrict__ gscale, int64_t N,
+                             float lr, float beta1, float beta2, float eps,
+                             float wd, float c1, float c2) {
+  // optional fused gradient clip: gscale points at a 1-element device
+  // scalar (min(1, max_norm/||g||)) computed by grad_sqnorm + host math —
+  // avoids a separate 21 GB bf16 read+write pass over the flat grad
+  const float gs = gscale ? *gscale : 1.f;
   const int64_t nv = N / 4;
   for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < nv;
        idx += (int64_t)gridDim.x * blockDim.x) {
@@ -22,7 +27,7 @@ __global__ void adamw_kernel(float* __restrict__ p32, bf16* __restrict__ p16,
     bf16x4 pb;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      const float gf = bf2f(gv.v[j]);
+      const float gf = bf2f(gv.v[j]) * gs;
       mv.v[j] = beta1 * mv.v[j] + (1.f - beta1) * gf;
       vv.v[j] = beta2 * vv.v[j] + (1.f - beta2) * gf * gf;
       const float mhat = mv.v[j] * c1;
@@ -39,7 +44,7 @@ __global__ void adamw_kernel(float* __restrict__ p32, bf16* __restrict__ p16,
   const int64_t tail = nv * 4;
   for (int64_t i = tail + blockIdx.x * blockDim.x + threadIdx.x; i < N;
        i += (int64_t)gridDim.x * blockDim.x) {
-    const float gf = bf2f(g[i]);
+    const float gf = bf2f(g[i]) * gs;
     m[i] = beta1 * m[i] + (1.f - beta1) * gf;
     v[i] = beta2 * v[i] + (1.f - beta2) * gf * gf;
     p32[i] -= lr * ((m[i] * c1) / (sqrtf(v[i] * c2) + eps) + wd * p32[i]);
@@ -48,16 +53,48 @@ __global__ void adamw_kernel(float* __restrict__ p32, bf16* __restrict__ p16,
 }
 
 PRIME_API int prime_adamw(hipStream_t stream, void* p32, void* p16,
-                          const void* g, void* m, void* v, int64_t N,
-                          double lr, double beta1, double beta2, double eps,
-                          double wd, int64_t step) {
+                          const void* g, void* m, void* v, const void* gscale,
+                          int64_t N, double lr, double beta1, double beta2,
+                          double eps, double wd, int64_t step) {
   const float b1 = (float)beta1, b2 = (float)beta2;
   const float c1 = 1.f / (1.f - powf(b1, (float)step));
   const float c2 = 1.f / (1.f - powf(b2, (float)step));
   int grid = prime_grid(N / 4 + 1, 256);
   hipLaunchKernelGGL(adamw_kernel, dim3(grid), dim3(256), 0, stream,
                      (float*)p32, (bf16*)p16, (const bf16*)g, (float*)m,
-                     (float*)v, N, (float)lr, b1, b2, (float)eps, (float)wd,
-                     c1, c2);
+                     (float*)v, (const float*)gscale, N, (float)lr, b1, b2,
+                     (float)eps, (float)wd, c1, c2);
+  return (int)hipGetLastError();
+}
+
+// single-pass squared L2 norm of a bf16 buffer (fp32 accumulation)
+__global__ void sqnorm_kernel(const bf16* __restrict__ g, float* __restrict__ out,
+                              int64_t N) {
+  __shared__ float scratch[16];
+  float acc = 0.f;
+  const int64_t nv = N / 8;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const bf16x8 v = reinterpret_cast<const bf16x8*>(g)[i];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float f = bf2f(v.v[j]);
+      acc += f * f;
+    }
+  }
+  for (int64_t i = nv * 8 + blockIdx.x * blockDim.x + threadIdx.x; i < N;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const float f = bf2f(g[i]);
+    acc += f * f;
+  }
+  acc = block_reduce_sum(acc, scratch);
+  if (threadIdx.x == 0) atomicAdd(out, acc);
+}
+
+PRIME_API int prime_grad_sqnorm(hipStream_t stream, const void* g, void* out,
+                                int64_t N) {
+  int grid = prime_grid(N / 8 + 1, 256);
+  hipLaunchKernelGGL(sqnorm_kernel, dim3(grid), dim3(256), 0, stream,
+                     (const bf16*)g, (float*)out, N);
   return (int)hipGetLastError();
 }

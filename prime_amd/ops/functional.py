@@ -299,14 +299,23 @@ def cross_entropy(logits, targets, ignore_index: int = -100) -> torch.Tensor:
 
 
 # ------------------------------------------------- raw (non-autograd) ops
-def fused_adamw(p32, p16, grad, m, v, *, lr, beta1, beta2, eps, wd, step):
+def fused_adamw(p32, p16, grad, m, v, *, lr, beta1, beta2, eps, wd, step,
+                gscale=None):
+    """One-pass AdamW; optional fused gradient clip via a 1-element device
+    scalar `gscale` (grads are multiplied in-register, never rewritten)."""
     check(
         lib().prime_adamw(
             stream_of(p32), ptr(p32), ptr(p16), ptr(grad), ptr(m), ptr(v),
-            p32.numel(), lr, beta1, beta2, eps, wd, step,
+            ptr(gscale), p32.numel(), lr, beta1, beta2, eps, wd, step,
         ),
         "adamw",
     )
+
+
+def grad_sqnorm(grad: torch.Tensor, out: torch.Tensor) -> None:
+    """out (1-elem fp32, pre-zeroed) += sum(grad^2), single bf16 pass."""
+    check(lib().prime_grad_sqnorm(stream_of(grad), ptr(grad), ptr(out), grad.numel()),
+          "grad_sqnorm")
 
 
 def pseudograd(outer32, master32, out_delta):

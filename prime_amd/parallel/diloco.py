@@ -117,10 +117,10 @@ class DilocoOptimizer:
             self._copy_stream = torch.cuda.Stream(device=dev)
 
     # ----------------------------------------------------------------- step
-    def step(self) -> bool:
+    def step(self, gscale=None) -> bool:
         """One inner step (grads already populated + locally averaged).
         Returns True when an outer sync happened."""
-        self.inner.step()
+        self.inner.step(gscale)
         self.inner_step_count += 1
         if self.inner_step_count % self.H != 0:
             return False

@@ -63,6 +63,7 @@ class FlatParamSpace:
             p.data = self.flat_w[o : o + k].view(shp)
             p.grad = self.flat_grad[o : o + k].view(shp)
         self.master32 = self.flat_w.float()
+        self._sq = None
 
     def zero_grad(self) -> None:
         self.flat_grad.zero_()
@@ -78,6 +79,18 @@ class FlatParamSpace:
         scale = (max_norm / (norm + 1e-6)).clamp(max=1.0)
         self.flat_grad.mul_(scale.to(self.flat_grad.dtype))
         return norm
+
+    def grad_clip_scale(self, max_norm: float) -> torch.Tensor:
+        """Deferred clip (GPU): returns the 1-element scale tensor
+        min(1, max_norm/||g||) consumed by the fused AdamW kernel — the
+        flat grad is never rewritten (saves a 2x21 GB pass at 10B)."""
+        from .. import ops
+
+        if self._sq is None:
+            self._sq = torch.zeros(1, device=self.device, dtype=torch.float32)
+        self._sq.zero_()
+        ops.grad_sqnorm(self.flat_grad, self._sq)
+        return (max_norm / (self._sq.sqrt() + 1e-6)).clamp(max=1.0)
 
     def state_dict(self) -> dict:
         return {"flat_w": self.flat_w, "master32": self.master32}
@@ -101,7 +114,7 @@ class FusedAdamW:
         self.v = torch.zeros_like(flat.master32)
         self.step_count = 0
 
-    def step(self) -> None:
+    def step(self, gscale=None) -> None:
         self.step_count += 1
         f = self.flat
         if f.device.type == "cuda":
@@ -109,6 +122,7 @@ class FusedAdamW:
                 f.master32, f.flat_w, f.flat_grad, self.m, self.v,
                 lr=self.lr, beta1=self.betas[0], beta2=self.betas[1],
                 eps=self.eps, wd=self.wd, step=self.step_count,
+                gscale=gscale,
             )
         else:
             ops.reference.adamw_step(

@@ -177,9 +177,14 @@ class Trainer:
             self.flat.finalize_grads()  # sharded units were reduce-scattered in bwd
         else:
             self.mesh.local_allreduce_grad(self.flat.flat_grad)
+        gscale = None
         if cfg.optim.grad_clip > 0:
-            self.flat.clip_grad_norm_(cfg.optim.grad_clip)
-        did_outer = self.diloco.step()
+            if self.device.type == "cuda" and not self.fsdp:
+                # deferred clip: scale folded into the fused AdamW pass
+                gscale = self.flat.grad_clip_scale(cfg.optim.grad_clip)
+            else:
+                self.flat.clip_grad_norm_(cfg.optim.grad_clip)
+        did_outer = self.diloco.step(gscale)
         self.step_count += 1
         if did_outer:
             self._maybe_checkpoint()

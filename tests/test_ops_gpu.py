@@ -293,3 +293,31 @@ def test_mfma32_layout_vs_matmul():
     torch.cuda.synchronize()
     want = _bf(A).float() @ _bf(B).float()
     torch.testing.assert_close(got.cpu(), want.cpu(), atol=1e-3, rtol=1e-3)
+
+
+def test_adamw_fused_clip_matches_explicit():
+    from prime_amd import ops
+
+    N = 8192
+    g = _bf(torch.randn(N) * 3)
+    p32a = torch.randn(N, device=DEV)
+    p32b = p32a.clone()
+    p16a, p16b = p32a.bfloat16(), p32b.bfloat16()
+    ma, va = torch.zeros(N, device=DEV), torch.zeros(N, device=DEV)
+    mb, vb = torch.zeros(N, device=DEV), torch.zeros(N, device=DEV)
+
+    # path a: explicit clip then adamw
+    max_norm = 1.0
+    norm = g.float().norm()
+    ga = (g.float() * (max_norm / (norm + 1e-6)).clamp(max=1.0)).bfloat16()
+    ops.fused_adamw(p32a, p16a, ga, ma, va, lr=1e-2, beta1=0.9, beta2=0.95,
+                    eps=1e-8, wd=0.0, step=1)
+    # path b: fused gscale
+    sq = torch.zeros(1, device=DEV)
+    ops.grad_sqnorm(g, sq)
+    torch.testing.assert_close(sq.sqrt().cpu(), norm.cpu().reshape(1), atol=1e-1, rtol=1e-3)
+    gs = (max_norm / (sq.sqrt() + 1e-6)).clamp(max=1.0)
+    ops.fused_adamw(p32b, p16b, g, mb, vb, lr=1e-2, beta1=0.9, beta2=0.95,
+                    eps=1e-8, wd=0.0, step=1, gscale=gs)
+    # bf16 rounding of the pre-scaled grad differs slightly; compare loosely
+    torch.testing.assert_close(p32a.cpu(), p32b.cpu(), atol=2e-4, rtol=1e-3)
