@@ -251,11 +251,25 @@ class Trainer:
             f"workers={self.mesh.n_workers}x{cfg.parallel.worker_size}, "
             f"H={cfg.diloco.H if cfg.diloco.enabled else 'off'}, device={self.device}"
         )
+        profiler = None
+        if cfg.metrics.torch_profiler_steps > 0 and self.mesh.rank == 0:
+            from torch.profiler import ProfilerActivity, profile, schedule
+
+            profiler = profile(
+                activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
+                schedule=schedule(wait=2, warmup=1,
+                                  active=cfg.metrics.torch_profiler_steps),
+                on_trace_ready=lambda p: p.export_chrome_trace(
+                    str(self.run_dir / "trace.json")),
+            )
+            profiler.start()
         t_start = time.perf_counter()
         window_t0, window_steps = t_start, 0
         last_loss = float("nan")
         loss = None
         for _ in range(cfg.steps):
+            if profiler is not None:
+                profiler.step()
             if self.stop_requested:
                 self.log.warning("stop requested: saving checkpoint and exiting")
                 if self.ckpt:
@@ -287,6 +301,8 @@ class Trainer:
                     f"tok/s={tps:,.0f} mfu={mfu(tps_gpu, self.flops_per_token):.3f}"
                 )
                 window_t0, window_steps = now, 0
+        if profiler is not None:
+            profiler.stop()
         if self.device.type == "cuda":
             torch.cuda.synchronize()
         total_t = time.perf_counter() - t_start

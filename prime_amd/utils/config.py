@@ -76,6 +76,7 @@ class MetricsConfig(_Strict):
     log_interval: int = 10
     jsonl: bool = True
     wandb: bool = False
+    torch_profiler_steps: int = 0   # trace steps [3, 3+N) to <run>/trace.json
 
 
 class TrainConfig(_Strict):

@@ -124,3 +124,16 @@ def test_generate_sampling_reproducible():
     a = generate(m, prompt, 5, temperature=0.8, top_k=10, seed=7)
     b = generate(m, prompt, 5, temperature=0.8, top_k=10, seed=7)
     assert a.tolist() == b.tolist()
+
+
+def test_large_configs_build_on_meta():
+    """Shape-validate the big presets without allocating (meta device)."""
+    from prime_amd.models import build_model
+
+    for name, lo, hi in [("llama_8b", 7.5e9, 8.6e9),
+                         ("intellect_10b", 9.5e9, 11e9),
+                         ("llama_70b", 68e9, 73e9)]:
+        with torch.device("meta"):
+            m = build_model(name)
+        n = sum(p.numel() for p in m.parameters())
+        assert lo < n < hi, (name, n)
