@@ -114,7 +114,6 @@ class DilocoOptimizer:
             self._g_theta = torch.zeros(self.chunk, device=dev, dtype=torch.float32)
             self._g_buf = torch.zeros(self.chunk, device=dev, dtype=torch.float32)
             self._g_delta = torch.zeros(self.chunk, device=dev, dtype=torch.float32)
-            self._copy_stream = torch.cuda.Stream(device=dev)
 
     # ----------------------------------------------------------------- step
     def step(self, gscale=None) -> bool:

@@ -84,8 +84,6 @@ class FlatParamSpace:
         """Deferred clip (GPU): returns the 1-element scale tensor
         min(1, max_norm/||g||) consumed by the fused AdamW kernel — the
         flat grad is never rewritten (saves a 2x21 GB pass at 10B)."""
-        from .. import ops
-
         if self._sq is None:
             self._sq = torch.zeros(1, device=self.device, dtype=torch.float32)
         self._sq.zero_()
