@@ -25,10 +25,28 @@ import torch
 
 
 class _PinnedStager:
-    """Reusable pinned-host mirror of a dict of device tensors."""
+    """Reusable pinned-host mirror of a dict of device tensors. Pinning is
+    best-effort: a 10B checkpoint stages ~200 GB — if the pin fails (or
+    host RAM is tight) the buffer falls back to pageable memory (slower
+    D2H, still async-ish) instead of risking the OOM killer."""
 
     def __init__(self):
         self.buffers: dict[str, torch.Tensor] = {}
+
+    @staticmethod
+    def _alloc(t: torch.Tensor) -> torch.Tensor:
+        try:
+            avail = None
+            for line in open("/proc/meminfo"):
+                if line.startswith("MemAvailable:"):
+                    avail = int(line.split()[1]) * 1024
+                    break
+            need = t.numel() * t.element_size()
+            if avail is None or need < 0.3 * avail:
+                return torch.empty_like(t, device="cpu", pin_memory=True)
+        except (OSError, RuntimeError):
+            pass
+        return torch.empty_like(t, device="cpu")
 
     def stage(self, state: dict[str, torch.Tensor], stream: torch.cuda.Stream) -> dict:
         out = {}
@@ -36,7 +54,7 @@ class _PinnedStager:
             for k, t in state.items():
                 buf = self.buffers.get(k)
                 if buf is None or buf.shape != t.shape or buf.dtype != t.dtype:
-                    buf = torch.empty_like(t, device="cpu", pin_memory=True)
+                    buf = self._alloc(t)
                     self.buffers[k] = buf
                 buf.copy_(t, non_blocking=True)
                 out[k] = buf

@@ -287,6 +287,21 @@ def stop(run: str = typer.Argument(...)):
     typer.secho(f"stop requested for {d.name} (graceful: checkpoints first)", fg="green")
 
 
+@train_app.command("delete")
+def delete(run: str = typer.Argument(...),
+           force: bool = typer.Option(False, "--force", "-f")):
+    """Delete a run's directory (refuses while RUNNING unless --force)."""
+    import shutil
+
+    d = _find_run(run)
+    st = _status(d)
+    if st.get("status") in ("RUNNING", "STARTING") and not force:
+        typer.secho("run appears to be RUNNING; stop it first or use --force", fg="red")
+        raise typer.Exit(1)
+    shutil.rmtree(d)
+    typer.secho(f"deleted {d.name}", fg="green")
+
+
 @train_app.command("restart")
 def restart(run: str = typer.Argument(...), detach: bool = typer.Option(False, "-d")):
     """Restart a run from its latest checkpoint."""

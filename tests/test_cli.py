@@ -165,3 +165,17 @@ def test_graceful_sigterm_stop(tmp_path, runs_dir):
     st = json.loads((run_dir / "status.json").read_text())
     assert st["status"] == "STOPPED", st
     assert (tmp_path / "ck").exists()
+
+
+def test_delete_run(tmp_path, runs_dir):
+    cfg = tmp_path / "d.toml"
+    cfg.write_text('run_name = "del_me"\nsteps = 1\n'
+                   '[model]\nname = "llama_test"\nseq_len = 64\n'
+                   '[data]\nmicro_batch_size = 1\n[metrics]\nlog_interval = 100\n')
+    r = runner.invoke(app, ["train", "run", str(cfg)])
+    assert r.exit_code == 0
+    rows = json.loads(runner.invoke(app, ["train", "list", "--json"]).output)
+    run_id = rows[0]["run"]
+    r = runner.invoke(app, ["train", "delete", run_id])
+    assert r.exit_code == 0
+    assert json.loads(runner.invoke(app, ["train", "list", "--json"]).output) == []
