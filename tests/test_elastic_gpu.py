@@ -26,13 +26,13 @@ def _worker(idx, port, host, q):
 
         cfg = TrainConfig(
             run_name=f"el_gpu_{idx}", steps=4,
-            model=ModelConfig(name="llama_test", seq_len=64),
+            model=ModelConfig(name="llama_150m", seq_len=128),
             diloco=DilocoConfig(H=2),
             parallel=ParallelConfig(elastic=True, heartbeat_interval=0.3,
                                     heartbeat_timeout=15.0),
             metrics=MetricsConfig(log_interval=100),
         )
-        cfg.data.micro_batch_size = 2
+        cfg.data.micro_batch_size = 1
         tr = Trainer(cfg, run_dir=f"/tmp/prime_amd_test/elgpu_{idx}")
         assert tr.device.type == "cuda"
         while len(tr.elastic.members()) < 2:
