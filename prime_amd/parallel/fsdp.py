@@ -34,25 +34,30 @@ from .mesh import ElasticDeviceMesh
 ALIGN = 64
 
 
-def _all_gather_flat(full: torch.Tensor, shard: torch.Tensor, group) -> None:
+def _all_gather_flat(full: torch.Tensor, shard: torch.Tensor, group,
+                     async_op: bool = False):
     if dist.get_backend(group) == "nccl":
-        dist.all_gather_into_tensor(full, shard, group=group)
-    else:
-        W = dist.get_world_size(group)
-        chunks = list(full.chunk(W))
-        dist.all_gather(chunks, shard.contiguous(), group=group)
+        return dist.all_gather_into_tensor(full, shard, group=group,
+                                           async_op=async_op)
+    W = dist.get_world_size(group)
+    chunks = list(full.chunk(W))
+    return dist.all_gather(chunks, shard.contiguous(), group=group,
+                           async_op=async_op)
 
 
 def _reduce_scatter_flat(out_shard: torch.Tensor, grad_full: torch.Tensor,
-                         group) -> None:
+                         group, async_op: bool = False):
     if dist.get_backend(group) == "nccl":
-        dist.reduce_scatter_tensor(out_shard, grad_full, group=group)
-    else:
-        dist.all_reduce(grad_full, group=group)
-        W = dist.get_world_size(group)
-        r = dist.get_rank(group)
-        n = grad_full.numel() // W
-        out_shard.copy_(grad_full[r * n : (r + 1) * n])
+        return dist.reduce_scatter_tensor(out_shard, grad_full, group=group,
+                                          async_op=async_op)
+    # gloo: all_reduce then slice (no reduce_scatter); the slice-copy must
+    # happen after completion, so async falls back to sync here
+    dist.all_reduce(grad_full, group=group)
+    W = dist.get_world_size(group)
+    r = dist.get_rank(group)
+    n = grad_full.numel() // W
+    out_shard.copy_(grad_full[r * n : (r + 1) * n])
+    return None
 
 
 class _Unit:
@@ -158,7 +163,14 @@ class ShardedParamSpace:
         self._grad_pool = [
             torch.zeros(max_unit, device=self.device, dtype=self.dtype) for _ in range(2)
         ]
-        self._rs_tmp = torch.empty(max_unit // self.W, device=self.device, dtype=self.dtype)
+        self._rs_tmp = [
+            torch.empty(max_unit // self.W, device=self.device, dtype=self.dtype)
+            for _ in range(2)
+        ]
+        # comm/compute overlap state
+        self._prefetch: dict[int, object] = {}   # unit idx -> in-flight gather
+        self._last_gathered = -1
+        self._pending_rs: list[tuple[object, _Unit, int]] = []
         self._install_hooks()
 
     # -------------------------------------------------------------- hooks
@@ -201,33 +213,84 @@ class ShardedParamSpace:
 
     # ---------------------------------------------------------- gather/free
     def gather_unit(self, unit: _Unit) -> None:
-        full = self._full_pool[unit.idx % 2][: unit.numel_padded]
-        shard = self.flat_w[unit.shard_off : unit.shard_off + unit.shard_len]
-        _all_gather_flat(full, shard, self.group)
+        """Materialize the unit; overlap: wait on a prefetched gather when
+        one is in flight, and issue the next predicted unit's all-gather
+        (idx+1 while ascending through forward, idx-1 through the backward
+        recompute walk) so it lands under this unit's compute."""
+        work = self._prefetch.pop(unit.idx, None)
+        if work is not None:
+            work.wait()
+        else:
+            self._issue_gather(unit, async_op=False)
+        full = self._full_pool[unit.idx % 2]
         for n, p in unit.params:
             o, k, shp = unit.offsets[n]
             p.data = full[o : o + k].view(shp)
+        # predict + prefetch (buffer-parity safe: pool[i%2] is only reused
+        # two gather steps later, after that unit's backward completed)
+        if unit.idx > self._last_gathered:
+            nxt = unit.idx + 1
+        else:
+            nxt = unit.idx - 1
+        self._last_gathered = unit.idx
+        if 0 <= nxt < len(self.units) and nxt not in self._prefetch:
+            w = self._issue_gather(self.units[nxt], async_op=True)
+            if w is not None:
+                self._prefetch[nxt] = w
+
+    def _issue_gather(self, unit: _Unit, async_op: bool):
+        full = self._full_pool[unit.idx % 2][: unit.numel_padded]
+        shard = self.flat_w[unit.shard_off : unit.shard_off + unit.shard_len]
+        return _all_gather_flat(full, shard, self.group, async_op=async_op)
 
     def release_unit(self, unit: _Unit) -> None:
         for _, p in unit.params:
             p.data = self._stub
 
     def reduce_scatter_unit(self, unit: _Unit) -> None:
-        g = self._grad_pool[unit.idx % 2][: unit.numel_padded]
-        out = self._rs_tmp[: unit.shard_len]
-        _reduce_scatter_flat(out, g, self.group)
+        """Issue the unit's grad reduce-scatter async (RCCL) and complete it
+        when its tmp buffer is next needed (or at finalize_grads) — the
+        collective overlaps the previous unit's recompute+backward."""
+        tmp_idx = unit.idx % 2
+        self._drain_rs(tmp_idx)
+        g = self._grad_pool[tmp_idx][: unit.numel_padded]
+        out = self._rs_tmp[tmp_idx][: unit.shard_len]
+        work = _reduce_scatter_flat(out, g, self.group, async_op=True)
+        if work is None:  # gloo sync fallback already filled `out`
+            self._apply_rs(unit, out)
+        else:
+            self._pending_rs.append((work, unit, tmp_idx))
+
+    def _apply_rs(self, unit: _Unit, out: torch.Tensor) -> None:
         # average over the worker's ranks + accumulate (grad_accum)
         self.flat_grad[unit.shard_off : unit.shard_off + unit.shard_len].add_(
             out, alpha=1.0 / self.W
         )
+
+    def _drain_rs(self, tmp_idx: int | None = None) -> None:
+        keep = []
+        for work, unit, ti in self._pending_rs:
+            if tmp_idx is None or ti == tmp_idx:
+                work.wait()
+                self._apply_rs(unit, self._rs_tmp[ti][: unit.shard_len])
+            else:
+                keep.append((work, unit, ti))
+        self._pending_rs = keep
 
     # -------------------------------------------- FlatParamSpace interface
     def zero_grad(self) -> None:
         self.flat_grad.zero_()
 
     def finalize_grads(self) -> None:
-        """Average the replicated region across the worker (sharded units
-        were reduce-scattered by the backward hooks)."""
+        """Drain in-flight reduce-scatters, then average the replicated
+        region across the worker."""
+        self._drain_rs()
+        # complete (not discard) any stale prefetch: a dangling in-flight
+        # gather landing after the optimizer step would overwrite a fresh
+        # gather with stale weights
+        for w in self._prefetch.values():
+            w.wait()
+        self._prefetch.clear()
         repl = self.flat_grad[self.repl_off :]
         dist.all_reduce(repl, group=self.group)
         repl.div_(self.W)
