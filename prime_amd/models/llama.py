@@ -40,7 +40,7 @@ class Attention(nn.Module):
         self.wqkv = nn.Linear(cfg.dim, (cfg.n_heads + 2 * cfg.n_kv_heads) * hd, bias=False)
         self.wo = nn.Linear(cfg.n_heads * hd, cfg.dim, bias=False)
 
-    def forward(self, x, cos, sin, cache=None, pos: int = 0):
+    def forward(self, x, cos, sin, cache=None, pos: int = 0, pos_dev=None):
         B, S, _ = x.shape
         cfg = self.cfg
         hd = cfg.head_dim
@@ -51,10 +51,17 @@ class Attention(nn.Module):
         q = q.view(B, S, cfg.n_heads, hd)
         k = k.view(B, S, cfg.n_kv_heads, hd)
         v = v.view(B, S, cfg.n_kv_heads, hd)
-        q = ops.apply_rope(q, cos, sin, pos_offset=pos)
-        k = ops.apply_rope(k, cos, sin, pos_offset=pos)
+        p32 = pos_dev["pos32"] if pos_dev is not None else None
+        q = ops.apply_rope(q, cos, sin, pos_offset=pos, pos_dev=p32)
+        k = ops.apply_rope(k, cos, sin, pos_offset=pos, pos_dev=p32)
         if cache is not None:
             kc, vc = cache  # [B, Smax, Hkv, hd]
+            if pos_dev is not None:  # graph-capturable dynamic position
+                kc.index_copy_(1, pos_dev["pos64"], k)
+                vc.index_copy_(1, pos_dev["pos64"], v)
+                o = ops.attn_decode(q, kc, vc, length=1,
+                                    len_dev=pos_dev["len32"])
+                return self.wo(o.reshape(B, 1, cfg.n_heads * hd))
             kc[:, pos : pos + S] = k
             vc[:, pos : pos + S] = v
             if S == 1:  # decode: memory-bound cache-streaming kernel
@@ -90,10 +97,10 @@ class Block(nn.Module):
         self.mlp_norm = RMSNorm(cfg.dim, cfg.norm_eps)
         self.mlp = MLP(cfg)
 
-    def forward(self, x, res, cos, sin, cache=None, pos: int = 0):
+    def forward(self, x, res, cos, sin, cache=None, pos: int = 0, pos_dev=None):
         y, s = ops.fused_add_rmsnorm(x, res, self.attn_norm.weight,
                                      self.attn_norm.eps)
-        a = self.attn(y, cos, sin, cache=cache, pos=pos)
+        a = self.attn(y, cos, sin, cache=cache, pos=pos, pos_dev=pos_dev)
         y2, s2 = ops.fused_add_rmsnorm(a, s, self.mlp_norm.weight,
                                        self.mlp_norm.eps)
         return self.mlp(y2), s2
@@ -136,9 +143,11 @@ class Llama(nn.Module):
         self.rope_cos = cos
         self.rope_sin = sin
 
-    def forward(self, tokens: torch.Tensor, caches=None, pos: int = 0) -> torch.Tensor:
+    def forward(self, tokens: torch.Tensor, caches=None, pos: int = 0,
+                pos_dev=None) -> torch.Tensor:
         """tokens [B,S] -> hidden states [B,S,dim] (pre-lm_head).
-        caches: optional per-layer (k,v) KV caches for inference."""
+        caches: optional per-layer (k,v) KV caches for inference; pos_dev:
+        device-scalar position dict for hipGraph-captured decode."""
         x = self.tok_embeddings(tokens)
         res = None
         cos, sin = self.rope_cos, self.rope_sin
@@ -147,7 +156,8 @@ class Llama(nn.Module):
                 x, res = checkpoint(blk, x, res, cos, sin, use_reentrant=False)
             else:
                 x, res = blk(x, res, cos, sin,
-                             cache=caches[i] if caches else None, pos=pos)
+                             cache=caches[i] if caches else None, pos=pos,
+                             pos_dev=pos_dev)
         y, _ = ops.fused_add_rmsnorm(x, res, self.norm.weight, self.norm.eps)
         return y
 

@@ -23,7 +23,7 @@ _SIGS = {
     "prime_rmsnorm_bwd": [ctypes.c_void_p] * 7 + [ctypes.c_int64] * 2 + [ctypes.c_double],
     "prime_add_rmsnorm_fwd": [ctypes.c_void_p] * 7 + [ctypes.c_int64] * 2 + [ctypes.c_double],
     "prime_add_rmsnorm_bwd": [ctypes.c_void_p] * 8 + [ctypes.c_int64] * 2 + [ctypes.c_double],
-    "prime_rope": [ctypes.c_void_p] * 5 + [ctypes.c_int64] * 4 + [ctypes.c_int, ctypes.c_int64],
+    "prime_rope": [ctypes.c_void_p] * 5 + [ctypes.c_int64] * 4 + [ctypes.c_int, ctypes.c_int64, ctypes.c_void_p],
     "prime_swiglu_fwd": [ctypes.c_void_p] * 3 + [ctypes.c_int64] * 2,
     "prime_swiglu_bwd": [ctypes.c_void_p] * 4 + [ctypes.c_int64] * 2,
     "prime_adamw": [ctypes.c_void_p] * 7 + [ctypes.c_int64] + [ctypes.c_double] * 5 + [ctypes.c_int64],
@@ -40,7 +40,7 @@ _SIGS = {
     "prime_mfma_probe": [ctypes.c_void_p] * 4,
     "prime_mfma_probe32": [ctypes.c_void_p] * 4,
     "prime_transpose_bshd": [ctypes.c_void_p] * 3 + [ctypes.c_int64] * 7,
-    "prime_attn_decode": [ctypes.c_void_p] * 5 + [ctypes.c_int64] * 6 + [ctypes.c_double],
+    "prime_attn_decode": [ctypes.c_void_p] * 5 + [ctypes.c_int64] * 6 + [ctypes.c_double, ctypes.c_void_p],
 }
 
 

@@ -16,7 +16,8 @@ template <int D>
 __global__ __launch_bounds__(128) void attn_decode_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, bf16* __restrict__ O, int B, int H, int Hkv,
-    int Smax, int L, float scale) {
+    int Smax, int L, float scale, const int* __restrict__ len_dev) {
+  if (len_dev) L = *len_dev;  // hipGraph decode: dynamic cache length
   const int bh = blockIdx.x;
   const int b = bh / H, h = bh - b * H;
   const int hkv = h / (H / Hkv);
@@ -103,18 +104,18 @@ PRIME_API int prime_attn_decode(hipStream_t stream, const void* Q,
                                 const void* K, const void* V, void* O,
                                 int64_t B, int64_t H, int64_t Hkv,
                                 int64_t Smax, int64_t L, int64_t D,
-                                double scale) {
+                                double scale, const void* len_dev) {
   if ((D != 64 && D != 128) || L < 1) return hipErrorInvalidValue;
   const int grid = (int)(B * H);
   if (D == 128)
     hipLaunchKernelGGL(attn_decode_kernel<128>, dim3(grid), dim3(128), 0,
                        stream, (const bf16*)Q, (const bf16*)K, (const bf16*)V,
                        (bf16*)O, (int)B, (int)H, (int)Hkv, (int)Smax, (int)L,
-                       (float)scale);
+                       (float)scale, (const int*)len_dev);
   else
     hipLaunchKernelGGL(attn_decode_kernel<64>, dim3(grid), dim3(128), 0,
                        stream, (const bf16*)Q, (const bf16*)K, (const bf16*)V,
                        (bf16*)O, (int)B, (int)H, (int)Hkv, (int)Smax, (int)L,
-                       (float)scale);
+                       (float)scale, (const int*)len_dev);
   return (int)hipGetLastError();
 }

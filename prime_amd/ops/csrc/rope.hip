@@ -12,7 +12,9 @@ __global__ void rope_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
                             const float* __restrict__ cos_t,
                             const float* __restrict__ sin_t, int64_t nrows,
                             int H, int S, int D, float sign,
-                            int64_t pos_offset) {
+                            int64_t pos_offset,
+                            const int* __restrict__ pos_dev) {
+  if (pos_dev) pos_offset = *pos_dev;  // hipGraph decode: dynamic position
   const int half = D / 2;  // multiple of 4 (head_dim 64/128)
   const int hv = half / 4;
   const int64_t total = nrows * hv;
@@ -41,13 +43,13 @@ __global__ void rope_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
 PRIME_API int prime_rope(hipStream_t stream, const void* x, void* y,
                          const void* cos_t, const void* sin_t, int64_t nrows,
                          int64_t H, int64_t S, int64_t D, int backward,
-                         int64_t pos_offset) {
+                         int64_t pos_offset, const void* pos_dev) {
   if (D % 8 != 0) return hipErrorInvalidValue;
   int64_t total = nrows * (D / 8);
   int grid = prime_grid(total, 256);
   hipLaunchKernelGGL(rope_kernel, dim3(grid), dim3(256), 0, stream,
                      (const bf16*)x, (bf16*)y, (const float*)cos_t,
                      (const float*)sin_t, nrows, (int)H, (int)S, (int)D,
-                     backward ? -1.f : 1.f, pos_offset);
+                     backward ? -1.f : 1.f, pos_offset, (const int*)pos_dev);
   return (int)hipGetLastError();
 }

@@ -72,7 +72,7 @@ def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tensor
 # ------------------------------------------------------------------ RoPE
 class _Rope(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, cos, sin, pos_offset: int):
+    def forward(ctx, x, cos, sin, pos_offset: int, pos_dev=None):
         if not _is_hip(x):
             ctx.cpu = (cos, sin, pos_offset)
             return ref.apply_rope(x, cos, sin, pos_offset)
@@ -82,7 +82,8 @@ class _Rope(torch.autograd.Function):
         y = torch.empty_like(x)
         check(
             lib().prime_rope(
-                stream_of(x), ptr(x), ptr(y), ptr(cos), ptr(sin), B * S * H, H, S, D, 0, pos_offset
+                stream_of(x), ptr(x), ptr(y), ptr(cos), ptr(sin), B * S * H,
+                H, S, D, 0, pos_offset, ptr(pos_dev),
             ),
             "rope_fwd",
         )
@@ -102,16 +103,19 @@ class _Rope(torch.autograd.Function):
         dx = torch.empty_like(dy)
         check(
             lib().prime_rope(
-                stream_of(dy), ptr(dy), ptr(dx), ptr(cos), ptr(sin), B * S * H, H, S, D, 1, off
+                stream_of(dy), ptr(dy), ptr(dx), ptr(cos), ptr(sin), B * S * H,
+                H, S, D, 1, off, ptr(None),
             ),
             "rope_bwd",
         )
-        return dx, None, None, None
+        return dx, None, None, None, None
 
 
-def apply_rope(x, cos, sin, pos_offset: int = 0):
-    """x: [B,S,H,D] bf16; cos/sin: [S_max, D/2] fp32 tables."""
-    return _Rope.apply(x, cos, sin, pos_offset)
+def apply_rope(x, cos, sin, pos_offset: int = 0, pos_dev=None):
+    """x: [B,S,H,D] bf16; cos/sin: [S_max, D/2] fp32 tables. `pos_dev`
+    (optional int32 device scalar) overrides pos_offset at kernel time
+    (hipGraph decode)."""
+    return _Rope.apply(x, cos, sin, pos_offset, pos_dev)
 
 
 # ---------------------------------------------------------------- SwiGLU
@@ -438,9 +442,11 @@ def fused_add_rmsnorm(x, res, w, eps: float = 1e-5):
 
 
 # ------------------------------------------------------- decode attention
-def attn_decode(q, k_cache, v_cache, length: int) -> torch.Tensor:
+def attn_decode(q, k_cache, v_cache, length: int, len_dev=None) -> torch.Tensor:
     """Single-token decode: q [B,H,D] (or [B,1,H,D]) against the first
-    `length` rows of k/v caches [B,Smax,Hkv,D]. Returns [B,H,D]."""
+    `length` rows of k/v caches [B,Smax,Hkv,D]. Returns [B,H,D]. `len_dev`
+    (optional int32 device scalar) overrides `length` at kernel time
+    (hipGraph decode)."""
     if q.dim() == 4:
         q = q.squeeze(1)
     B, H, D = q.shape
@@ -454,7 +460,7 @@ def attn_decode(q, k_cache, v_cache, length: int) -> torch.Tensor:
     check(
         lib().prime_attn_decode(
             stream_of(q), ptr(q), ptr(k_cache), ptr(v_cache), ptr(o),
-            B, H, Hkv, Smax, length, D, D**-0.5,
+            B, H, Hkv, Smax, length, D, D**-0.5, ptr(len_dev),
         ),
         "attn_decode",
     )

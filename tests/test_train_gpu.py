@@ -92,3 +92,18 @@ def test_generate_on_gpu(tmp_path):
     out = generate(m, prompt, max_new_tokens=8, temperature=0.0)
     assert out.shape == (2, 41)
     assert (out[:, :33] == prompt).all()
+
+
+def test_generate_graphed_matches_eager(tmp_path):
+    import torch
+
+    from prime_amd.models import build_model
+    from prime_amd.models.generate import generate
+
+    torch.manual_seed(0)
+    m = build_model("llama_150m").to("cuda", dtype=torch.bfloat16)
+    m.reset_rope("cuda")
+    prompt = torch.randint(0, m.cfg.vocab_size, (2, 17), device="cuda")
+    eager = generate(m, prompt, max_new_tokens=12, temperature=0.0, use_graph=False)
+    graphed = generate(m, prompt, max_new_tokens=12, temperature=0.0, use_graph=True)
+    assert eager.tolist() == graphed.tolist()

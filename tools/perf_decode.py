@@ -21,15 +21,17 @@ def main():
     m = m.to(dtype=torch.bfloat16)
     m.reset_rope("cuda")
     prompt = torch.randint(0, m.cfg.vocab_size, (bs, 128), device="cuda")
-    generate(m, prompt, 8)  # warmup
-    torch.cuda.synchronize()
-    t0 = time.perf_counter()
-    out = generate(m, prompt, new)
-    torch.cuda.synchronize()
-    dt = time.perf_counter() - t0
-    tps = bs * new / dt
-    print(f"{model_name} bs={bs}: {tps:,.0f} decode tok/s "
-          f"({1e3*dt/new:.2f} ms/step, {out.shape[1]} total len)")
+    for graphed in (False, True):
+        generate(m, prompt, 8, use_graph=graphed)  # warmup
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        out = generate(m, prompt, new, use_graph=graphed)
+        torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        tps = bs * new / dt
+        mode = "graph" if graphed else "eager"
+        print(f"{model_name} bs={bs} {mode}: {tps:,.0f} decode tok/s "
+              f"({1e3*dt/new:.2f} ms/tok, {out.shape[1]} total len)")
 
 
 if __name__ == "__main__":
