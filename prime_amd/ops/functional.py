@@ -96,7 +96,7 @@ class _Rope(torch.autograd.Function):
         if ctx.cpu is not None:
             cos, sin, off = ctx.cpu
             # inverse rotation
-            return ref.apply_rope(dy, cos, -sin, off), None, None, None
+            return ref.apply_rope(dy, cos, -sin, off), None, None, None, None
         cos, sin = ctx.saved_tensors
         B, S, H, D, off = ctx.meta
         dy = dy.contiguous()
