@@ -28,7 +28,9 @@ def generate(
     use_graph: bool = True,
 ) -> torch.Tensor:
     """tokens [B, L0] -> [B, L0 + max_new_tokens] (greedy when
-    temperature == 0)."""
+    temperature == 0). With use_graph on CUDA, the per-token decode pass
+    runs as a cached hipGraph replay (capture amortizes across calls of
+    the same (batch, length-bucket))."""
     model.eval()
     cfg = model.cfg
     dev = tokens.device
@@ -37,14 +39,30 @@ def generate(
     total = L0 + max_new_tokens
     if total > cfg.max_seq:
         raise ValueError(f"{total} tokens exceeds max_seq {cfg.max_seq}")
-    smax = _pad64(total)
-    caches = [
-        (
-            torch.zeros(B, smax, cfg.n_kv_heads, cfg.head_dim, device=dev, dtype=dtype),
-            torch.zeros(B, smax, cfg.n_kv_heads, cfg.head_dim, device=dev, dtype=dtype),
-        )
-        for _ in range(cfg.n_layers)
-    ]
+    graphing = use_graph and tokens.is_cuda and max_new_tokens > 2
+    # bucket the cache length so repeat calls reuse the captured graph
+    smax = min(_pad64(cfg.max_seq), _pad64(max(total, 1024))) if graphing \
+        else _pad64(total)
+    sessions = getattr(model, "_decode_sessions", None)
+    if sessions is None:
+        sessions = model._decode_sessions = {}
+    key = (B, smax, dev.index)
+    sess = None
+    if graphing and key in sessions:
+        sess = sessions[key]
+        caches = sess["caches"]
+    else:
+        caches = [
+            (
+                torch.zeros(B, smax, cfg.n_kv_heads, cfg.head_dim, device=dev, dtype=dtype),
+                torch.zeros(B, smax, cfg.n_kv_heads, cfg.head_dim, device=dev, dtype=dtype),
+            )
+            for _ in range(cfg.n_layers)
+        ]
+        if graphing:
+            # capture BEFORE prefill: the warmup/capture passes write junk
+            # rows into the cache, which the prefill then overwrites
+            sess = sessions[key] = _build_session(model, caches, B, dev)
     gen = torch.Generator(device="cpu")
     if seed is not None:
         gen.manual_seed(seed)
@@ -60,8 +78,8 @@ def generate(
     cur = _sample(logits, temperature, top_k, gen)
     out.append(cur)
     pos = L0
-    if use_graph and tokens.is_cuda and max_new_tokens > 8:
-        rest = _decode_graphed(model, caches, cur, pos, max_new_tokens - 1,
+    if sess is not None and max_new_tokens > 1:
+        rest = _decode_graphed(sess, cur, pos, max_new_tokens - 1,
                                temperature, top_k, gen)
         out.extend(rest)
     else:
@@ -74,21 +92,18 @@ def generate(
     return torch.cat(out, dim=1)
 
 
-def _decode_graphed(model, caches, first_tok, pos0, n_tokens, temperature,
-                    top_k, gen):
-    """hipGraph-captured decode: the whole per-token pass (embed -> 42x
-    {norms, qkv GEMM, rope, cache write, decode attention, o/mlp GEMMs} ->
-    lm_head) replays as ONE graph launch — the eager loop is launch-bound
-    (~6 kernels x n_layers x ~20 us per token). Dynamic state (position,
-    cache length) lives in device scalars the kernels read at run time and
-    the graph itself increments."""
-    dev = first_tok.device
+def _build_session(model, caches, B, dev):
+    """Capture the per-token decode pass (embed -> n_layers x {norms, qkv
+    GEMM, rope, cache write, decode attention, o/mlp GEMMs} -> lm_head) as
+    ONE replayable hipGraph. Dynamic state (position, cache length) lives
+    in device scalars the kernels read at run time; the graph itself
+    increments them. Capture cost amortizes across generate() calls."""
     pos_dev = {
-        "pos32": torch.tensor(pos0, dtype=torch.int32, device=dev),
-        "pos64": torch.tensor([pos0], dtype=torch.int64, device=dev),
-        "len32": torch.tensor(pos0 + 1, dtype=torch.int32, device=dev),
+        "pos32": torch.zeros((), dtype=torch.int32, device=dev),
+        "pos64": torch.zeros(1, dtype=torch.int64, device=dev),
+        "len32": torch.ones((), dtype=torch.int32, device=dev),
     }
-    static_tok = first_tok.clone()
+    static_tok = torch.zeros(B, 1, dtype=torch.int64, device=dev)
 
     def step():
         h = model(static_tok, caches=caches, pos=0, pos_dev=pos_dev)
@@ -98,24 +113,27 @@ def _decode_graphed(model, caches, first_tok, pos0, n_tokens, temperature,
         pos_dev["len32"].add_(1)
         return logits
 
-    # warmup on a side stream (per torch CUDA-graphs contract), then reset
+    # warmup on a side stream (per torch CUDA-graphs contract)
     s = torch.cuda.Stream()
     s.wait_stream(torch.cuda.current_stream())
     with torch.cuda.stream(s):
         for _ in range(2):
             step()
-            pos_dev["pos32"].fill_(pos0)
-            pos_dev["pos64"].fill_(pos0)
-            pos_dev["len32"].fill_(pos0 + 1)
     torch.cuda.current_stream().wait_stream(s)
-
     graph = torch.cuda.CUDAGraph()
     with torch.cuda.graph(graph):
         static_logits = step()
+    return {"caches": caches, "graph": graph, "pos_dev": pos_dev,
+            "static_tok": static_tok, "static_logits": static_logits}
+
+
+def _decode_graphed(sess, first_tok, pos0, n_tokens, temperature, top_k, gen):
+    pos_dev = sess["pos_dev"]
     pos_dev["pos32"].fill_(pos0)
     pos_dev["pos64"].fill_(pos0)
     pos_dev["len32"].fill_(pos0 + 1)
-
+    sess["static_tok"].copy_(first_tok)
+    graph, static_logits = sess["graph"], sess["static_logits"]
     out = []
     for _ in range(n_tokens):
         graph.replay()
@@ -124,7 +142,7 @@ def _decode_graphed(model, caches, first_tok, pos0, n_tokens, temperature,
         else:
             nxt = _sample(static_logits, temperature, top_k, gen)
         out.append(nxt)
-        static_tok.copy_(nxt)
+        sess["static_tok"].copy_(nxt)
     return out
 
 
