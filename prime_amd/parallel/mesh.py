@@ -103,9 +103,16 @@ class ElasticDeviceMesh:
         if not self.cfg.quant_outer:
             ring.allreduce_fp32(delta32, group=group, average=True)
             return
+        import os
+
         W = self.n_workers
         R = len([o for o in range(1, W) if ring._gcd(o, W) == 1])
-        if W > 2 and R > 1 and delta32.numel() % (R * W * ring.QBLK) == 0:
+        # multi-ring engages R xGMI links concurrently but is opt-in for
+        # now (PRIME_AMD_MULTIRING=1): the outer sync is amortized over H
+        # steps, and an unattended scale run should not bet on concurrent
+        # grouped P2P until it has been validated on an 8-GPU node
+        multi = os.environ.get("PRIME_AMD_MULTIRING", "0") == "1"
+        if multi and W > 2 and R > 1 and delta32.numel() % (R * W * ring.QBLK) == 0:
             ring.ring_allreduce_int8_multi(delta32, group=group, average=True)
         else:
             ring.ring_allreduce_int8(delta32, group=group, average=True)
