@@ -159,3 +159,28 @@ class CheckpointManager:
                 return None
             fname = cands[0]
         return torch.load(fname, map_location=map_location, weights_only=False)
+
+
+def export_safetensors(ckpt_dir, model_name: str, out_path,
+                       overrides: dict | None = None) -> int:
+    """Explode a checkpoint's flat fp32 master into per-parameter bf16
+    tensors and write a safetensors file (interop export). Returns the
+    number of tensors written."""
+    from safetensors.torch import save_file
+
+    from ..models import build_model
+    from ..parallel.flat import FlatParamSpace
+
+    mgr = CheckpointManager(ckpt_dir)
+    payload = mgr.load()
+    if payload is None:
+        raise FileNotFoundError(f"no checkpoint under {ckpt_dir}")
+    model = build_model(model_name, **(overrides or {}))
+    flat = FlatParamSpace(model)
+    flat.load_flat_(payload["tensors"]["master32"])
+    tensors = {}
+    for n, _ in flat.params:
+        o, k, shp = flat.offsets[n]
+        tensors[n] = flat.flat_w[o : o + k].view(shp).to(torch.bfloat16).contiguous()
+    save_file(tensors, str(out_path), metadata={"model": model_name})
+    return len(tensors)

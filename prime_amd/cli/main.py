@@ -440,6 +440,8 @@ def eval_cmd(
 def generate_cmd(
     model: str = typer.Option("llama_150m", help="model preset"),
     prompt_tokens: str = typer.Option("1,2,3,4", help="comma-separated token ids"),
+    prompt: Optional[str] = typer.Option(None, help="text prompt (needs --tokenizer)"),
+    tokenizer: Optional[str] = typer.Option(None, help="local tokenizer.json"),
     max_new: int = typer.Option(32),
     temperature: float = typer.Option(0.0),
     top_k: int = typer.Option(0),
@@ -452,6 +454,11 @@ def generate_cmd(
     from ..models import build_model
     from ..models.generate import generate
 
+    tok = None
+    if tokenizer:
+        from ..utils.tokenizer import load_tokenizer
+
+        tok = load_tokenizer(tokenizer)
     torch.manual_seed(seed)
     dev = "cuda" if torch.cuda.is_available() else "cpu"
     m = build_model(model)
@@ -468,10 +475,42 @@ def generate_cmd(
             typer.secho("no checkpoint found", fg="red")
             raise typer.Exit(1)
         flat.load_flat_(payload["tensors"]["master32"].to(dev))
-    toks = torch.tensor([[int(t) for t in prompt_tokens.split(",")]], device=dev)
+    if prompt is not None:
+        if tok is None:
+            typer.secho("--prompt requires --tokenizer", fg="red")
+            raise typer.Exit(2)
+        from ..utils.tokenizer import encode
+
+        ids = encode(tok, prompt)
+    else:
+        ids = [int(t) for t in prompt_tokens.split(",")]
+    toks = torch.tensor([ids], device=dev)
     out = generate(m, toks, max_new, temperature=temperature, top_k=top_k,
                    seed=seed)
-    typer.echo(",".join(str(int(t)) for t in out[0]))
+    if tok is not None:
+        from ..utils.tokenizer import decode as tok_decode
+
+        typer.echo(tok_decode(tok, [int(t) for t in out[0]]))
+    else:
+        typer.echo(",".join(str(int(t)) for t in out[0]))
+
+
+# ------------------------------------------------------------------ export
+@app.command("export")
+def export_cmd(
+    checkpoint: str = typer.Argument(..., help="checkpoint dir"),
+    out: str = typer.Argument("model.safetensors"),
+    model: str = typer.Option("llama_150m"),
+):
+    """Export a checkpoint's weights to safetensors (per-parameter bf16)."""
+    from ..ckpt.manager import export_safetensors
+
+    try:
+        n = export_safetensors(checkpoint, model, out)
+    except FileNotFoundError as e:
+        typer.secho(str(e), fg="red")
+        raise typer.Exit(1)
+    typer.secho(f"wrote {n} tensors to {out}", fg="green")
 
 
 # ------------------------------------------------------------------ store

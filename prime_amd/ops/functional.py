@@ -182,6 +182,15 @@ def _bshd_ok(t: torch.Tensor) -> bool:
         and t.storage_offset() % 8 == 0
 
 
+def _check_flash_shape(S: int, D: int) -> None:
+    if D not in (64, 128) or S % 64 != 0:
+        raise NotImplementedError(
+            f"flash attention kernels support head_dim in {{64, 128}} and "
+            f"seq_len % 64 == 0 (got head_dim={D}, seq_len={S}); pad the "
+            "sequence or pick a head_dim-compatible model config"
+        )
+
+
 class _FlashAttention(torch.autograd.Function):
     """Stride-aware [B,S,H,D] domain: consumes q/k/v views of the packed
     qkv GEMM output directly (no transposes / .contiguous() on the hot
@@ -191,6 +200,7 @@ class _FlashAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal: bool, scale: float):
         B, S, H, D = q.shape
+        _check_flash_shape(S, D)
         Hkv = k.shape[2]
         if not (_bshd_ok(q) and _bshd_ok(k) and _bshd_ok(v)):
             q, k, v = q.contiguous(), k.contiguous(), v.contiguous()

@@ -57,3 +57,19 @@ def test_trainer_resume(tmp_path):
     res = tr2.run()  # continues training
     assert res["steps"] == inner_end + 2
     tr2.close()
+
+
+def test_export_safetensors(tmp_path):
+    from safetensors.torch import load_file
+
+    tr = Trainer(_train_cfg(2, tmp_path), run_dir=tmp_path / "runE")
+    tr.run()
+    tr.close()
+    from prime_amd.ckpt.manager import export_safetensors
+
+    out = tmp_path / "m.safetensors"
+    n = export_safetensors(tmp_path / "ck", "llama_test", out)
+    assert n > 0 and out.exists()
+    tensors = load_file(str(out))
+    assert "tok_embeddings.weight" in tensors
+    assert tensors["layers.0.attn.wqkv.weight"].shape[1] == 64
