@@ -129,7 +129,20 @@ class DilocoOptimizer:
     def outer_step(self) -> None:
         if self.elastic is not None:
             # boundary: agree on membership + rebuild the cross-worker group
-            self._view = self.elastic.sync()
+            from .elastic import EvictedError
+
+            try:
+                self._view = self.elastic.sync()
+            except EvictedError:
+                # stalled past the heartbeat timeout and got evicted:
+                # re-register, adopt a live peer's outer state, skip this
+                # boundary, and contribute again from the next one
+                self.elastic.rejoin()
+                payload = self.elastic.bootstrap_from_peer()
+                if payload is not None:
+                    self.load_bootstrap(payload)
+                self.rejoined = getattr(self, "rejoined", 0) + 1
+                return
         with self._outer_lock:
             if self.outer_device == "gpu":
                 self._outer_step_resident()

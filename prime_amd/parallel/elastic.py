@@ -267,6 +267,21 @@ class ElasticWorker:
                 continue
         return None
 
+    # -------------------------------------------------------------- rejoin
+    def rejoin(self) -> None:
+        """Re-register after an eviction (e.g. a long stall): new worker id,
+        fresh heartbeat identity, re-advertised checkpoint server. The
+        caller should re-bootstrap state from a peer before contributing."""
+        seq = int(self.store.add("join_seq", 1))
+        old = self.wid
+        self.wid = f"{seq:06d}-rejoin-{old.split('-', 1)[1]}"
+        self._peer_tracks.clear()
+        self.epoch = int(self._get_str("epoch", "0"))
+        self._register()
+        if self._ckpt_srv is not None:
+            host, port = self._ckpt_srv.getsockname()
+            self.store.set(f"ckptsrv/{self.wid}", f"{host}:{port}")
+
     # --------------------------------------------------------------- leave
     def close(self, leaving: bool = True) -> None:
         if leaving:

@@ -214,3 +214,88 @@ def test_elastic_trainer_two_workers():
     assert out[0]["outer_steps"] == 2
     assert out[0]["head"] == out[1]["head"]
     assert out[0]["outer"] == out[1]["outer"]
+
+
+def _stall_worker(port):
+    """Worker that stalls past the heartbeat timeout mid-run, gets evicted,
+    and must auto-rejoin at its next boundary."""
+    import torch
+
+    from prime_amd.parallel.elastic import ElasticWorker
+    from prime_amd.parallel.diloco import DilocoOptimizer
+    from prime_amd.parallel.flat import FlatParamSpace, FusedAdamW
+    from prime_amd.parallel.mesh import ElasticDeviceMesh, MeshConfig
+    from prime_amd.models import build_model
+
+    torch.manual_seed(0)
+    mesh = ElasticDeviceMesh(MeshConfig())
+    model = build_model("llama_test")
+    flat = FlatParamSpace(model)
+    el = ElasticWorker(port=port, host_store=False, worker_name="staller",
+                       heartbeat_interval=0.2, heartbeat_timeout=1.5,
+                       ckpt_provider=lambda: dl.live_state())
+    dl = DilocoOptimizer(flat, mesh, FusedAdamW(flat), H=1, elastic=el)
+    while len(el.members()) < 2:
+        time.sleep(0.05)
+    dl.outer_step()           # boundary 1: both present
+    # stall: stop heartbeating long enough to be evicted
+    el._stop.set()
+    el._hb_thread.join()
+    el.store.set("test/stalling", "1")
+    el.store.wait(["test/evicted_me"])
+    el._stop.clear()
+    import threading
+
+    el._hb_thread = threading.Thread(target=el._hb_loop, daemon=True)
+    el._hb_thread.start()
+    dl.outer_step()           # detects eviction -> rejoin + bootstrap
+    assert getattr(dl, "rejoined", 0) == 1
+    el.store.set("test/rejoined", "1")
+    dl.outer_step()           # participates again (world 2)
+    el.store.set("test/staller_done", "1")
+    el.close()
+    return {"rejoined": dl.rejoined, "outer": dl.outer_step_count}
+
+
+def _survivor2_worker(port):
+    import torch
+
+    from prime_amd.parallel.elastic import ElasticWorker
+    from prime_amd.parallel.diloco import DilocoOptimizer
+    from prime_amd.parallel.flat import FlatParamSpace, FusedAdamW
+    from prime_amd.parallel.mesh import ElasticDeviceMesh, MeshConfig
+    from prime_amd.models import build_model
+
+    torch.manual_seed(0)
+    mesh = ElasticDeviceMesh(MeshConfig())
+    model = build_model("llama_test")
+    flat = FlatParamSpace(model)
+    el = ElasticWorker(port=port, host_store=True, worker_name="surv2",
+                       heartbeat_interval=0.2, heartbeat_timeout=1.5,
+                       ckpt_provider=lambda: dl.live_state())
+    dl = DilocoOptimizer(flat, mesh, FusedAdamW(flat), H=1, elastic=el)
+    while len(el.members()) < 2:
+        time.sleep(0.05)
+    dl.outer_step()           # boundary 1
+    el.store.wait(["test/stalling"])
+    time.sleep(2.0)           # staller's heartbeat goes stale
+    dl.outer_step()           # evicts the staller (world 1)
+    el.store.set("test/evicted_me", "1")
+    worlds = [dl._view.world]
+    el.store.wait(["test/rejoined"])  # deterministic: staller re-registered
+    dl.outer_step()           # staller rejoined -> world 2 again
+    worlds.append(dl._view.world)
+    el.store.wait(["test/staller_done"])
+    el.close()
+    return {"worlds": worlds}
+
+
+def test_evicted_worker_auto_rejoins():
+    port = free_port()
+    out = _spawn([
+        (_survivor2_worker, {}, (port,)),
+        (_stall_worker, {}, (port,)),
+    ], timeout=240)
+    assert out[1]["rejoined"] == 1
+    assert out[0]["worlds"][0] == 1   # after eviction
+    assert out[0]["worlds"][1] == 2   # after rejoin
