@@ -391,6 +391,31 @@ def config_delete(name: str):
     typer.secho(f"deleted context '{name}'", fg="green")
 
 
+# ----------------------------------------------------------- prepare-data
+@app.command("prepare-data")
+def prepare_data(
+    text_file: str = typer.Argument(..., help="input UTF-8 text file"),
+    out: str = typer.Argument(..., help="output token .bin (uint16/uint32)"),
+    tokenizer: str = typer.Option(..., help="local tokenizer.json"),
+    vocab_threshold: int = typer.Option(65535, help="uint16 if vocab fits"),
+):
+    """Tokenize a text file into the flat token .bin the token_file
+    dataloader consumes."""
+    import numpy as np
+
+    from ..utils.tokenizer import encode, load_tokenizer
+
+    tok = load_tokenizer(tokenizer)
+    text = Path(text_file).read_text()
+    ids = encode(tok, text)
+    vocab = tok.get_vocab_size()
+    dtype = np.uint16 if vocab <= vocab_threshold else np.uint32
+    arr = np.asarray(ids, dtype=dtype)
+    arr.tofile(out)
+    typer.secho(f"wrote {len(arr):,} tokens (vocab {vocab}, {dtype.__name__}) to {out}",
+                fg="green")
+
+
 # ------------------------------------------------------------------- eval
 @app.command("eval")
 def eval_cmd(

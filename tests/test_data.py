@@ -73,3 +73,35 @@ def test_token_file_resume(tmp_path):
     b.load_state_dict(st)
     xb, _ = b.next_batch(torch.device("cpu"))
     assert torch.equal(xa, xb)
+
+
+def test_prepare_data_roundtrip(tmp_path):
+    """Build a tiny tokenizer locally (no network), prepare a .bin, train a
+    step from it — the full real-data path."""
+    tokenizers = __import__("tokenizers")
+    from tokenizers.models import WordLevel
+    from tokenizers.pre_tokenizers import Whitespace
+    from tokenizers.trainers import WordLevelTrainer
+
+    text = ("the quick brown fox jumps over the lazy dog " * 200).strip()
+    src = tmp_path / "corpus.txt"
+    src.write_text(text)
+    tok = tokenizers.Tokenizer(WordLevel(unk_token="<unk>"))
+    tok.pre_tokenizer = Whitespace()
+    tok.train_from_iterator([text], WordLevelTrainer(special_tokens=["<unk>"]))
+    tok_path = tmp_path / "tok.json"
+    tok.save(str(tok_path))
+
+    from typer.testing import CliRunner
+
+    from prime_amd.cli.main import app
+
+    out_bin = tmp_path / "toks.bin"
+    r = CliRunner().invoke(app, ["prepare-data", str(src), str(out_bin),
+                                 "--tokenizer", str(tok_path)])
+    assert r.exit_code == 0, r.output
+    cfg = DataConfig(kind="token_file", path=str(out_bin), seq_len=64,
+                     micro_batch_size=2, shuffle=True)
+    dl = build_dataloader(cfg, 1000, 0, 1)
+    x, y = dl.next_batch(torch.device("cpu"))
+    assert x.shape == (2, 64) and int(x.max()) < tok.get_vocab_size()
