@@ -53,3 +53,39 @@ def test_diloco_single_worker_no_comm():
     # world=1: outer step still applies (self-average is identity)
     outs = run_distributed(_diloco_worker, 1, args=(2, 2, True), timeout=300)
     assert outs[0]["outer_steps"] == 1
+
+
+def _mixed_worker(rank, world):
+    """2 workers x 2 DP ranks: inner grad all-reduce + outer int8 ring."""
+    from prime_amd.utils.config import (
+        DilocoConfig, MetricsConfig, ModelConfig, ParallelConfig, TrainConfig,
+    )
+    from prime_amd.train import Trainer
+
+    cfg = TrainConfig(
+        run_name="mixed",
+        steps=4,
+        model=ModelConfig(name="llama_test", seq_len=64),
+        diloco=DilocoConfig(H=2),
+        parallel=ParallelConfig(worker_size=2),
+        metrics=MetricsConfig(log_interval=100),
+    )
+    cfg.data.micro_batch_size = 2
+    tr = Trainer(cfg, run_dir=f"/tmp/prime_amd_test/mixed_r{rank}")
+    res = tr.run()
+    out = {
+        "outer": res["outer_steps"],
+        "head": tr.flat.flat_w[:16].tolist(),
+        "worker": tr.mesh.worker_id,
+        "n_workers": tr.mesh.n_workers,
+    }
+    tr.close()
+    return out
+
+
+def test_mixed_topology_2x2():
+    outs = run_distributed(_mixed_worker, 4, timeout=300)
+    assert outs[0]["n_workers"] == 2
+    assert outs[0]["outer"] == 2
+    # after the final outer boundary every rank holds identical params
+    assert outs[0]["head"] == outs[1]["head"] == outs[2]["head"] == outs[3]["head"]
