@@ -252,7 +252,13 @@ class _FlashAttention(torch.autograd.Function):
         # split the causal q loop across the grid: the unsplit grid is only
         # B*Hkv*S/64 blocks with 1..S/64 trip-count imbalance
         base_blocks = B * Hkv * (S // 64)
-        splits = max(1, min(8, 8192 // max(1, base_blocks), S // 64))
+        import os
+
+        env = os.environ.get("PRIME_AMD_DKV_SPLITS")
+        if env:
+            splits = max(1, min(int(env), S // 64))
+        else:
+            splits = max(1, min(8, 8192 // max(1, base_blocks), S // 64))
         ws = torch.empty(2, splits, B, Hkv, S, D, device=q.device,
                          dtype=torch.float32)
         check(
