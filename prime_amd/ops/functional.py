@@ -258,7 +258,9 @@ class _FlashAttention(torch.autograd.Function):
         if env:
             splits = max(1, min(int(env), S // 64))
         else:
-            splits = max(1, min(8, 8192 // max(1, base_blocks), S // 64))
+            # A/B on MI355X @10B shapes: splits=2 (167 TF) > 4 (161) > 8 (153)
+            # — workspace traffic outweighs balance beyond ~4096 blocks
+            splits = max(1, min(8, 4096 // max(1, base_blocks), S // 64))
         ws = torch.empty(2, splits, B, Hkv, S, D, device=q.device,
                          dtype=torch.float32)
         check(
