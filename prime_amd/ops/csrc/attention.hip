@@ -1,8 +1,12 @@
-// Flash attention v2 (causal/non-causal, GQA) for gfx950 — MFMA
-// v_mfma_f32_16x16x32_bf16 tiles, online softmax, FA2-style split backward.
+// Flash attention (causal/non-causal, GQA) for gfx950 — MFMA
+// v_mfma_f32_16x16x32_bf16 tiles, online softmax, FA2-style split backward
+// (dQ q-outer; dK/dV kv-outer with a grid-split q loop + fp32 partial
+// workspaces). A 32x32 swapped-operand forward (flash_fwd32_kernel) is
+// kept behind PRIME_ATTN_V4=1 — measured occupancy-bound (see profiles).
 //
-// v2 design (profiled v1 at ~70 TF: redundant per-wave global reads and
-// unswizzled LDS dominated):
+// Design (profiled v1 at ~70 TF: redundant per-wave global reads and
+// unswizzled LDS dominated; measured after fixes: fwd ~240 TF / bwd ~168
+// TF at the 10B shape):
 //  - K/V(+transposed) tiles are staged COOPERATIVELY once per block with
 //    __builtin_amdgcn_global_load_lds width-16 (no VGPR round trip), with
 //    the T2 XOR-16B swizzle applied via the pre-swizzled SOURCE address
