@@ -183,8 +183,15 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
         tmax = fmaxf(tmax, v);
       }
       tmax = grp16_max(tmax);
-      const float mnew = fmaxf(m[r], tmax);
-      alpha[r] = __expf(m[r] - mnew);
+      // defer-max (guide T13): if this tile's max does not exceed the
+      // running max by more than THR, keep the old max — P values stay
+      // bounded by e^THR (fp32 accumulates fine) and the O-wide rescale
+      // pass is skipped (alpha == 1)
+      const float THR = 8.f;
+      float mnew = m[r];
+      if (tmax > m[r] + 0.f) mnew = (tmax - m[r] <= THR && m[r] > NEG_INF)
+                                        ? m[r] : fmaxf(m[r], tmax);
+      alpha[r] = (mnew == m[r]) ? 1.f : __expf(m[r] - mnew);
       float psum = 0.f;
 #pragma unroll
       for (int sub = 0; sub < 4; ++sub) {
@@ -195,10 +202,15 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
       l[r] = l[r] * alpha[r] + grp16_sum(psum);
       m[r] = mnew;
     }
+    float amin = 1.f;
 #pragma unroll
-    for (int dt = 0; dt < DT; ++dt)
+    for (int r = 0; r < 4; ++r) amin = fminf(amin, alpha[r]);
+    if (__ballot(amin < 1.f)) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) o_acc[dt][r] *= alpha[r];
+      for (int dt = 0; dt < DT; ++dt)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) o_acc[dt][r] *= alpha[r];
+    }
     // ---- P (C-layout) -> swizzled per-wave LDS (A-layout source)
 #pragma unroll
     for (int sub = 0; sub < 4; ++sub)
