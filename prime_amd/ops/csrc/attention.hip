@@ -187,10 +187,17 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
       // running max by more than THR, keep the old max — P values stay
       // bounded by e^THR (fp32 accumulates fine) and the O-wide rescale
       // pass is skipped (alpha == 1)
+#ifndef PRIME_DEFER_MAX
+#define PRIME_DEFER_MAX 1
+#endif
+#if PRIME_DEFER_MAX
       const float THR = 8.f;
       float mnew = m[r];
       if (tmax > m[r] + 0.f) mnew = (tmax - m[r] <= THR && m[r] > NEG_INF)
                                         ? m[r] : fmaxf(m[r], tmax);
+#else
+      const float mnew = fmaxf(m[r], tmax);
+#endif
       alpha[r] = (mnew == m[r]) ? 1.f : __expf(m[r] - mnew);
       float psum = 0.f;
 #pragma unroll
