@@ -58,6 +58,15 @@ __device__ __forceinline__ float grp16_sum(float x) {
 
 #define NEG_INF (-1e30f)
 
+#ifndef PRIME_FWD_SETPRIO
+#define PRIME_FWD_SETPRIO 0
+#endif
+#if PRIME_FWD_SETPRIO
+#define FWD_PRIO(x) __builtin_amdgcn_s_setprio(x)
+#else
+#define FWD_PRIO(x)
+#endif
+
 // ---- swizzled-tile helpers --------------------------------------------
 // A [ROWS][COLS] bf16 tile lives linearly in LDS; byte (row, colb) is
 // stored at row*COLS*2 + (colb ^ ((row&7)<<4)). Stage from a strided
@@ -161,6 +170,7 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
 
     // ---- S = scale * Q K^T
     f32x4 s[4];
+    FWD_PRIO(1);
 #pragma unroll
     for (int sub = 0; sub < 4; ++sub) {
       f32x4 acc{0.f, 0.f, 0.f, 0.f};
@@ -169,6 +179,7 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
         acc = mfma16(qf[ds], ld8_swz<D>(k_lds, sub * 16 + li, ds * 64 + lg * 16), acc);
       s[sub] = acc;
     }
+    FWD_PRIO(0);
     // ---- online softmax per q-row (reg r), row owned by 16-lane group
     float alpha[4];
 #pragma unroll
@@ -229,6 +240,7 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
         st16_swz<64>(p_lds, lg * 4 + r, sub * 16 + li, f2bf(s[sub][r]));
     wave_lds_fence();
     // ---- O += P V
+    FWD_PRIO(1);
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
@@ -236,6 +248,7 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
         o_acc[dt] = mfma16(ld8_swz<64>(p_lds, li, ks * 64 + lg * 16),
                            ld8_swz<64>(vt_lds, dt * 16 + li, ks * 64 + lg * 16),
                            o_acc[dt]);
+    FWD_PRIO(0);
   }
   // ---- epilogue
 #pragma unroll
