@@ -63,12 +63,14 @@ class _PinnedStager:
 
 class CheckpointManager:
     def __init__(self, path: str | Path, keep: int = 3, async_save: bool = True,
-                 worker_id: int = 0, is_leader: bool = True):
+                 worker_id: int = 0, is_leader: bool = True,
+                 remote_path: str | Path | None = None):
         self.root = Path(path)
         self.keep = keep
         self.async_save = async_save
         self.worker_id = worker_id
         self.is_leader = is_leader
+        self.remote = Path(remote_path) if remote_path else None
         self._stager = _PinnedStager()
         self._stream = None
         self._pending: threading.Thread | None = None
@@ -108,6 +110,13 @@ class CheckpointManager:
             self._finalize(tag_dir, outer_step, meta)
 
     def _finalize(self, tag_dir: Path, outer_step: int, meta: dict) -> None:
+        if self.remote is not None:
+            # remote copy (e.g. NFS / fuse-mounted object store): performed
+            # on the background persist thread, never the training thread
+            dst = self.remote / tag_dir.name
+            dst.mkdir(parents=True, exist_ok=True)
+            for f in tag_dir.glob(f"worker{self.worker_id}.pt"):
+                shutil.copy2(f, dst / f.name)
         if self.is_leader and self.worker_id == 0:
             (tag_dir / "meta.json").write_text(
                 json.dumps({"outer_step": outer_step, "time": time.time(), **{
@@ -115,6 +124,9 @@ class CheckpointManager:
                     if isinstance(v, (int, float, str, bool, type(None)))
                 }})
             )
+            if self.remote is not None and (tag_dir / "meta.json").exists():
+                shutil.copy2(tag_dir / "meta.json",
+                             self.remote / tag_dir.name / "meta.json")
             latest = self.root / "latest"
             tmp = self.root / ".latest.tmp"
             if tmp.is_symlink() or tmp.exists():

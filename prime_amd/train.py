@@ -151,6 +151,7 @@ class Trainer:
                 cfg.checkpoint.path or (self.run_dir / "ckpt"),
                 keep=cfg.checkpoint.keep, async_save=cfg.checkpoint.async_save,
                 worker_id=self.mesh.worker_id, is_leader=self.mesh.is_leader,
+                remote_path=cfg.checkpoint.remote_path,
             )
         self.step_count = 0
         if cfg.checkpoint.resume and self.ckpt:

@@ -67,6 +67,7 @@ class ParallelConfig(_Strict):
 class CheckpointConfig(_Strict):
     interval: int = 0                 # outer steps between checkpoints; 0=off
     path: str | None = None
+    remote_path: str | None = None    # secondary copy (NFS/fuse mount)
     keep: int = 3
     async_save: bool = True
     resume: str | None = None

@@ -73,3 +73,11 @@ def test_export_safetensors(tmp_path):
     tensors = load_file(str(out))
     assert "tok_embeddings.weight" in tensors
     assert tensors["layers.0.attn.wqkv.weight"].shape[1] == 64
+
+
+def test_remote_copy(tmp_path):
+    mgr = CheckpointManager(tmp_path / "ck", keep=2, async_save=False,
+                            remote_path=tmp_path / "remote")
+    mgr.save(1, {"x": torch.arange(4, dtype=torch.float32)}, {"step": 1})
+    assert (tmp_path / "remote" / "step_1" / "worker0.pt").exists()
+    assert (tmp_path / "remote" / "step_1" / "meta.json").exists()
