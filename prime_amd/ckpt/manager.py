@@ -64,12 +64,14 @@ class _PinnedStager:
 class CheckpointManager:
     def __init__(self, path: str | Path, keep: int = 3, async_save: bool = True,
                  worker_id: int = 0, is_leader: bool = True,
-                 remote_path: str | Path | None = None):
+                 remote_path: str | Path | None = None, shard_rank: int = 0):
         self.root = Path(path)
         self.keep = keep
         self.async_save = async_save
         self.worker_id = worker_id
         self.is_leader = is_leader
+        # FSDP: every rank of a worker persists its own shard file
+        self.shard_rank = shard_rank
         self.remote = Path(remote_path) if remote_path else None
         self._stager = _PinnedStager()
         self._stream = None
@@ -83,7 +85,7 @@ class CheckpointManager:
         self.wait()
         tag_dir = self.root / f"step_{outer_step}"
         tag_dir.mkdir(parents=True, exist_ok=True)
-        fname = tag_dir / f"worker{self.worker_id}.pt"
+        fname = tag_dir / self._fname()
 
         on_gpu = any(t.is_cuda for t in tensors.values())
         if on_gpu and self.async_save:
@@ -109,13 +111,18 @@ class CheckpointManager:
             torch.save({"tensors": host, "meta": meta}, fname)
             self._finalize(tag_dir, outer_step, meta)
 
+    def _fname(self) -> str:
+        if self.shard_rank:
+            return f"worker{self.worker_id}_shard{self.shard_rank}.pt"
+        return f"worker{self.worker_id}.pt"
+
     def _finalize(self, tag_dir: Path, outer_step: int, meta: dict) -> None:
         if self.remote is not None:
             # remote copy (e.g. NFS / fuse-mounted object store): performed
             # on the background persist thread, never the training thread
             dst = self.remote / tag_dir.name
             dst.mkdir(parents=True, exist_ok=True)
-            for f in tag_dir.glob(f"worker{self.worker_id}.pt"):
+            for f in tag_dir.glob(self._fname()):
                 shutil.copy2(f, dst / f.name)
         if self.is_leader and self.worker_id == 0:
             (tag_dir / "meta.json").write_text(
@@ -163,7 +170,7 @@ class CheckpointManager:
         tag_dir = Path(tag) if tag else self.latest_tag()
         if tag_dir is None or not tag_dir.exists():
             return None
-        fname = tag_dir / f"worker{self.worker_id}.pt"
+        fname = tag_dir / self._fname()
         if not fname.exists():
             # joining worker: adopt any peer's snapshot (live recovery fallback)
             cands = sorted(tag_dir.glob("worker*.pt"))

@@ -217,6 +217,11 @@ class ShardedParamSpace:
         one is in flight, and issue the next predicted unit's all-gather
         (idx+1 while ascending through forward, idx-1 through the backward
         recompute walk) so it lands under this unit's compute."""
+        # unit idx-2 shares this unit's grad-pool parity: its in-flight
+        # reduce-scatter must complete before this unit's backward hooks
+        # overwrite the buffer (gather precedes the unit's backward, so
+        # draining here orders the streams correctly)
+        self._drain_rs(unit.idx % 2)
         work = self._prefetch.pop(unit.idx, None)
         if work is not None:
             work.wait()

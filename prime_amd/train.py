@@ -152,6 +152,8 @@ class Trainer:
                 keep=cfg.checkpoint.keep, async_save=cfg.checkpoint.async_save,
                 worker_id=self.mesh.worker_id, is_leader=self.mesh.is_leader,
                 remote_path=cfg.checkpoint.remote_path,
+                # FSDP: optimizer/outer state is per-shard -> every rank saves
+                shard_rank=self.mesh.worker_rank if self.fsdp else 0,
             )
         self.step_count = 0
         if cfg.checkpoint.resume and self.ckpt:
@@ -196,8 +198,8 @@ class Trainer:
             return
         if self.diloco.outer_step_count % self.cfg.checkpoint.interval != 0:
             return
-        if not self.mesh.is_leader:
-            return
+        if not (self.mesh.is_leader or self.fsdp):
+            return  # FSDP: every rank owns a distinct shard and must save
         self.save_checkpoint()
 
     def save_checkpoint(self) -> None:
@@ -321,7 +323,7 @@ class Trainer:
         self.log.info(f"run complete: {result}", type_="result")
         return result
 
-    def close(self) -> None:
+    def close(self, destroy_pg: bool = True) -> None:
         self.wandb.close()
         if self.elastic is not None:
             self.elastic.close(leaving=True)
@@ -329,7 +331,8 @@ class Trainer:
             self.ckpt.wait()
         self.metrics.close()
         self.log.close()
-        self.mesh.destroy()
+        if destroy_pg:
+            self.mesh.destroy()
 
 
 def train_from_config(cfg: TrainConfig, run_dir=None) -> dict:
