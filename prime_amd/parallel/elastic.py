@@ -33,7 +33,6 @@ from datetime import timedelta
 from typing import Callable, Optional
 
 import torch
-import torch.distributed as dist
 from torch.distributed import PrefixStore, ProcessGroupGloo, TCPStore
 
 
