@@ -111,9 +111,15 @@ class DilocoOptimizer:
                 self.theta_outer[c0:c1].copy_(flat.master32[c0:c1])
             self.outer_buf = _host_buf(self._n_comm)
             dev = flat.device
-            self._g_theta = torch.zeros(self.chunk, device=dev, dtype=torch.float32)
-            self._g_buf = torch.zeros(self.chunk, device=dev, dtype=torch.float32)
+            # double-buffered chunk scratch: h2d of chunk i+1 and d2h of
+            # chunk i-1 run on a dedicated copy stream under chunk i's
+            # pseudograd/ring/Nesterov (hipMemcpyAsync + pinned host)
+            self._g_theta = [torch.zeros(self.chunk, device=dev, dtype=torch.float32) for _ in range(2)]
+            self._g_buf = [torch.zeros(self.chunk, device=dev, dtype=torch.float32) for _ in range(2)]
             self._g_delta = torch.zeros(self.chunk, device=dev, dtype=torch.float32)
+            self._copy_stream = torch.cuda.Stream(device=dev)
+            self._h2d_ev = [torch.cuda.Event(), torch.cuda.Event()]
+            self._cmp_ev = [torch.cuda.Event(), torch.cuda.Event()]
 
     # ----------------------------------------------------------------- step
     def step(self, gscale=None) -> bool:
@@ -201,20 +207,42 @@ class DilocoOptimizer:
             f.load_flat_(theta[:n])
 
     def _outer_step_streamed(self) -> None:
-        """Chunked outer step with pinned-host theta/momentum."""
+        """Chunked outer step with pinned-host theta/momentum, pipelined:
+        the copy stream uploads chunk i+1 and drains chunk i-1 while the
+        compute stream runs chunk i's {pseudograd, ring, Nesterov}."""
         f = self.flat
         n = f.numel_padded
-        gt, gb, gd = self._g_theta, self._g_buf, self._g_delta
-        for c0 in range(0, self._n_comm, self.chunk):
+        gd = self._g_delta
+        chunks = list(range(0, self._n_comm, self.chunk))
+        cs = self._copy_stream
+        main = torch.cuda.current_stream(f.device)
+
+        def h2d(i: int) -> None:
+            c0 = chunks[i]
             c1 = min(c0 + self.chunk, self._n_comm)
-            k = c1 - c0                     # == self.chunk (n_comm is padded)
+            b = i % 2
+            with torch.cuda.stream(cs):
+                self._g_theta[b][: c1 - c0].copy_(self.theta_outer[c0:c1], non_blocking=True)
+                self._g_buf[b][: c1 - c0].copy_(self.outer_buf[c0:c1], non_blocking=True)
+                self._h2d_ev[b].record(cs)
+
+        cs.wait_stream(main)  # master32 writes (inner steps) visible
+        h2d(0)
+        for i, c0 in enumerate(chunks):
+            c1 = min(c0 + self.chunk, self._n_comm)
+            k = c1 - c0
             live = max(0, min(n, c1) - c0)  # elements backed by real params
-            gt[:k].copy_(self.theta_outer[c0:c1], non_blocking=True)
-            gb[:k].copy_(self.outer_buf[c0:c1], non_blocking=True)
+            b = i % 2
+            gt, gb = self._g_theta[b], self._g_buf[b]
+            main.wait_event(self._h2d_ev[b])
             if live > 0:
                 ops.pseudograd(gt[:live], f.master32[c0:c0 + live], gd[:live])
             if live < k:
                 gd[live:k].zero_()
+            if i + 1 < len(chunks):
+                # buffer (i+1)%2 is free: chunk i-1's d2h was enqueued on the
+                # copy stream before this h2d (stream order preserves it)
+                h2d(i + 1)
             self._allreduce(gd[:k])
             if live > 0:
                 ops.nesterov_outer(
@@ -222,8 +250,11 @@ class DilocoOptimizer:
                     f.flat_w[c0:c0 + live], gb[:live], gd[:live],
                     lr=self.outer_lr, mu=self.outer_momentum,
                 )
-            self.theta_outer[c0:c1].copy_(gt[:k], non_blocking=True)
-            self.outer_buf[c0:c1].copy_(gb[:k], non_blocking=True)
+            self._cmp_ev[b].record(main)
+            with torch.cuda.stream(cs):
+                cs.wait_event(self._cmp_ev[b])
+                self.theta_outer[c0:c1].copy_(gt[:k], non_blocking=True)
+                self.outer_buf[c0:c1].copy_(gb[:k], non_blocking=True)
         torch.cuda.synchronize()
 
     # ------------------------------------------------------------ ckpt
