@@ -117,9 +117,14 @@ class DilocoOptimizer:
             self._g_theta = [torch.zeros(self.chunk, device=dev, dtype=torch.float32) for _ in range(2)]
             self._g_buf = [torch.zeros(self.chunk, device=dev, dtype=torch.float32) for _ in range(2)]
             self._g_delta = torch.zeros(self.chunk, device=dev, dtype=torch.float32)
-            self._copy_stream = torch.cuda.Stream(device=dev)
+            # separate streams per direction: PCIe Gen5 is full duplex and
+            # the SDMA engines are per-direction — one shared stream would
+            # serialize h2d and d2h at 2x the wall time
+            self._h2d_stream = torch.cuda.Stream(device=dev)
+            self._d2h_stream = torch.cuda.Stream(device=dev)
             self._h2d_ev = [torch.cuda.Event(), torch.cuda.Event()]
             self._cmp_ev = [torch.cuda.Event(), torch.cuda.Event()]
+            self._d2h_ev = [torch.cuda.Event(), torch.cuda.Event()]
 
     # ----------------------------------------------------------------- step
     def step(self, gscale=None) -> bool:
@@ -214,19 +219,21 @@ class DilocoOptimizer:
         n = f.numel_padded
         gd = self._g_delta
         chunks = list(range(0, self._n_comm, self.chunk))
-        cs = self._copy_stream
+        up, down = self._h2d_stream, self._d2h_stream
         main = torch.cuda.current_stream(f.device)
 
         def h2d(i: int) -> None:
             c0 = chunks[i]
             c1 = min(c0 + self.chunk, self._n_comm)
             b = i % 2
-            with torch.cuda.stream(cs):
+            with torch.cuda.stream(up):
+                if i >= 2:  # buffer reuse: chunk i-2's d2h must have drained
+                    up.wait_event(self._d2h_ev[b])
                 self._g_theta[b][: c1 - c0].copy_(self.theta_outer[c0:c1], non_blocking=True)
                 self._g_buf[b][: c1 - c0].copy_(self.outer_buf[c0:c1], non_blocking=True)
-                self._h2d_ev[b].record(cs)
+                self._h2d_ev[b].record(up)
 
-        cs.wait_stream(main)  # master32 writes (inner steps) visible
+        up.wait_stream(main)  # master32 writes (inner steps) visible
         h2d(0)
         for i, c0 in enumerate(chunks):
             c1 = min(c0 + self.chunk, self._n_comm)
@@ -251,10 +258,11 @@ class DilocoOptimizer:
                     lr=self.outer_lr, mu=self.outer_momentum,
                 )
             self._cmp_ev[b].record(main)
-            with torch.cuda.stream(cs):
-                cs.wait_event(self._cmp_ev[b])
+            with torch.cuda.stream(down):
+                down.wait_event(self._cmp_ev[b])
                 self.theta_outer[c0:c1].copy_(gt[:k], non_blocking=True)
                 self.outer_buf[c0:c1].copy_(gb[:k], non_blocking=True)
+                self._d2h_ev[b].record(down)
         torch.cuda.synchronize()
 
     # ------------------------------------------------------------ ckpt
