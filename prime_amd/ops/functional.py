@@ -28,6 +28,11 @@ class _RMSNorm(torch.autograd.Function):
         x = x.contiguous()
         shp = x.shape
         D = shp[-1]
+        if D % 8 != 0:
+            raise NotImplementedError(
+                f"rmsnorm kernel needs hidden dim % 8 == 0 (got {D}): the "
+                "bf16x8 vector loads require 16 B rows"
+            )
         R = x.numel() // D
         y = torch.empty_like(x)
         rstd = torch.empty(R, device=x.device, dtype=torch.float32)
@@ -127,6 +132,10 @@ class _SwiGLU(torch.autograd.Function):
             return ref.swiglu(gu)
         gu = gu.contiguous()
         I = gu.shape[-1] // 2
+        if I % 8 != 0:
+            raise NotImplementedError(
+                f"swiglu kernel needs intermediate dim % 8 == 0 (got {I})"
+            )
         R = gu.numel() // (2 * I)
         out = torch.empty(*gu.shape[:-1], I, device=gu.device, dtype=gu.dtype)
         check(lib().prime_swiglu_fwd(stream_of(gu), ptr(gu), ptr(out), R, I), "swiglu_fwd")
