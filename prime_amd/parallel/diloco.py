@@ -145,7 +145,11 @@ class DilocoOptimizer:
             from .elastic import EvictedError
 
             try:
-                self._view = self.elastic.sync()
+                steps_this_round = self.inner_step_count - getattr(
+                    self, "_last_boundary_step", 0)
+                self._view = self.elastic.sync(
+                    contribution=min(1.0, steps_this_round / max(1, self.H)))
+                self._last_boundary_step = self.inner_step_count
             except EvictedError:
                 # stalled past the heartbeat timeout and got evicted:
                 # re-register, adopt a live peer's outer state, skip this
@@ -176,6 +180,9 @@ class DilocoOptimizer:
 
                 v = self._view
                 if v is not None and v.world > 1:
+                    scale = v.my_scale()
+                    if scale != 1.0:
+                        delta_chunk.mul_(scale)  # weighted outer average
                     if delta_chunk.is_cuda:
                         host = delta_chunk.to("cpu")
                         ring_allreduce_int8_pg(host, v.pg, v.my_index, v.world)

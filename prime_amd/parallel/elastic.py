@@ -46,10 +46,21 @@ class ElasticView:
     members: list[str]
     my_index: int
     pg: Optional[ProcessGroupGloo]
+    weights: list[float] | None = None  # per-member contribution weights
 
     @property
     def world(self) -> int:
         return len(self.members)
+
+    def my_scale(self) -> float:
+        """Pre-scale for this worker's delta so the ring's plain average
+        yields the WEIGHTED mean: delta_i * (w_i * W / sum(w))."""
+        if not self.weights:
+            return 1.0
+        tot = sum(self.weights)
+        if tot <= 0:
+            return 1.0
+        return self.weights[self.my_index] * len(self.weights) / tot
 
 
 @dataclass
@@ -153,12 +164,15 @@ class ElasticWorker:
                 pass
 
     # -------------------------------------------------------- boundary sync
-    def sync(self, build_pg: bool = True) -> ElasticView:
+    def sync(self, build_pg: bool = True, contribution: float = 1.0) -> ElasticView:
         """Called by every leader at an outer-step boundary; returns the
-        agreed membership view and a fresh gloo group over it."""
+        agreed membership view and a fresh gloo group over it.
+        `contribution` (e.g. inner steps completed this round / H) becomes
+        this worker's weight in the outer average — a freshly-joined
+        worker with a partial round contributes proportionally."""
         while True:
             n = self.epoch + 1
-            self.store.set(f"ready/{n}/{self.wid}", "1")
+            self.store.set(f"ready/{n}/{self.wid}", f"{max(0.0, contribution):.6f}")
             view_key = f"view/{n}"
             t0 = _now()
             members: list[str] = []
@@ -182,7 +196,13 @@ class ElasticWorker:
                     PrefixStore(f"pg/{n}/", self.store),
                     members.index(self.wid), len(members),
                 )
-            return ElasticView(n, members, members.index(self.wid), pg)
+            weights = []
+            for w in members:
+                try:
+                    weights.append(float(self.store.get(f"ready/{n}/{w}").decode()))
+                except Exception:  # noqa: BLE001 — legacy "1" or missing
+                    weights.append(1.0)
+            return ElasticView(n, members, members.index(self.wid), pg, weights)
 
     def _try_arbiter(self, n: int, t_wait_start: float) -> bool:
         """Attempt the arbiter role; returns True once view/<n> exists."""

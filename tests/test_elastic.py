@@ -299,3 +299,35 @@ def test_evicted_worker_auto_rejoins():
     assert out[1]["rejoined"] == 1
     assert out[0]["worlds"][0] == 1   # after eviction
     assert out[0]["worlds"][1] == 2   # after rejoin
+
+
+def _weighted_worker(port, host, contribution):
+    import torch
+
+    from prime_amd.parallel.elastic import ElasticWorker, ring_allreduce_int8_pg
+    from prime_amd.ops import QBLK
+
+    w = ElasticWorker(port=port, host_store=host, worker_name="wt",
+                      heartbeat_interval=0.3, heartbeat_timeout=5.0,
+                      ckpt_provider=lambda: {})
+    while len(w.members()) < 2:
+        time.sleep(0.05)
+    view = w.sync(contribution=contribution)
+    delta = torch.full((view.world * QBLK,), 1.0 if host else 3.0)
+    delta.mul_(view.my_scale())
+    ring_allreduce_int8_pg(delta, view.pg, view.my_index, view.world)
+    w.close()
+    return {"weights": view.weights, "avg": float(delta[0])}
+
+
+def test_weighted_outer_average():
+    port = free_port()
+    # worker A contributed a full round (w=1.0), worker B half (w=0.5):
+    # weighted mean of (1.0, 3.0) = (1*1 + 0.5*3)/1.5 = 5/3
+    out = _spawn([
+        (_weighted_worker, {}, (port, True, 1.0)),
+        (_weighted_worker, {}, (port, False, 0.5)),
+    ])
+    assert sorted(out[0]["weights"]) == [0.5, 1.0]
+    assert abs(out[0]["avg"] - 5.0 / 3.0) < 0.05
+    assert abs(out[1]["avg"] - 5.0 / 3.0) < 0.05
