@@ -172,7 +172,16 @@ class ElasticWorker:
         worker with a partial round contributes proportionally."""
         while True:
             n = self.epoch + 1
-            self.store.set(f"ready/{n}/{self.wid}", f"{max(0.0, contribution):.6f}")
+            try:
+                self.store.set(f"ready/{n}/{self.wid}",
+                               f"{max(0.0, contribution):.6f}")
+            except Exception as e:  # noqa: BLE001
+                raise RuntimeError(
+                    "elastic store unreachable — the TCPStore host is gone. "
+                    "Run a standalone registry (`prime-amd store`) so worker "
+                    "churn cannot take the store down, then restart workers "
+                    f"against it. ({e})"
+                ) from e
             view_key = f"view/{n}"
             t0 = _now()
             members: list[str] = []
