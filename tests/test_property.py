@@ -1,0 +1,59 @@
+"""Property-based tests (hypothesis) for the pure numeric building blocks."""
+import numpy as np
+import torch
+from hypothesis import given, settings, strategies as st
+
+from prime_amd.ops import QBLK
+from prime_amd.ops import reference as ref
+from prime_amd.parallel.ring import _gcd, _modinv, _ring_pos
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.integers(2, 16), st.integers(1, 15))
+def test_ring_pos_bijective(W, o):
+    if _gcd(o % W if o % W else 1, W) != 1 or o % W == 0:
+        return
+    o = o % W
+    # ownership after reduce-scatter covers every partition exactly once
+    assert sorted(_ring_pos(r, o, W, 1) for r in range(W)) == list(range(W))
+    # modinv correctness
+    assert (o * _modinv(o, W)) % W == 1
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.integers(1, 4), st.floats(1e-6, 1e4))
+def test_quant_roundtrip_bound(nblk, scale):
+    g = torch.Generator().manual_seed(nblk)
+    x = torch.randn(nblk * QBLK, generator=g) * scale
+    q, s = ref.quant_int8_blockwise(x)
+    back = ref.dequant_int8_blockwise(q, s)
+    # per-block error bounded by that block's scale (amax/127)
+    xb = x.view(nblk, QBLK)
+    bb = back.view(nblk, QBLK)
+    for b in range(nblk):
+        bound = xb[b].abs().max() / 127 + 1e-12
+        assert (bb[b] - xb[b]).abs().max() <= bound * 1.01
+
+
+@settings(max_examples=20, deadline=None)
+@given(st.integers(1, 8), st.integers(1, 64))
+def test_rope_orthogonality(heads, seq):
+    cos, sin = ref.rope_tables(16, 128)
+    g = torch.Generator().manual_seed(heads * 100 + seq)
+    x = torch.randn(1, seq, heads, 16, generator=g)
+    y = ref.apply_rope(x, cos, sin)
+    torch.testing.assert_close(x.norm(dim=-1), y.norm(dim=-1), atol=1e-4, rtol=1e-4)
+    back = ref.apply_rope(y, cos, -sin)
+    torch.testing.assert_close(back, x, atol=1e-5, rtol=1e-5)
+
+
+@settings(max_examples=20, deadline=None)
+@given(st.integers(2, 64), st.integers(2, 6))
+def test_softmax_reference_attention_rows_sum(seq, heads):
+    g = torch.Generator().manual_seed(seq * heads)
+    q = torch.randn(1, seq, heads, 16, generator=g)
+    k = torch.randn(1, seq, heads, 16, generator=g)
+    v = torch.ones(1, seq, heads, 16)
+    # attention over constant V returns constant rows (probabilities sum to 1)
+    o = ref.attention(q, k, v, causal=True)
+    torch.testing.assert_close(o, torch.ones_like(o), atol=1e-5, rtol=1e-5)
