@@ -106,6 +106,9 @@ def main() -> None:
             "dtype": "bf16",
             "data": "synthetic",
             "mfu": mfu,
+            # BASELINE.md convention: record the peak used for MFU (DENSE
+            # bf16; AMD's 5 PF headline is 2:1-sparse)
+            "mfu_peak_flops": 2.5e15,
             "config": {
                 "model": args.model,
                 "global_batch": args.micro_batch * n_gpus,

@@ -324,6 +324,36 @@ def run_cmd(config: str, detach: bool) -> None:
     run(config, nproc=0, detach=detach)
 
 
+@train_app.command("report")
+def report(run: str = typer.Argument(...)):
+    """Summarize a run: status, loss trajectory, throughput, failures."""
+    from ..utils.metrics import read_metrics
+
+    d = _find_run(run)
+    st = _status(d)
+    rows = read_metrics(d / "metrics.jsonl")
+    typer.secho(f"run {d.name}: {st.get('status', '?')}", bold=True)
+    fa = st.get("failure_analysis")
+    if fa:
+        typer.secho(f"  failure: {fa['category']} — {fa['hint']}", fg="red")
+    if not rows:
+        typer.echo("  no metrics recorded")
+        return
+    losses = [r["loss"] for r in rows if "loss" in r]
+    tps = [r["tokens_per_sec"] for r in rows if "tokens_per_sec" in r]
+    mfus = [r["mfu"] for r in rows if "mfu" in r]
+    typer.echo(f"  steps logged : {rows[0]['step']} .. {rows[-1]['step']}")
+    if losses:
+        typer.echo(f"  loss         : first {losses[0]:.4f}  min {min(losses):.4f}  last {losses[-1]:.4f}")
+    if tps:
+        typer.echo(f"  tokens/sec   : mean {sum(tps)/len(tps):,.0f}  last {tps[-1]:,.0f}")
+    if mfus:
+        typer.echo(f"  MFU          : mean {sum(mfus)/len(mfus):.3f}  last {mfus[-1]:.3f}")
+    res = st.get("result")
+    if res:
+        typer.echo(f"  result       : {json.dumps(res)}")
+
+
 @train_app.command("models")
 def models():
     """List model presets (reference: `prime train models`)."""

@@ -64,7 +64,7 @@ class Trainer:
             {"model": cfg.model.name, "H": cfg.diloco.H},
         )
 
-        torch.manual_seed(1234)
+        torch.manual_seed(cfg.seed)
         # construct directly on the target device: CPU-side init of a 10B
         # model takes minutes; on-GPU init is seconds
         with torch.device(self.device):

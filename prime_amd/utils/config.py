@@ -83,6 +83,7 @@ class MetricsConfig(_Strict):
 class TrainConfig(_Strict):
     run_name: str = "run"
     steps: int = 100
+    seed: int = 1234
     device: str | None = None         # cuda | cpu (default: auto)
     model: ModelConfig = Field(default_factory=ModelConfig)
     data: DataSection = Field(default_factory=DataSection)

@@ -179,3 +179,16 @@ def test_delete_run(tmp_path, runs_dir):
     r = runner.invoke(app, ["train", "delete", run_id])
     assert r.exit_code == 0
     assert json.loads(runner.invoke(app, ["train", "list", "--json"]).output) == []
+
+
+def test_report_command(tmp_path, runs_dir):
+    cfg = tmp_path / "r.toml"
+    cfg.write_text('run_name = "rep"\nsteps = 2\n'
+                   '[model]\nname = "llama_test"\nseq_len = 64\n'
+                   '[data]\nmicro_batch_size = 2\n[metrics]\nlog_interval = 1\n')
+    r = runner.invoke(app, ["train", "run", str(cfg)])
+    assert r.exit_code == 0
+    run_id = json.loads(runner.invoke(app, ["train", "list", "--json"]).output)[0]["run"]
+    r = runner.invoke(app, ["train", "report", run_id])
+    assert r.exit_code == 0
+    assert "loss" in r.output and "tokens/sec" in r.output
