@@ -6,6 +6,11 @@ worker (worker_size=1, weak scaling: fixed per-GPU batch); the outer int8
 ring all-reduce runs every H=100 inner steps (and once during warmup so the
 path is exercised even when K < H). Rank 0 prints ONE JSON line with the
 whole-job tokens/sec.
+
+Host-memory sizing at N=8 x 10B: each rank keeps fp32 theta_outer +
+momentum on the host (2 x 42 GB), i.e. ~672 GB across the node; the
+allocation is pinned only when MemAvailable comfortably covers it
+(PRIME_AMD_OUTER_PIN to override) and falls back to pageable otherwise.
 """
 from __future__ import annotations
 
