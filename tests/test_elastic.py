@@ -331,3 +331,45 @@ def test_weighted_outer_average():
     assert sorted(out[0]["weights"]) == [0.5, 1.0]
     assert abs(out[0]["avg"] - 5.0 / 3.0) < 0.05
     assert abs(out[1]["avg"] - 5.0 / 3.0) < 0.05
+
+
+def _departing_trainer(port, host, steps):
+    from prime_amd.train import Trainer
+    from prime_amd.utils.config import (
+        DilocoConfig, MetricsConfig, ModelConfig, ParallelConfig, TrainConfig,
+    )
+
+    cfg = TrainConfig(
+        run_name=f"dep_{host}",
+        steps=steps,
+        model=ModelConfig(name="llama_test", seq_len=64),
+        diloco=DilocoConfig(H=2),
+        parallel=ParallelConfig(elastic=True, heartbeat_interval=0.3,
+                                heartbeat_timeout=8.0),
+        metrics=MetricsConfig(log_interval=100),
+    )
+    cfg.data.micro_batch_size = 2
+    tr = Trainer(cfg, run_dir=f"/tmp/prime_amd_test/depart_{host}_{steps}")
+    while len(tr.elastic.members()) < 2:
+        time.sleep(0.05)
+    res = tr.run()
+    last_world = tr.diloco._view.world if tr.diloco._view else 1
+    tr.close()  # graceful leave: sets leaving/<wid> + deregisters
+    return {"outer_steps": res["outer_steps"], "last_world": last_world}
+
+
+def test_graceful_departure_mid_run():
+    """Worker A runs 4 steps (2 boundaries) then leaves; worker B runs 8
+    steps — its later boundaries must proceed at world=1, not hang."""
+    port = free_port()
+    env = {"PRIME_GLOBAL_PORT": port, "WORLD_SIZE": 1, "RANK": 0,
+           "MASTER_ADDR": "127.0.0.1"}
+    out = _spawn([
+        (_departing_trainer, {**env, "PRIME_GLOBAL_HOST": 1,
+                              "MASTER_PORT": free_port()}, (port, True, 8)),
+        (_departing_trainer, {**env, "PRIME_GLOBAL_HOST": 0,
+                              "MASTER_PORT": free_port()}, (port, False, 4)),
+    ], timeout=300)
+    assert out[1]["outer_steps"] == 2          # departed after 2 boundaries
+    assert out[0]["outer_steps"] == 4          # survivor completed all
+    assert out[0]["last_world"] == 1           # finished alone
