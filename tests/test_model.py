@@ -137,3 +137,22 @@ def test_large_configs_build_on_meta():
             m = build_model(name)
         n = sum(p.numel() for p in m.parameters())
         assert lo < n < hi, (name, n)
+
+
+def test_evaluate_perplexity_token_file(tmp_path):
+    import numpy as np
+
+    from prime_amd.data import DataConfig
+    from prime_amd.models.evaluate import evaluate_perplexity
+
+    toks = np.random.default_rng(0).integers(0, 256, 5000).astype(np.uint16)
+    p = tmp_path / "t.bin"
+    toks.tofile(p)
+    m = build_model("llama_test")
+    res = evaluate_perplexity(
+        m, DataConfig(kind="token_file", path=str(p), seq_len=64,
+                      micro_batch_size=2),
+        n_batches=3,
+    )
+    assert res["tokens"] == 3 * 2 * 64
+    assert 1 < res["perplexity"] < 1e4
