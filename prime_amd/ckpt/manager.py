@@ -60,6 +60,20 @@ class _PinnedStager:
                 out[k] = buf
         return out
 
+    def stage_cpu(self, state: dict[str, torch.Tensor]) -> dict:
+        """Synchronous copy of CPU tensors into reusable staging buffers so
+        the caller may keep mutating the originals while a background
+        thread serializes the snapshot."""
+        out = {}
+        for k, t in state.items():
+            buf = self.buffers.get(k)
+            if buf is None or buf.shape != t.shape or buf.dtype != t.dtype:
+                buf = torch.empty_like(t, device="cpu")
+                self.buffers[k] = buf
+            buf.copy_(t)
+            out[k] = buf
+        return out
+
 
 class CheckpointManager:
     def __init__(self, path: str | Path, keep: int = 3, async_save: bool = True,
@@ -95,9 +109,18 @@ class CheckpointManager:
             host = self._stager.stage(
                 {k: t for k, t in tensors.items() if t.is_cuda}, self._stream
             )
-            host.update({k: t for k, t in tensors.items() if not t.is_cuda})
+            # CPU tensors (host-offloaded theta_outer/outer_buf) are copied
+            # into reusable staging buffers: a later outer step must not
+            # mutate what the background torch.save is serializing.
+            host.update(self._stager.stage_cpu(
+                {k: t for k, t in tensors.items() if not t.is_cuda}
+            ))
             ev = torch.cuda.Event()
             ev.record(self._stream)
+            # The compute stream must not mutate master32/m/v (the very next
+            # train_step's fused AdamW does) while the side stream is still
+            # reading them for the D2H staging copies.
+            torch.cuda.current_stream().wait_event(ev)
 
             def _persist():
                 ev.synchronize()

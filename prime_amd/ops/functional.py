@@ -303,16 +303,23 @@ class _CrossEntropy(torch.autograd.Function):
         logits = logits.contiguous()
         t32 = targets.to(torch.int32).contiguous()
         loss = torch.empty(R, device=logits.device, dtype=torch.float32)
-        dlogits = torch.empty_like(logits)
+        # under no_grad (eval/perplexity) skip the [R,V] dlogits write —
+        # at V=128k that is a full extra bf16 tensor per eval batch
+        need_grad = torch.is_grad_enabled() and logits.requires_grad
+        dlogits = torch.empty_like(logits) if need_grad else logits  # dummy ptr
         check(
             lib().prime_cross_entropy(
                 stream_of(logits), ptr(logits), ptr(t32), ptr(loss), ptr(dlogits),
-                R, V, 1.0, ignore_index, 1,
+                R, V, 1.0, ignore_index, 1 if need_grad else 0,
             ),
             "cross_entropy",
         )
         n_valid = (targets != ignore_index).sum().clamp(min=1)
-        ctx.save_for_backward(dlogits, n_valid)
+        if need_grad:
+            ctx.save_for_backward(dlogits, n_valid)
+        else:
+            ctx.save_for_backward(torch.empty(0, device=logits.device,
+                                              dtype=logits.dtype), n_valid)
         return loss.sum() / n_valid.float()
 
     @staticmethod

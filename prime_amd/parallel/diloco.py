@@ -52,6 +52,8 @@ class DilocoOptimizer:
         chunk_elems: int = 256 * 1024 * 1024,  # 1 GB fp32 per streamed chunk
         elastic=None,  # ElasticWorker (leader rank only)
         elastic_mode: bool = False,  # True on ALL ranks of an elastic worker
+        shard_client=None,  # ElasticShardClient (non-leader FSDP ranks)
+        sharded: bool = False,  # flat is a ShardedParamSpace (FSDP)
     ):
         self.flat = flat
         self.mesh = mesh
@@ -61,7 +63,10 @@ class DilocoOptimizer:
         self.H = H
         self.elastic = elastic
         self.elastic_mode = elastic_mode or elastic is not None
+        self.shard_client = shard_client
+        self.sharded = sharded
         self._view = None
+        self._scale = 1.0  # this worker's outer-average pre-scale
         import threading
 
         self._outer_lock = threading.Lock()
@@ -140,32 +145,154 @@ class DilocoOptimizer:
         return True
 
     def outer_step(self) -> None:
-        if self.elastic is not None:
-            # boundary: agree on membership + rebuild the cross-worker group
-            from .elastic import EvictedError
-
-            try:
-                steps_this_round = self.inner_step_count - getattr(
-                    self, "_last_boundary_step", 0)
-                self._view = self.elastic.sync(
-                    contribution=min(1.0, steps_this_round / max(1, self.H)))
-                self._last_boundary_step = self.inner_step_count
-            except EvictedError:
-                # stalled past the heartbeat timeout and got evicted:
-                # re-register, adopt a live peer's outer state, skip this
-                # boundary, and contribute again from the next one
-                self.elastic.rejoin()
-                payload = self.elastic.bootstrap_from_peer()
-                if payload is not None:
-                    self.load_bootstrap(payload)
-                self.rejoined = getattr(self, "rejoined", 0) + 1
-                return
+        if self.elastic_mode and not self._elastic_boundary():
+            return  # evicted this boundary: all ranks skip together
         with self._outer_lock:
             if self.outer_device == "gpu":
                 self._outer_step_resident()
             else:
                 self._outer_step_streamed()
         self.outer_step_count += 1
+
+    def _elastic_boundary(self) -> bool:
+        """Outer-boundary membership agreement for elastic workers.
+
+        The leader runs the TCPStore sync (possibly getting evicted and
+        rejoining); every local rank then learns the SAME branch via an
+        object broadcast — without it, non-leader ranks of a multi-rank
+        worker would walk into the chunk loop's collectives alone and
+        deadlock. Sharded (FSDP) ranks additionally build their own
+        shard-aligned cross-worker gloo group. Returns False when this
+        boundary must be skipped (just evicted)."""
+        from .elastic import ElasticView, EvictedError
+
+        info = None
+        if self.elastic is not None:
+            try:
+                steps_this_round = self.inner_step_count - getattr(
+                    self, "_last_boundary_step", 0)
+                self._view = self.elastic.sync(
+                    contribution=min(1.0, steps_this_round / max(1, self.H)))
+                self._last_boundary_step = self.inner_step_count
+                v = self._view
+                info = {"evicted": False, "bootstrapped": False,
+                        "epoch": v.epoch, "world": v.world,
+                        "my_index": v.my_index, "scale": v.my_scale(),
+                        "wid": self.elastic.wid}
+            except EvictedError:
+                # stalled past the heartbeat timeout and got evicted:
+                # re-register, adopt a live peer's outer state, skip this
+                # boundary, and contribute again from the next one
+                self.elastic.rejoin()
+                payload = self.elastic.bootstrap_from_peer()
+                boot = payload is not None
+                if boot:
+                    self.load_bootstrap(payload)
+                self.rejoined = getattr(self, "rejoined", 0) + 1
+                self._view = None
+                info = {"evicted": True, "bootstrapped": boot,
+                        "wid": self.elastic.wid}
+        lg = self.mesh.local_group
+        src = self.mesh.worker_id * self.mesh.cfg.worker_size
+        if lg is not None:
+            import torch.distributed as dist
+
+            obj = [info]
+            dist.broadcast_object_list(obj, src=src, group=lg)
+            info = obj[0]
+            if self.shard_client is not None and info and info.get("wid"):
+                self.shard_client.set_wid(info["wid"])
+        if info is None:  # single-rank non-leader cannot happen; be safe
+            return True
+        self._scale = float(info.get("scale", 1.0))
+        if info["evicted"]:
+            if info["bootstrapped"] and lg is not None:
+                if self.sharded:
+                    if self.shard_client is not None:
+                        payload = self.shard_client.bootstrap_shard()
+                        if payload is not None:
+                            self.load_bootstrap(payload)
+                else:
+                    self._bcast_outer_state(src)
+            self._last_boundary_step = self.inner_step_count
+            return False
+        # sharded non-leader ranks: build this epoch's shard-aligned group
+        if (self.sharded and self.shard_client is not None
+                and info["world"] > 1):
+            pg = self.shard_client.build_pg(info["epoch"], info["my_index"],
+                                            info["world"])
+            self._view = ElasticView(info["epoch"], [""] * info["world"],
+                                     info["my_index"], pg, None)
+        elif self.elastic is None:
+            self._view = (ElasticView(info["epoch"], [""], 0, None, None)
+                          if info["world"] <= 1 else self._view)
+        return True
+
+    def _bcast_outer_state(self, src: int) -> None:
+        """Replicate the leader's (just-bootstrapped) outer state to the
+        other ranks of a DP elastic worker, so a rejoin cannot leave leader
+        and non-leader parameters diverged."""
+        import torch.distributed as dist
+
+        lg = self.mesh.local_group
+        dev = self.flat.device
+        n = self.flat.numel_padded
+        counts = torch.tensor([self.inner_step_count, self.outer_step_count],
+                              dtype=torch.int64, device=dev)
+        dist.broadcast(counts, src=src, group=lg)
+        self.inner_step_count = int(counts[0])
+        self.outer_step_count = int(counts[1])
+        self._last_boundary_step = self.inner_step_count
+        if self.outer_device == "gpu":
+            dist.broadcast(self.theta_outer, src=src, group=lg)
+            dist.broadcast(self.outer_buf, src=src, group=lg)
+        else:
+            # host-offloaded state: chunk through the device scratch
+            scratch = self._g_theta[0]
+            for c0 in range(0, self._n_comm, self.chunk):
+                c1 = min(c0 + self.chunk, self._n_comm)
+                k = c1 - c0
+                scratch[:k].copy_(self.theta_outer[c0:c1])
+                dist.broadcast(scratch[:k], src=src, group=lg)
+                self.theta_outer[c0:c1].copy_(scratch[:k])
+                scratch[:k].copy_(self.outer_buf[c0:c1])
+                dist.broadcast(scratch[:k], src=src, group=lg)
+                self.outer_buf[c0:c1].copy_(scratch[:k])
+        th = self.theta_outer[:n]
+        self.flat.load_flat_(th.to(self.flat.master32.device)
+                             if th.device != self.flat.master32.device else th)
+
+    def init_bootstrap(self) -> bool:
+        """Startup live recovery: the leader pulls a live peer's outer
+        state; a multi-rank worker then syncs so every rank starts
+        consistent (FSDP ranks fetch their own shard from the peer's
+        matching rank). Returns True when state was adopted."""
+        info = None
+        if self.elastic is not None:
+            payload = self.elastic.bootstrap_from_peer()
+            boot = payload is not None
+            if boot:
+                self.load_bootstrap(payload)
+            info = {"bootstrapped": boot, "wid": self.elastic.wid}
+        lg = self.mesh.local_group
+        if self.elastic_mode and lg is not None:
+            import torch.distributed as dist
+
+            src = self.mesh.worker_id * self.mesh.cfg.worker_size
+            obj = [info]
+            dist.broadcast_object_list(obj, src=src, group=lg)
+            info = obj[0]
+            if self.shard_client is not None and info and info.get("wid"):
+                self.shard_client.set_wid(info["wid"])
+            if info and info["bootstrapped"]:
+                if self.sharded:
+                    if self.shard_client is not None:
+                        payload = self.shard_client.bootstrap_shard()
+                        if payload is not None:
+                            self.load_bootstrap(payload)
+                else:
+                    self._bcast_outer_state(src)
+        return bool(info and info["bootstrapped"])
 
     def _allreduce(self, delta_chunk: torch.Tensor) -> None:
         """Average a (padded) delta chunk across workers.
@@ -175,21 +302,23 @@ class DilocoOptimizer:
         worker's other ranks over the local (RCCL) group. Static mode: the
         mesh's pre-built outer group (int8 ring over RCCL P2P)."""
         if self.elastic_mode:
-            if self.elastic is not None:
+            # who rings cross-worker: the leader always; with FSDP every
+            # rank rings its OWN shard over its shard-aligned group
+            rings = self.elastic is not None or (
+                self.sharded and self.shard_client is not None)
+            v = self._view
+            if rings and v is not None and v.world > 1:
                 from .elastic import ring_allreduce_int8_pg
 
-                v = self._view
-                if v is not None and v.world > 1:
-                    scale = v.my_scale()
-                    if scale != 1.0:
-                        delta_chunk.mul_(scale)  # weighted outer average
-                    if delta_chunk.is_cuda:
-                        host = delta_chunk.to("cpu")
-                        ring_allreduce_int8_pg(host, v.pg, v.my_index, v.world)
-                        delta_chunk.copy_(host, non_blocking=True)
-                    else:
-                        ring_allreduce_int8_pg(delta_chunk, v.pg, v.my_index, v.world)
-            if self.mesh.local_group is not None:
+                if self._scale != 1.0:
+                    delta_chunk.mul_(self._scale)  # weighted outer average
+                if delta_chunk.is_cuda:
+                    host = delta_chunk.to("cpu")
+                    ring_allreduce_int8_pg(host, v.pg, v.my_index, v.world)
+                    delta_chunk.copy_(host, non_blocking=True)
+                else:
+                    ring_allreduce_int8_pg(delta_chunk, v.pg, v.my_index, v.world)
+            if not self.sharded and self.mesh.local_group is not None:
                 import torch.distributed as dist
 
                 src = self.mesh.worker_id * self.mesh.cfg.worker_size
@@ -296,6 +425,10 @@ class DilocoOptimizer:
             self.outer_buf[:k].copy_(payload["outer_buf"][:k].to(self.outer_buf.device))
             self.inner_step_count = int(payload["inner_step"])
             self.outer_step_count = int(payload["outer_step"])
+            # the adopted step count is the peer's, not steps WE trained:
+            # without resetting the boundary marker the first outer sync
+            # would clamp this joiner's contribution to 1.0
+            self._last_boundary_step = self.inner_step_count
             th = self.theta_outer[:n]
             self.flat.load_flat_(th.to(self.flat.master32.device)
                                  if th.device != self.flat.master32.device else th)

@@ -98,6 +98,7 @@ class ElasticWorker:
         )
         # unique ordered worker id: join sequence number + name
         seq = int(self.store.add("join_seq", 1))
+        self.join_seq = seq  # worker-unique (data sharding uses it)
         self.wid = f"{seq:06d}-{worker_name or os.getpid()}"
         self.epoch = int(self._get_str("epoch", "0"))
         self._hb_seq = 0
@@ -121,16 +122,31 @@ class ElasticWorker:
         self.store.set(f"members/{self.wid}", "1")
         self.store.set(f"hb/{self.wid}", "0")
 
+    def _index_cas(self, transform) -> list[str]:
+        """Apply a pure transform to the member_index mirror with a
+        compare_set retry loop — a plain get/set read-modify-write lets two
+        concurrent joiners each read the old index and overwrite the
+        other's registration (this key is the membership source of truth)."""
+        for _ in range(64):
+            exists = self.store.check(["member_index"])
+            cur = self.store.get("member_index").decode() if exists else "[]"
+            new = json.dumps(transform(json.loads(cur)))
+            if new == cur:
+                return json.loads(cur)
+            got = self.store.compare_set("member_index", cur if exists else "", new)
+            if got.decode() == new:
+                return json.loads(new)
+        raise RuntimeError("member_index compare_set did not converge")
+
     def members(self) -> list[str]:
         """Sorted registry (join order == lexicographic by construction)."""
         # TCPStore has no key listing; keep a mirror index
-        idx = json.loads(self._get_str("member_index", "[]"))
-        if self.wid not in idx:
-            idx = sorted(set(idx) | {self.wid})
-            self.store.set("member_index", json.dumps(idx))
+        idx = self._index_cas(lambda i: sorted(set(i) | {self.wid}))
         live = [w for w in idx if self.store.check([f"members/{w}"])]
         if live != idx:
-            self.store.set("member_index", json.dumps(live))
+            dead = set(idx) - set(live)
+            live = self._index_cas(lambda i: sorted(set(i) - dead))
+            live = [w for w in live if self.store.check([f"members/{w}"])]
         return live
 
     # ------------------------------------------------------------ heartbeat
@@ -201,8 +217,11 @@ class ElasticWorker:
                 continue  # missed this round's cut; try next boundary
             pg = None
             if build_pg and len(members) > 1:
+                # shard-aligned prefix: rank r of every worker rings over
+                # pg/<epoch>/r<r>/ — the leader is shard 0; non-leader ranks
+                # of an FSDP worker build theirs via ElasticShardClient
                 pg = ProcessGroupGloo(
-                    PrefixStore(f"pg/{n}/", self.store),
+                    PrefixStore(f"pg/{n}/r0/", self.store),
                     members.index(self.wid), len(members),
                 )
             weights = []
@@ -251,6 +270,8 @@ class ElasticWorker:
         self._ckpt_srv = srv
         host, port = srv.getsockname()
         self.store.set(f"ckptsrv/{self.wid}", f"{host}:{port}")
+        # shard-aligned alias: the leader serves shard 0 of an FSDP worker
+        self.store.set(f"ckptsrv/{self.wid}/r0", f"{host}:{port}")
 
         def _serve():
             while not self._stop.is_set():
@@ -301,6 +322,7 @@ class ElasticWorker:
         fresh heartbeat identity, re-advertised checkpoint server. The
         caller should re-bootstrap state from a peer before contributing."""
         seq = int(self.store.add("join_seq", 1))
+        self.join_seq = seq
         old = self.wid
         self.wid = f"{seq:06d}-rejoin-{old.split('-', 1)[1]}"
         self._peer_tracks.clear()
@@ -309,6 +331,7 @@ class ElasticWorker:
         if self._ckpt_srv is not None:
             host, port = self._ckpt_srv.getsockname()
             self.store.set(f"ckptsrv/{self.wid}", f"{host}:{port}")
+            self.store.set(f"ckptsrv/{self.wid}/r0", f"{host}:{port}")
 
     # --------------------------------------------------------------- leave
     def close(self, leaving: bool = True) -> None:
@@ -318,6 +341,115 @@ class ElasticWorker:
                 self._evict(self.wid)
             except Exception:  # noqa: BLE001
                 pass
+        self._stop.set()
+        if self._ckpt_srv is not None:
+            try:
+                self._ckpt_srv.close()
+            except OSError:
+                pass
+
+
+class ElasticShardClient:
+    """A non-leader rank's handle on the elastic fabric (FSDP workers).
+
+    The leader (ElasticWorker) owns membership; this client lets shard
+    rank r of the worker (a) build the shard-aligned cross-worker gloo
+    group pg/<epoch>/r<r>/ once the leader has broadcast the agreed view,
+    (b) serve its OWN shard for live peer recovery under
+    ckptsrv/<leader-wid>/r<r>, and (c) bootstrap its shard from the
+    matching rank of any live peer worker (peers must share worker_size)."""
+
+    def __init__(self, shard_rank: int, addr: str = None, port: int = None,
+                 ckpt_provider: Callable[[], dict] | None = None,
+                 host_ip: str | None = None):
+        self.addr = addr or os.environ.get("PRIME_GLOBAL_ADDR", "127.0.0.1")
+        self.port = int(port or os.environ.get("PRIME_GLOBAL_PORT", 29777))
+        self.host_ip = host_ip or os.environ.get("PRIME_HOST_IP", "127.0.0.1")
+        self.shard_rank = shard_rank
+        self.wid: str | None = None  # leader's wid, set via set_wid()
+        self.store = TCPStore(self.addr, self.port, is_master=False,
+                              wait_for_workers=False,
+                              timeout=timedelta(seconds=60))
+        self._stop = threading.Event()
+        self._ckpt_provider = ckpt_provider
+        self._ckpt_srv: Optional[socket.socket] = None
+        if ckpt_provider is not None:
+            self._start_ckpt_server()
+
+    def set_wid(self, wid: str) -> None:
+        """(Re)publish this shard's checkpoint server under the leader's
+        current wid (changes after an eviction+rejoin)."""
+        if wid == self.wid:
+            return
+        self.wid = wid
+        if self._ckpt_srv is not None:
+            host, port = self._ckpt_srv.getsockname()
+            self.store.set(f"ckptsrv/{wid}/r{self.shard_rank}", f"{host}:{port}")
+
+    def build_pg(self, epoch: int, my_index: int, world: int) -> ProcessGroupGloo:
+        return ProcessGroupGloo(
+            PrefixStore(f"pg/{epoch}/r{self.shard_rank}/", self.store),
+            my_index, world,
+        )
+
+    def _start_ckpt_server(self) -> None:
+        srv = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+        srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        srv.bind((self.host_ip, 0))
+        srv.listen(4)
+        self._ckpt_srv = srv  # store key published by set_wid()
+
+        def _serve():
+            while not self._stop.is_set():
+                try:
+                    srv.settimeout(1.0)
+                    conn, _ = srv.accept()
+                except socket.timeout:
+                    continue
+                except OSError:
+                    return
+                try:
+                    payload = self._ckpt_provider()
+                    buf = io.BytesIO()
+                    torch.save(payload, buf)
+                    raw = buf.getvalue()
+                    conn.sendall(struct.pack("<Q", len(raw)))
+                    conn.sendall(raw)
+                except Exception:  # noqa: BLE001 — keep serving other peers
+                    pass
+                finally:
+                    conn.close()
+
+        threading.Thread(target=_serve, daemon=True).start()
+
+    def bootstrap_shard(self) -> dict | None:
+        """Fetch this shard's live state from the matching rank of any
+        live peer worker."""
+        try:
+            exists = self.store.check(["member_index"])
+            idx = json.loads(self.store.get("member_index").decode()) if exists else []
+        except Exception:  # noqa: BLE001
+            return None
+        for wid in idx:
+            # skip ourselves and evicted members (their shard ckptsrv keys
+            # outlive _evict, which only knows the leader-level key)
+            if wid == self.wid or not self.store.check([f"members/{wid}"]):
+                continue
+            key = f"ckptsrv/{wid}/r{self.shard_rank}"
+            if not self.store.check([key]):
+                continue
+            host, port = self.store.get(key).decode().rsplit(":", 1)
+            try:
+                with socket.create_connection((host, int(port)), timeout=30) as c:
+                    hdr = _recv_exact(c, 8)
+                    (length,) = struct.unpack("<Q", hdr)
+                    raw = _recv_exact(c, length)
+                return torch.load(io.BytesIO(raw), map_location="cpu", weights_only=False)
+            except OSError:
+                continue
+        return None
+
+    def close(self) -> None:
         self._stop.set()
         if self._ckpt_srv is not None:
             try:
@@ -346,12 +478,14 @@ def ring_allreduce_int8_pg(delta: torch.Tensor, pg, rank: int, world: int,
     as prime_amd.parallel.ring (which needs a registered group)."""
     from .. import ops
     from ..ops import QBLK
-    from .ring import _dequant_add, _quant
+    from .ring import _dequant_add, _pad_run, _quant
 
     if world == 1:
         return
     n = delta.numel()
-    assert n % (world * QBLK) == 0
+    if _pad_run(delta, world * QBLK,
+                lambda b: ring_allreduce_int8_pg(b, pg, rank, world, average)):
+        return
     part = n // world
     parts = [delta[i * part : (i + 1) * part] for i in range(world)]
     nxt, prv = (rank + 1) % world, (rank - 1) % world

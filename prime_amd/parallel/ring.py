@@ -35,20 +35,37 @@ def _dequant_add(q, scales, dst, accumulate: bool):
             dst.copy_(v)
 
 
+def _pad_run(delta: torch.Tensor, align: int, fn) -> bool:
+    """Run `fn(buf)` on a zero-padded copy of `delta` when its size is not
+    a multiple of `align`, copying the result back. Keeps the ring usable
+    for ANY live world size (an elastic fleet can land on 3/5/6/7 workers
+    exactly when fault tolerance matters), not just divisors of the
+    construction-time alignment. Returns True when it ran padded."""
+    n = delta.numel()
+    if n % align == 0:
+        return False
+    pad_n = (n + align - 1) // align * align
+    buf = torch.zeros(pad_n, dtype=delta.dtype, device=delta.device)
+    buf[:n].copy_(delta)
+    fn(buf)
+    delta.copy_(buf[:n])
+    return True
+
+
 def ring_allreduce_int8(
     delta: torch.Tensor,
     group: dist.ProcessGroup | None = None,
     average: bool = True,
 ) -> None:
     """In-place sum (or average) of fp32 `delta` across `group` with int8
-    compression. Requires delta.numel() % (W * QBLK) == 0 — callers pad
-    (FlatParamSpace buffers are padded by DilocoOptimizer)."""
+    compression. Sizes not divisible by W*QBLK are zero-padded per call."""
     W = dist.get_world_size(group)
     if W == 1:
         return
     rank = dist.get_rank(group)
     n = delta.numel()
-    assert n % (W * QBLK) == 0, f"delta size {n} not divisible by W*QBLK={W * QBLK}"
+    if _pad_run(delta, W * QBLK, lambda b: ring_allreduce_int8(b, group, average)):
+        return
     part = n // W
     parts = [delta[i * part : (i + 1) * part] for i in range(W)]
     nxt = (rank + 1) % W
@@ -136,9 +153,9 @@ def ring_allreduce_int8_multi(
     if R <= 1:
         return ring_allreduce_int8(delta, group, average)
     n = delta.numel()
-    assert n % (R * W * QBLK) == 0, (
-        f"delta size {n} not divisible by R*W*QBLK={R * W * QBLK}"
-    )
+    if _pad_run(delta, R * W * QBLK,
+                lambda b: ring_allreduce_int8_multi(b, group, average, n_rings)):
+        return
     sub = n // R
     subs = [delta[r * sub : (r + 1) * sub] for r in range(R)]
     part = sub // W

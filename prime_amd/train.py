@@ -108,6 +108,7 @@ class Trainer:
             eps=cfg.optim.eps, weight_decay=cfg.optim.weight_decay,
         )
         self.elastic = None
+        self.shard_client = None
         if cfg.parallel.elastic and self.mesh.is_leader:
             from .parallel.elastic import ElasticWorker
 
@@ -118,6 +119,15 @@ class Trainer:
                 heartbeat_timeout=cfg.parallel.heartbeat_timeout,
                 ckpt_provider=lambda: self.diloco.live_state(),
             )
+        elif cfg.parallel.elastic and self.fsdp:
+            # non-leader FSDP rank: own store client + shard ckpt server +
+            # shard-aligned cross-worker ring (see parallel/elastic.py)
+            from .parallel.elastic import ElasticShardClient
+
+            self.shard_client = ElasticShardClient(
+                shard_rank=self.mesh.worker_rank,
+                ckpt_provider=lambda: self.diloco.live_state(),
+            )
         self.diloco = DilocoOptimizer(
             self.flat, self.mesh, self.inner,
             outer_lr=cfg.diloco.outer_lr,
@@ -126,18 +136,34 @@ class Trainer:
             outer_device=cfg.diloco.outer_device,
             elastic=self.elastic,
             elastic_mode=cfg.parallel.elastic,
+            shard_client=self.shard_client,
+            sharded=self.fsdp,
         )
-        if self.elastic is not None:
-            payload = self.elastic.bootstrap_from_peer()
-            if payload is not None:
-                self.diloco.load_bootstrap(payload)
+        if cfg.parallel.elastic:
+            if self.diloco.init_bootstrap():
                 self.log.info(
-                    f"live-recovered from peer at outer step {payload['outer_step']}"
+                    "live-recovered from peer at outer step "
+                    f"{self.diloco.outer_step_count}"
                 )
 
+        # DiLoCo-worker data index: distinct workers must see distinct
+        # streams. Elastic workers are separate torchrun jobs with identical
+        # rank/world, so fold a worker-unique index (config override, else
+        # the elastic join sequence) into the seed.
+        widx = cfg.data.worker_index
+        if widx is None and cfg.parallel.elastic:
+            widx = self.elastic.join_seq if self.elastic is not None else 0
+            if self.mesh.local_group is not None:
+                import torch.distributed as dist
+
+                t = torch.tensor([widx], dtype=torch.int64, device=self.device)
+                # elastic: this whole torchrun job is one worker, leader = 0
+                dist.broadcast(t, src=0, group=self.mesh.local_group)
+                widx = int(t[0])
         data_cfg = DataConfig(
             kind=cfg.data.kind, path=cfg.data.path, seq_len=cfg.model.seq_len,
-            micro_batch_size=cfg.data.micro_batch_size, seed=cfg.data.seed,
+            micro_batch_size=cfg.data.micro_batch_size,
+            seed=cfg.data.seed + 100003 * (widx or 0),
             shuffle=cfg.data.shuffle,
         )
         self.data = build_dataloader(
@@ -327,6 +353,8 @@ class Trainer:
         self.wandb.close()
         if self.elastic is not None:
             self.elastic.close(leaving=True)
+        if self.shard_client is not None:
+            self.shard_client.close()
         if self.ckpt:
             self.ckpt.wait()
         self.metrics.close()

@@ -33,6 +33,9 @@ class DataSection(_Strict):
     grad_accum: int = 1
     seed: int = 1234
     shuffle: bool = True
+    # explicit DiLoCo-worker data index; elastic mode derives one from the
+    # join sequence when unset so workers never train identical streams
+    worker_index: int | None = None
 
 
 class OptimConfig(_Strict):

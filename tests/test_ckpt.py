@@ -81,3 +81,13 @@ def test_remote_copy(tmp_path):
     mgr.save(1, {"x": torch.arange(4, dtype=torch.float32)}, {"step": 1})
     assert (tmp_path / "remote" / "step_1" / "worker0.pt").exists()
     assert (tmp_path / "remote" / "step_1" / "meta.json").exists()
+
+
+def test_cpu_tensors_staged_not_live(tmp_path):
+    """A later outer step mutating host-resident theta_outer in place must
+    not corrupt the snapshot a background save is serializing."""
+    mgr = CheckpointManager(tmp_path / "ck", async_save=False)
+    t = torch.arange(8, dtype=torch.float32)
+    staged = mgr._stager.stage_cpu({"theta_outer": t})
+    t.mul_(100.0)  # in-place mutation after staging
+    assert staged["theta_outer"][3].item() == 3.0

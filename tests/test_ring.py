@@ -84,3 +84,42 @@ def test_multi_ring_int8_three_ranks():
     for err, scale, _ in outs:
         assert err < scale * 6 / 127 + 1e-6, err
     assert outs[0][2] == outs[2][2]
+
+
+def _unaligned_worker(rank, world, multi):
+    """Ring with a size NOT divisible by W*QBLK — exercises per-call
+    padding (an elastic fleet can land on any live world size)."""
+    import torch.distributed as dist
+
+    from prime_amd.parallel import ring
+    from prime_amd.ops import QBLK
+
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    torch.manual_seed(3)
+    n = world * QBLK * 2 + 177  # deliberately unaligned
+    full = torch.randn(world, n)
+    mine = full[rank].clone()
+    want = full.mean(0)
+    if multi:
+        ring.ring_allreduce_int8_multi(mine, average=True)
+    else:
+        ring.ring_allreduce_int8(mine, average=True)
+    err = (mine - want).abs().max().item()
+    scale = full.abs().max().item()
+    dist.barrier()
+    dist.destroy_process_group()
+    return err, scale, mine[-8:].tolist()
+
+
+def test_ring_int8_unaligned_three_ranks():
+    outs = run_distributed(_unaligned_worker, 3, args=(False,))
+    for err, scale, _ in outs:
+        assert err < scale * 6 / 127 + 1e-6, err
+    assert outs[0][2] == outs[1][2] == outs[2][2]
+
+
+def test_multi_ring_int8_unaligned_five_ranks():
+    outs = run_distributed(_unaligned_worker, 5, args=(True,))
+    for err, scale, _ in outs:
+        assert err < scale * 10 / 127 + 1e-6, err
+    assert outs[0][2] == outs[4][2]
