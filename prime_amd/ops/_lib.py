@@ -12,7 +12,7 @@ from typing import Optional
 
 import torch
 
-from .build import LIB_PATH, build
+from .build import LIB_PATH, build, needs_build
 
 _LIB: Optional[ctypes.CDLL] = None
 _TRIED = False
@@ -49,8 +49,9 @@ def _load() -> ctypes.CDLL:
 
     override = os.environ.get("PRIME_AMD_LIB_PATH")  # A/B testing of builds
     path = override or str(LIB_PATH)
-    if override is None and not LIB_PATH.exists():
-        # last resort: try building (hipcc cross-compiles without a GPU)
+    if override is None and needs_build():
+        # stale (source-hash stamp mismatch) or missing: rebuild — hipcc
+        # cross-compiles without a GPU and exists on every target image
         build(verbose=True)
     lib = ctypes.CDLL(path)
     for name, argtypes in _SIGS.items():

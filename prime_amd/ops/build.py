@@ -21,12 +21,27 @@ def sources() -> list[Path]:
     return sorted(CSRC.glob("*.hip"))
 
 
+def source_hash() -> str:
+    """Content hash of every source + the arch — mtimes are useless after
+    a fresh checkout (everything gets the checkout time, so a stale
+    prebuilt .so would win silently)."""
+    import hashlib
+
+    h = hashlib.sha256()
+    h.update(ARCH.encode())
+    for s in sources() + sorted(CSRC.glob("*.h")):
+        h.update(s.name.encode())
+        h.update(s.read_bytes())
+    return h.hexdigest()
+
+
 def needs_build() -> bool:
     if not LIB_PATH.exists():
         return True
-    lib_mtime = LIB_PATH.stat().st_mtime
-    deps = sources() + list(CSRC.glob("*.h"))
-    return any(s.stat().st_mtime > lib_mtime for s in deps)
+    stamp = LIB_PATH.with_suffix(".so.hash")
+    if not stamp.exists():
+        return True
+    return stamp.read_text().strip() != source_hash()
 
 
 def build(force: bool = False, verbose: bool = True) -> Path:
@@ -50,6 +65,7 @@ def build(force: bool = False, verbose: bool = True) -> Path:
     if verbose:
         print(f"[prime_amd] building {LIB_PATH.name}: {' '.join(cmd)}", flush=True)
     subprocess.run(cmd, check=True, cwd=str(CSRC))
+    LIB_PATH.with_suffix(".so.hash").write_text(source_hash() + "\n")
     return LIB_PATH
 
 

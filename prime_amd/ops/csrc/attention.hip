@@ -265,13 +265,18 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
 }
 
 
-// ------------------------------------------------- forward v4 (32x32 MFMA)
-// Swapped-operand structure (guide §B attn, 8-warp ladder): St = K·Q^T with
-// v_mfma_f32_32x32x16_bf16 puts each q-row's scores LANE-LOCAL (lane pair
-// (l, l^32) splits the row), so the online softmax is in-register (no
-// cross-16-lane shuffle chains) and P feeds the PV A-fragment directly
-// after ONE lane-pair exchange — the per-tile P LDS bounce of the 16x16
-// kernel disappears. 4 waves x 32 q-rows = 128-row q tiles.
+// ------------------------------------------------- forward v5 (32x32 MFMA)
+// 8-wave swapped-operand structure (guide §B attn, 8-warp 32x32 ladder):
+// St = K·Q^T with v_mfma_f32_32x32x16_bf16 puts each q-row's scores
+// LANE-LOCAL (lane pair (l, l^32) splits the row), so the online softmax
+// is fully in-register and P feeds the PV A-fragment after one lane-pair
+// exchange (v_cvt_pk_bf16_f32 pack + v_permlane32_swap_b32 — guide T12).
+// 8 waves x 32 q-rows = 256-row q tiles sharing one K/Vt stage (vs the
+// 16x16 kernel's per-wave P LDS bounce + 4x staging traffic). The round-1
+// v4 (4-wave, shfl-based exchange, always-rescale) measured 130 TF; the
+// deltas here are the rest of the technique stack: defer-max rescale
+// skipping (T13), cheap P exchange, per-wave causal trip clipping, and
+// setprio around the MFMA clusters (T5).
 //
 // 32x32x16 fragment maps (HW-verified, test_mfma32_layout_vs_matmul):
 //   A[32][16]: lane holds A[lane&31][(lane>>5)*8 + j]
@@ -283,23 +288,44 @@ __device__ __forceinline__ f32x16 mfma32(short8 a, short8 b, f32x16 c) {
   return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
 }
 
-template <int D>  // D == 128 only
-__global__ __launch_bounds__(256) void flash_fwd32_kernel(
+// pack two f32 into one dword of 2 bf16 (no builtin on gfx950 — guide T12)
+__device__ __forceinline__ unsigned cvt_pk_bf16(float lo_, float hi_) {
+  unsigned r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo_), "v"(hi_));
+  return r;
+}
+
+// v_permlane32_swap_b32 a, b: a's lanes>=32 swap with b's lanes<32 —
+// after the swap, lane l<32 sees (a_l, a_{l+32}) and lane l>=32 sees
+// (b_{l-32}, b_l) in (a, b)
+__device__ __forceinline__ void permlane32_swap(unsigned& a, unsigned& b) {
+  asm volatile("v_permlane32_swap_b32 %0, %1" : "+v"(a), "+v"(b));
+}
+
+// FLAGS bits: 1 = setprio around MFMA clusters, 2 = defer-max (THR=8)
+template <int D, int FLAGS>
+__global__ __launch_bounds__(512, 2) void flash_fwd_v5_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ Vt, bf16* __restrict__ O, float* __restrict__ lse,
     int B, int H, int Hkv, int S, float scale, int causal,
     int64_t sqb, int64_t sqs, int64_t sqh,
     int64_t skb, int64_t sks, int64_t skh) {
-  constexpr int DSL = D / 16;  // 16-wide d slices for the K dim (8)
-  const int n_qt = S / 128;
-  const int bh = blockIdx.x / n_qt;
-  const int qt = n_qt - 1 - (blockIdx.x - bh * n_qt);  // longest trips first
+  constexpr int DSL = D / 16;   // 16-wide d slices for the K dim
+  constexpr int DT32 = D / 32;  // 32-wide output d tiles
+  const int n_qt = S / 256;
+  // XCD-aware bijective remap (guide T1): consecutive qt-tiles of one
+  // (b,h) land on ONE XCD's L2, which then reuses the same K/V stream
+  const int nwg = gridDim.x;
+  int vb = blockIdx.x;
+  if ((nwg & 7) == 0) vb = (blockIdx.x & 7) * (nwg >> 3) + (blockIdx.x >> 3);
+  const int bh = vb / n_qt;
+  const int qt = n_qt - 1 - (vb - bh * n_qt);  // longest trips first
   const int b = bh / H, h = bh - b * H;
   const int hkv = h / (H / Hkv);
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int lo = lane & 31, hi = lane >> 5;
-  const int q0w = qt * 128 + wid * 32;  // this wave's first q row
+  const int q0w = qt * 256 + wid * 32;  // this wave's first q row
 
   const bf16* Qb = Q + b * sqb + h * sqh;
   const bf16* Kb = K + b * skb + hkv * skh;
@@ -307,10 +333,10 @@ __global__ __launch_bounds__(256) void flash_fwd32_kernel(
 
   __shared__ bf16 k_lds[2][64 * D];
   __shared__ bf16 vt_lds[2][D * 64];
-  __shared__ float bcast_all[4][32];  // per-wave alpha / inv-l broadcast
+  __shared__ float bcast_all[8][32];  // per-wave alpha / inv-l broadcast
   float* bcast = bcast_all[wid];
 
-  // Q^T B-fragments, pre-scaled by 1/sqrt(D): lane holds Q[q0w+lo][ds*16+hi*8+j]
+  // Q^T B-fragments, pre-scaled: lane holds Q[q0w+lo][ds*16+hi*8+j]
   short8 qf[DSL];
 #pragma unroll
   for (int ds = 0; ds < DSL; ++ds) {
@@ -326,29 +352,33 @@ __global__ __launch_bounds__(256) void flash_fwd32_kernel(
   }
 
   float m_run = NEG_INF, l_run = 0.f;
-  f32x16 o_acc[4];  // O[32q x 128d]: 4 col-tiles of 32 d
+  f32x16 o_acc[DT32];
 #pragma unroll
-  for (int dt = 0; dt < 4; ++dt)
+  for (int dt = 0; dt < DT32; ++dt)
 #pragma unroll
     for (int r = 0; r < 16; ++r) o_acc[dt][r] = 0.f;
 
-  const int kv_end = causal ? (qt * 128 + 128) : S;
-  // prefetch-behind-barrier: at 2 waves/SIMD (VGPR-bound) there is no
-  // other-wave overlap to hide a synchronous stage; give the next tile's
-  // loads the whole current tile's compute to land
-  stage_tile<64, D>(Kb, sks, k_lds[0], threadIdx.x);
-  stage_tile<D, 64>(Vtb, S, vt_lds[0], threadIdx.x);
+  const int kv_end = causal ? (qt * 256 + 256) : S;
+  // per-wave causal clip: this wave's rows end at q0w+31, so tiles past
+  // q0w+32 are fully masked — skip their compute (they still stage+sync)
+  const int my_kv_end = causal ? (q0w + 32) : S;
+  stage_tile<64, D, 512>(Kb, sks, k_lds[0], threadIdx.x);
+  stage_tile<D, 64, 512>(Vtb, S, vt_lds[0], threadIdx.x);
   int idx = 0;
   for (int kv = 0; kv < kv_end; kv += 64, idx ^= 1) {
-    __syncthreads();
+    __syncthreads();  // prev tile consumed + this tile's stage drained
     if (kv + 64 < kv_end) {
-      stage_tile<64, D>(Kb + (int64_t)(kv + 64) * sks, sks, k_lds[idx ^ 1], threadIdx.x);
-      stage_tile<D, 64>(Vtb + kv + 64, S, vt_lds[idx ^ 1], threadIdx.x);
+      // prefetch-behind-barrier: these loads get the whole tile's compute
+      // to land (the NEXT loop iteration's syncthreads drains them)
+      stage_tile<64, D, 512>(Kb + (int64_t)(kv + 64) * sks, sks,
+                             k_lds[idx ^ 1], threadIdx.x);
+      stage_tile<D, 64, 512>(Vtb + kv + 64, S, vt_lds[idx ^ 1], threadIdx.x);
     }
+    if (kv >= my_kv_end) continue;  // fully-masked tile for this wave
 
-    // ---- St = (K q^T): two 32x32 C tiles over the 64-key block.
-    // A-frag rows = k (lane&31), cols = d slice; read from swizzled K LDS.
+    // ---- St = (K q^T): two 32x32 C tiles over the 64-key block
     f32x16 st[2];
+    if (FLAGS & 1) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
 #pragma unroll
@@ -359,8 +389,9 @@ __global__ __launch_bounds__(256) void flash_fwd32_kernel(
         st[t] = mfma32(kf, qf[ds], st[t]);
       }
     }
-    // ---- online softmax: lane pair (lo, hi) and (lo, hi^1) split the
-    // 64 scores of q row q0w+lo; st[t][r] is k = kv + t*32 + crow(r,hi)
+    if (FLAGS & 1) __builtin_amdgcn_s_setprio(0);
+    // ---- online softmax: st[t][r] is k = kv + t*32 + crow(r,hi) for
+    // q row q0w+lo; lane pair (l, l^32) splits the row's 64 scores
     float tmax = NEG_INF;
 #pragma unroll
     for (int t = 0; t < 2; ++t)
@@ -371,80 +402,68 @@ __global__ __launch_bounds__(256) void flash_fwd32_kernel(
         tmax = fmaxf(tmax, st[t][r]);
       }
     tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
-    const float mnew = fmaxf(m_run, tmax);
-    const float alpha = __expf(m_run - mnew);
+    // defer-max (guide T13): skip the O-wide rescale while the running
+    // max still bounds P by e^THR (fp32 accumulation tolerates it)
+    const float thr = (FLAGS & 2) ? 8.f : 0.f;
+    float alpha = 1.f;
+    if (__ballot(tmax > m_run + thr)) {
+      const float mnew = fmaxf(m_run, tmax);
+      alpha = __expf(m_run - mnew);  // 1 for rows that did not grow
+      m_run = mnew;
+      if (hi == 0) bcast[lo] = alpha;
+      wave_lds_fence();
+#pragma unroll
+      for (int dt = 0; dt < DT32; ++dt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r)
+          o_acc[dt][r] *= bcast[(r & 3) + 8 * (r >> 2) + 4 * hi];
+    }
     float psum = 0.f;
 #pragma unroll
     for (int t = 0; t < 2; ++t)
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        const float pv = (st[t][r] <= NEG_INF) ? 0.f : __expf(st[t][r] - mnew);
+        const float pv = (st[t][r] <= NEG_INF) ? 0.f : __expf(st[t][r] - m_run);
         st[t][r] = pv;
         psum += pv;
       }
     psum += __shfl_xor(psum, 32, 64);
     l_run = l_run * alpha + psum;
-    m_run = mnew;
 
-    // ---- O *= alpha, broadcast per C row (rows are crow(r,hi), but alpha
-    // lives on lane q=lo): per-wave LDS broadcast
-    if (hi == 0) bcast[lo] = alpha;
-    wave_lds_fence();
-#pragma unroll
-    for (int dt = 0; dt < 4; ++dt)
-#pragma unroll
-      for (int r = 0; r < 16; ++r)
-        o_acc[dt][r] *= bcast[(r & 3) + 8 * (r >> 2) + 4 * hi];
-
-    // ---- P -> PV A-fragments: pack to bf16 and exchange the partner
-    // half-chunks (4 values = 2 dwords) per 16-k slice
-    short8 pa[4];
+    // ---- P -> PV A-fragments (guide T12): per 16-k slice, pack the two
+    // 4-value r-groups to bf16 dwords and permlane32_swap — lane l<32
+    // keeps x and receives partner's x; lane>=32 receives partner's y and
+    // keeps y. Result is branch-free and identical on both halves.
+    short8 pa[2 * 2];
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
 #pragma unroll
       for (int kst = 0; kst < 2; ++kst) {
-        // my kept chunk: r group g_keep = 2*kst + hi; my sent chunk g_send
-        // = 2*kst + (1-hi); both packed as 2 dwords of 2 bf16
-        const int g_keep = 2 * kst + hi;
-        const int g_send = 2 * kst + 1 - hi;
-        unsigned keep0, keep1, send0, send1;
-        {
-          const bf16 a0 = f2bf(st[t][4 * g_keep + 0]), a1 = f2bf(st[t][4 * g_keep + 1]);
-          const bf16 a2 = f2bf(st[t][4 * g_keep + 2]), a3 = f2bf(st[t][4 * g_keep + 3]);
-          keep0 = (unsigned)reinterpret_cast<const unsigned short&>(a0) |
-                  ((unsigned)reinterpret_cast<const unsigned short&>(a1) << 16);
-          keep1 = (unsigned)reinterpret_cast<const unsigned short&>(a2) |
-                  ((unsigned)reinterpret_cast<const unsigned short&>(a3) << 16);
-          const bf16 b0 = f2bf(st[t][4 * g_send + 0]), b1 = f2bf(st[t][4 * g_send + 1]);
-          const bf16 b2 = f2bf(st[t][4 * g_send + 2]), b3 = f2bf(st[t][4 * g_send + 3]);
-          send0 = (unsigned)reinterpret_cast<const unsigned short&>(b0) |
-                  ((unsigned)reinterpret_cast<const unsigned short&>(b1) << 16);
-          send1 = (unsigned)reinterpret_cast<const unsigned short&>(b2) |
-                  ((unsigned)reinterpret_cast<const unsigned short&>(b3) << 16);
-        }
-        const unsigned got0 = __shfl_xor((int)send0, 32, 64);
-        const unsigned got1 = __shfl_xor((int)send1, 32, 64);
-        // A-frag element order j0..j7 = k 16*kst + 8*hi + j; first 4 come
-        // from the hi'=0 owner, last 4 from the hi'=1 owner
-        unsigned w0, w1, w2, w3;
-        if (hi == 0) { w0 = keep0; w1 = keep1; w2 = got0; w3 = got1; }
-        else         { w0 = got0;  w1 = got1;  w2 = keep0; w3 = keep1; }
+        const int g0 = 4 * (2 * kst), g1 = 4 * (2 * kst + 1);
+        unsigned x0 = cvt_pk_bf16(st[t][g0 + 0], st[t][g0 + 1]);
+        unsigned x1 = cvt_pk_bf16(st[t][g0 + 2], st[t][g0 + 3]);
+        unsigned y0 = cvt_pk_bf16(st[t][g1 + 0], st[t][g1 + 1]);
+        unsigned y1 = cvt_pk_bf16(st[t][g1 + 2], st[t][g1 + 3]);
+        permlane32_swap(x0, y0);
+        permlane32_swap(x1, y1);
         short8 frag;
-        reinterpret_cast<unsigned*>(&frag)[0] = w0;
-        reinterpret_cast<unsigned*>(&frag)[1] = w1;
-        reinterpret_cast<unsigned*>(&frag)[2] = w2;
-        reinterpret_cast<unsigned*>(&frag)[3] = w3;
+        reinterpret_cast<unsigned*>(&frag)[0] = x0;
+        reinterpret_cast<unsigned*>(&frag)[1] = x1;
+        reinterpret_cast<unsigned*>(&frag)[2] = y0;
+        reinterpret_cast<unsigned*>(&frag)[3] = y1;
         pa[t * 2 + kst] = frag;
       }
     }
     // ---- O += P V  (B-frags from swizzled Vt LDS)
+    if (FLAGS & 1) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt)
+    for (int dt = 0; dt < DT32; ++dt)
 #pragma unroll
       for (int ks = 0; ks < 4; ++ks)
         o_acc[dt] = mfma32(pa[ks],
                            ld8_swz<64>(vt_lds[idx], dt * 32 + lo, ks * 32 + hi * 16),
                            o_acc[dt]);
+    if (FLAGS & 1) __builtin_amdgcn_s_setprio(0);
   }
   // ---- epilogue: O /= l (per-row broadcast), write + lse
   if (hi == 0) bcast[lo] = (l_run > 0.f) ? 1.f / l_run : 0.f;
@@ -455,7 +474,7 @@ __global__ __launch_bounds__(256) void flash_fwd32_kernel(
     const float inv = bcast[qr];
     bf16* orow = O + (((int64_t)(b * S + q0w + qr)) * H + h) * D;
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt)
+    for (int dt = 0; dt < DT32; ++dt)
       orow[dt * 32 + lo] = f2bf(o_acc[dt][r] * inv);
   }
   if (hi == 0)
@@ -478,6 +497,158 @@ __global__ void attn_delta_kernel(const bf16* __restrict__ dO,
     }
     acc = wave_reduce_sum(acc);
     if (lane == 0) delta[r] = acc;
+  }
+}
+
+// ------------------------------------------------ backward dQ v5 (32x32)
+// Same 8-wave swapped-operand structure as the v5 forward: St = K·Q^T and
+// dPt = V·dO^T give lane-local q-rows (q = lane&31), so lse/delta are one
+// scalar per lane and dS never touches LDS — it is packed straight into
+// the dQ-accumulation A-fragments with cvt_pk + permlane32_swap.
+template <int D, int FLAGS>
+__global__ __launch_bounds__(512, 2) void flash_bwd_dq_v5_kernel(
+    const bf16* __restrict__ Q, const bf16* __restrict__ K,
+    const bf16* __restrict__ V, const bf16* __restrict__ Kt,
+    const bf16* __restrict__ dO, const float* __restrict__ lse,
+    const float* __restrict__ delta, bf16* __restrict__ dQ, int B, int H,
+    int Hkv, int S, float scale, int causal,
+    int64_t sqb, int64_t sqs, int64_t sqh,
+    int64_t skb, int64_t sks, int64_t skh,
+    int64_t svb, int64_t svs, int64_t svh) {
+  constexpr int DSL = D / 16;
+  constexpr int DT32 = D / 32;
+  const int n_qt = S / 256;
+  const int nwg = gridDim.x;
+  int vb = blockIdx.x;
+  if ((nwg & 7) == 0) vb = (blockIdx.x & 7) * (nwg >> 3) + (blockIdx.x >> 3);
+  const int bh = vb / n_qt;
+  const int qt = n_qt - 1 - (vb - bh * n_qt);
+  const int b = bh / H, h = bh - b * H;
+  const int hkv = h / (H / Hkv);
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lo = lane & 31, hi = lane >> 5;
+  const int q0w = qt * 256 + wid * 32;
+
+  const bf16* Qb = Q + b * sqb + h * sqh;
+  const bf16* Kb = K + b * skb + hkv * skh;
+  const bf16* Vb = V + b * svb + hkv * svh;
+  const bf16* Ktb = Kt + ((int64_t)(b * Hkv + hkv) * D) * S;
+  const bf16* dOb = dO + ((int64_t)b * S * H + h) * D;  // row stride H*D
+  const float* lse_b = lse + (int64_t)b * S * H + h;    // stride H
+  const float* dl_b = delta + (int64_t)b * S * H + h;
+
+  __shared__ bf16 k_lds[2][64 * D];
+  __shared__ bf16 v_lds[2][64 * D];
+  __shared__ bf16 kt_lds[2][D * 64];
+
+  // Q^T / dO^T B-fragments for this wave's 32 q rows (q = lo)
+  short8 qf[DSL], dof[DSL];
+#pragma unroll
+  for (int ds = 0; ds < DSL; ++ds) {
+    const bf16* src = Qb + (int64_t)(q0w + lo) * sqs + ds * 16 + hi * 8;
+    short8 raw = ld8(src);
+    short8 sc;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      bf16 v = reinterpret_cast<const bf16*>(&raw)[j];
+      reinterpret_cast<bf16*>(&sc)[j] = f2bf(bf2f(v) * scale);
+    }
+    qf[ds] = sc;
+    dof[ds] = ld8(dOb + (int64_t)(q0w + lo) * H * D + ds * 16 + hi * 8);
+  }
+  const float lse_q = lse_b[(int64_t)(q0w + lo) * H];
+  const float dl_q = dl_b[(int64_t)(q0w + lo) * H];
+
+  f32x16 dq_acc[DT32];
+#pragma unroll
+  for (int dt = 0; dt < DT32; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dq_acc[dt][r] = 0.f;
+
+  const int kv_end = causal ? (qt * 256 + 256) : S;
+  const int my_kv_end = causal ? (q0w + 32) : S;
+  stage_tile<64, D, 512>(Kb, sks, k_lds[0], threadIdx.x);
+  stage_tile<64, D, 512>(Vb, svs, v_lds[0], threadIdx.x);
+  stage_tile<D, 64, 512>(Ktb, S, kt_lds[0], threadIdx.x);
+  int idx = 0;
+  for (int kv = 0; kv < kv_end; kv += 64, idx ^= 1) {
+    __syncthreads();
+    if (kv + 64 < kv_end) {
+      stage_tile<64, D, 512>(Kb + (int64_t)(kv + 64) * sks, sks,
+                             k_lds[idx ^ 1], threadIdx.x);
+      stage_tile<64, D, 512>(Vb + (int64_t)(kv + 64) * svs, svs,
+                             v_lds[idx ^ 1], threadIdx.x);
+      stage_tile<D, 64, 512>(Ktb + kv + 64, S, kt_lds[idx ^ 1], threadIdx.x);
+    }
+    if (kv >= my_kv_end) continue;
+
+    // ---- St = K·Q^T (pre-scaled), dPt = V·dO^T: C[k][q], q = lo
+    f32x16 st[2], dpt[2];
+    if (FLAGS & 1) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) { st[t][r] = 0.f; dpt[t][r] = 0.f; }
+#pragma unroll
+      for (int ds = 0; ds < DSL; ++ds) {
+        const short8 kf = ld8_swz<D>(k_lds[idx], t * 32 + lo, ds * 32 + hi * 16);
+        const short8 vf = ld8_swz<D>(v_lds[idx], t * 32 + lo, ds * 32 + hi * 16);
+        st[t] = mfma32(kf, qf[ds], st[t]);
+        dpt[t] = mfma32(vf, dof[ds], dpt[t]);
+      }
+    }
+    if (FLAGS & 1) __builtin_amdgcn_s_setprio(0);
+    // ---- dS = P * (dP - delta) * scale, all in-register (q = lo)
+#pragma unroll
+    for (int t = 0; t < 2; ++t)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kg = kv + t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float p = __expf(st[t][r] - lse_q);
+        if (causal && kg > q0w + lo) p = 0.f;
+        st[t][r] = p * (dpt[t][r] - dl_q) * scale;
+      }
+    // ---- dS -> A-fragments (same lane-pair exchange as the fwd P)
+    short8 da[2 * 2];
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+#pragma unroll
+      for (int kst = 0; kst < 2; ++kst) {
+        const int g0 = 4 * (2 * kst), g1 = 4 * (2 * kst + 1);
+        unsigned x0 = cvt_pk_bf16(st[t][g0 + 0], st[t][g0 + 1]);
+        unsigned x1 = cvt_pk_bf16(st[t][g0 + 2], st[t][g0 + 3]);
+        unsigned y0 = cvt_pk_bf16(st[t][g1 + 0], st[t][g1 + 1]);
+        unsigned y1 = cvt_pk_bf16(st[t][g1 + 2], st[t][g1 + 3]);
+        permlane32_swap(x0, y0);
+        permlane32_swap(x1, y1);
+        short8 frag;
+        reinterpret_cast<unsigned*>(&frag)[0] = x0;
+        reinterpret_cast<unsigned*>(&frag)[1] = x1;
+        reinterpret_cast<unsigned*>(&frag)[2] = y0;
+        reinterpret_cast<unsigned*>(&frag)[3] = y1;
+        da[t * 2 + kst] = frag;
+      }
+    }
+    // ---- dQ += dS·K (B-frags from swizzled Kt LDS)
+    if (FLAGS & 1) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int dt = 0; dt < DT32; ++dt)
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks)
+        dq_acc[dt] = mfma32(da[ks],
+                            ld8_swz<64>(kt_lds[idx], dt * 32 + lo, ks * 32 + hi * 16),
+                            dq_acc[dt]);
+    if (FLAGS & 1) __builtin_amdgcn_s_setprio(0);
+  }
+  // ---- epilogue: C[q][d] rows q = crow(r,hi)
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int qr = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    bf16* row = dQ + (((int64_t)(b * S + q0w + qr)) * H + h) * D;
+#pragma unroll
+    for (int dt = 0; dt < DT32; ++dt)
+      row[dt * 32 + lo] = f2bf(dq_acc[dt][r]);
   }
 }
 
@@ -581,8 +752,8 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
 // -------------------------------------------------------- backward dK, dV
 // Stages Q[64][D], dO[64][D] (B-operands for St and dPt), Qt[D][64] and
 // dOt[D][64] (B-operands for dK and dV accumulation).
-template <int D>
-__global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
+template <int D, int NW = 4>  // NW waves x 16 k-rows per block
+__global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ Qt,
     const bf16* __restrict__ K, const bf16* __restrict__ V,
     const bf16* __restrict__ dO, const bf16* __restrict__ dOt,
@@ -594,7 +765,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
     int64_t svb, int64_t svs, int64_t svh) {
   constexpr int DS = D / 32;
   constexpr int DT = D / 16;
-  const int n_kt = S / 64;
+  const int n_kt = S / (NW * 16);
   const int per_split = (B * Hkv) * n_kt;
   const int split = blockIdx.x / per_split;
   const int rem = blockIdx.x - split * per_split;
@@ -605,7 +776,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int lg = lane >> 4, li = lane & 15;
-  const int k0 = kt * 64 + wid * 16;
+  const int k0 = kt * (NW * 16) + wid * 16;
 
   const bf16* Kb = K + b * skb + hkv * skh;
   const bf16* Vb = V + b * svb + hkv * svh;
@@ -614,8 +785,8 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
   __shared__ bf16 do_lds[64 * D];
   __shared__ bf16 qt_lds[D * 64];
   __shared__ bf16 dot_lds[D * 64];
-  __shared__ bf16 pt_lds_all[4][16 * 64];
-  __shared__ bf16 dst_lds_all[4][16 * 64];
+  __shared__ bf16 pt_lds_all[NW][16 * 64];
+  __shared__ bf16 dst_lds_all[NW][16 * 64];
   bf16* pt_lds = pt_lds_all[wid];
   bf16* dst_lds = dst_lds_all[wid];
 
@@ -640,13 +811,13 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
     const bf16* dOtb = dOt + ((int64_t)(b * H + h) * D) * S;
     const float* lse_b = lse + (int64_t)b * S * H + h;
     const float* dl_b = delta + (int64_t)b * S * H + h;
-    const int q_start = (causal ? kt * 64 : 0) + split * 64;
+    const int q_start = (causal ? (kt * (NW * 16)) / 64 * 64 : 0) + split * 64;
     for (int q0g = q_start; q0g < S; q0g += 64 * splits) {
       __syncthreads();
-      stage_tile<64, D>(Qb + (int64_t)q0g * sqs, sqs, q_lds, threadIdx.x);
-      stage_tile<64, D>(dOb + (int64_t)q0g * H * D, (int64_t)H * D, do_lds, threadIdx.x);
-      stage_tile<D, 64>(Qtb + q0g, S, qt_lds, threadIdx.x);
-      stage_tile<D, 64>(dOtb + q0g, S, dot_lds, threadIdx.x);
+      stage_tile<64, D, NW * 64>(Qb + (int64_t)q0g * sqs, sqs, q_lds, threadIdx.x);
+      stage_tile<64, D, NW * 64>(dOb + (int64_t)q0g * H * D, (int64_t)H * D, do_lds, threadIdx.x);
+      stage_tile<D, 64, NW * 64>(Qtb + q0g, S, qt_lds, threadIdx.x);
+      stage_tile<D, 64, NW * 64>(dOtb + q0g, S, dot_lds, threadIdx.x);
       __syncthreads();
 #pragma unroll
       for (int sub = 0; sub < 4; ++sub) {
@@ -740,16 +911,27 @@ PRIME_API int prime_flash_fwd(hipStream_t stream, const void* Q, const void* K,
                               int64_t sqb, int64_t sqs, int64_t sqh,
                               int64_t skb, int64_t sks, int64_t skh) {
   if (S % 64 != 0 || (D != 64 && D != 128)) return hipErrorInvalidValue;
-  static const char* v4env = getenv("PRIME_ATTN_V4");
-  // v4 measured slower than v2 at 10B shapes (130 vs 202 TF/s: 235
-  // VGPRs -> 2 waves/SIMD starves latency hiding); opt-in via PRIME_ATTN_V4=1
-  const bool use_v4 = (D == 128) && (S % 128 == 0) && (v4env && v4env[0] == '1');
-  if (use_v4) {
-    const int grid = (int)(B * H * (S / 128));
-    hipLaunchKernelGGL(flash_fwd32_kernel<128>, dim3(grid), dim3(256), 0,
-                       stream, (const bf16*)Q, (const bf16*)K, (const bf16*)Vt,
-                       (bf16*)O, (float*)lse, (int)B, (int)H, (int)Hkv, (int)S,
-                       (float)scale, (int)causal, sqb, sqs, sqh, skb, sks, skh);
+  static const char* v5env = getenv("PRIME_ATTN_V5");
+  static const char* flagenv = getenv("PRIME_ATTN_FLAGS");
+  // measured on MI355X (15-iter sweep): FLAGS=2 417 TF > 3 (406) > 0 (387)
+  // > 1 (382) — defer-max pays, setprio slightly negative here
+  const int flags = flagenv ? atoi(flagenv) : 2;
+  const bool use_v5 =
+      (D == 128) && (S % 256 == 0) && !(v5env && v5env[0] == '0');
+  if (use_v5) {
+    const int grid = (int)(B * H * (S / 256));
+#define LAUNCH_V5(F)                                                         \
+    hipLaunchKernelGGL((flash_fwd_v5_kernel<128, F>), dim3(grid), dim3(512), \
+                       0, stream, (const bf16*)Q, (const bf16*)K,            \
+                       (const bf16*)Vt, (bf16*)O, (float*)lse, (int)B,       \
+                       (int)H, (int)Hkv, (int)S, (float)scale, (int)causal,  \
+                       sqb, sqs, sqh, skb, sks, skh)
+    switch (flags & 3) {
+      case 0: LAUNCH_V5(0); break;
+      case 1: LAUNCH_V5(1); break;
+      case 2: LAUNCH_V5(2); break;
+      default: LAUNCH_V5(3); break;
+    }
     return (int)hipGetLastError();
   }
 #define LAUNCH_FWD(DD, NW)                                                  \
@@ -788,6 +970,22 @@ PRIME_API int prime_flash_bwd_dq(hipStream_t stream, const void* Q,
                                  int64_t skb, int64_t sks, int64_t skh,
                                  int64_t svb, int64_t svs, int64_t svh) {
   if (S % 64 != 0 || (D != 64 && D != 128)) return hipErrorInvalidValue;
+  static const char* v5env = getenv("PRIME_ATTN_V5");
+  static const char* flagenv = getenv("PRIME_ATTN_FLAGS");
+  const int flags = flagenv ? atoi(flagenv) : 2;
+  if ((D == 128) && (S % 256 == 0) && !(v5env && v5env[0] == '0')) {
+    const int grid = (int)(B * H * (S / 256));
+#define LAUNCH_DQ5(F)                                                       \
+    hipLaunchKernelGGL((flash_bwd_dq_v5_kernel<128, F>), dim3(grid),        \
+                       dim3(512), 0, stream, (const bf16*)Q,                \
+                       (const bf16*)K, (const bf16*)V, (const bf16*)Kt,     \
+                       (const bf16*)dO, (const float*)lse,                  \
+                       (const float*)delta, (bf16*)dQ, (int)B, (int)H,      \
+                       (int)Hkv, (int)S, (float)scale, (int)causal,         \
+                       sqb, sqs, sqh, skb, sks, skh, svb, svs, svh)
+    if (flags & 1) LAUNCH_DQ5(1); else LAUNCH_DQ5(0);
+    return (int)hipGetLastError();
+  }
   const int grid = (int)(B * H * (S / 64));
   DISPATCH_D(flash_bwd_dq_kernel, (const bf16*)Q, (const bf16*)K,
              (const bf16*)V, (const bf16*)Kt, (const bf16*)dO,
@@ -809,12 +1007,26 @@ PRIME_API int prime_flash_bwd_dkv(hipStream_t stream, const void* Q,
                                   int64_t skb, int64_t sks, int64_t skh,
                                   int64_t svb, int64_t svs, int64_t svh) {
   if (S % 64 != 0 || (D != 64 && D != 128)) return hipErrorInvalidValue;
-  const int grid = (int)(B * Hkv * (S / 64) * splits);
-  DISPATCH_D(flash_bwd_dkv_kernel, (const bf16*)Q, (const bf16*)Qt,
-             (const bf16*)K, (const bf16*)V, (const bf16*)dO, (const bf16*)dOt,
-             (const float*)lse, (const float*)delta, (float*)wsK, (float*)wsV,
-             (int)B, (int)H, (int)Hkv, (int)S, (float)scale, (int)causal,
-             (int)splits, sqb, sqs, sqh, skb, sks, skh, svb, svs, svh);
+  // 8-wave blocks: one q/do/qt/dot stage feeds 128 k-rows (half the
+  // staging traffic of the 4-wave version); fall back to 4 waves when the
+  // sequence doesn't tile by 128
+  static const char* nw8env = getenv("PRIME_ATTN_DKV8");
+  const bool nw8 = (S % 128 == 0) && !(nw8env && nw8env[0] == '0');
+#define LAUNCH_DKV(DD, NWV)                                                  \
+    hipLaunchKernelGGL((flash_bwd_dkv_kernel<DD, NWV>),                      \
+                       dim3((int)(B * Hkv * (S / (NWV * 16)) * splits)),     \
+                       dim3(NWV * 64), 0, stream, (const bf16*)Q,            \
+                       (const bf16*)Qt, (const bf16*)K, (const bf16*)V,      \
+                       (const bf16*)dO, (const bf16*)dOt,                    \
+                       (const float*)lse, (const float*)delta, (float*)wsK,  \
+                       (float*)wsV, (int)B, (int)H, (int)Hkv, (int)S,        \
+                       (float)scale, (int)causal, (int)splits, sqb, sqs,     \
+                       sqh, skb, sks, skh, svb, svs, svh)
+  if (nw8) {
+    if (D == 128) LAUNCH_DKV(128, 8); else LAUNCH_DKV(64, 8);
+  } else {
+    if (D == 128) LAUNCH_DKV(128, 4); else LAUNCH_DKV(64, 4);
+  }
   int err = hipGetLastError();
   if (err) return err;
   int rgrid = prime_grid(B * Hkv * S * D, 256);
