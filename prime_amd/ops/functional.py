@@ -297,15 +297,16 @@ def flash_attention(q, k, v, causal: bool = True) -> torch.Tensor:
 # --------------------------------------------------------- cross entropy
 class _CrossEntropy(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, logits, targets, ignore_index: int):
-        # logits [R, V] bf16, targets [R] int32/int64
+    def forward(ctx, logits, targets, ignore_index: int, need_grad: bool):
+        # logits [R, V] bf16, targets [R] int32/int64. need_grad is decided
+        # by the caller: Function.forward always runs under no_grad, so
+        # torch.is_grad_enabled() here would be False even in training.
         R, V = logits.shape
         logits = logits.contiguous()
         t32 = targets.to(torch.int32).contiguous()
         loss = torch.empty(R, device=logits.device, dtype=torch.float32)
         # under no_grad (eval/perplexity) skip the [R,V] dlogits write —
         # at V=128k that is a full extra bf16 tensor per eval batch
-        need_grad = torch.is_grad_enabled() and logits.requires_grad
         dlogits = torch.empty_like(logits) if need_grad else logits  # dummy ptr
         check(
             lib().prime_cross_entropy(
@@ -326,14 +327,15 @@ class _CrossEntropy(torch.autograd.Function):
     def backward(ctx, grad_out):
         dlogits, n_valid = ctx.saved_tensors
         g = dlogits * (grad_out.float() / n_valid.float()).to(dlogits.dtype)
-        return g, None, None
+        return g, None, None, None
 
 
 def cross_entropy(logits, targets, ignore_index: int = -100) -> torch.Tensor:
     """Fused CE: mean loss over non-ignored tokens; grad computed in-kernel."""
     if not _is_hip(logits):
         return ref.cross_entropy(logits, targets, ignore_index)
-    return _CrossEntropy.apply(logits, targets, ignore_index)
+    need_grad = torch.is_grad_enabled() and logits.requires_grad
+    return _CrossEntropy.apply(logits, targets, ignore_index, need_grad)
 
 
 # ------------------------------------------------- raw (non-autograd) ops
