@@ -22,6 +22,16 @@ from .. import ops
 from .configs import LlamaConfig, get_config
 
 
+class PrimeLinear(nn.Linear):
+    """nn.Linear (no bias) routed through layout-tuned hipBLASLt calls
+    (ops.tuned_linear): NT forward, NT dX against a per-step cached W^T,
+    NN dW via an LDS-tiled dY transpose. State-dict compatible with
+    nn.Linear."""
+
+    def forward(self, x):
+        return ops.tuned_linear(x, self.weight)
+
+
 class RMSNorm(nn.Module):
     def __init__(self, dim: int, eps: float):
         super().__init__()
@@ -37,8 +47,8 @@ class Attention(nn.Module):
         super().__init__()
         self.cfg = cfg
         hd = cfg.head_dim
-        self.wqkv = nn.Linear(cfg.dim, (cfg.n_heads + 2 * cfg.n_kv_heads) * hd, bias=False)
-        self.wo = nn.Linear(cfg.n_heads * hd, cfg.dim, bias=False)
+        self.wqkv = PrimeLinear(cfg.dim, (cfg.n_heads + 2 * cfg.n_kv_heads) * hd, bias=False)
+        self.wo = PrimeLinear(cfg.n_heads * hd, cfg.dim, bias=False)
 
     def forward(self, x, cos, sin, cache=None, pos: int = 0, pos_dev=None):
         B, S, _ = x.shape
@@ -77,8 +87,8 @@ class Attention(nn.Module):
 class MLP(nn.Module):
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
-        self.w_gateup = nn.Linear(cfg.dim, 2 * cfg.intermediate, bias=False)
-        self.w_down = nn.Linear(cfg.intermediate, cfg.dim, bias=False)
+        self.w_gateup = PrimeLinear(cfg.dim, 2 * cfg.intermediate, bias=False)
+        self.w_down = PrimeLinear(cfg.intermediate, cfg.dim, bias=False)
 
     def forward(self, x):
         return self.w_down(ops.swiglu(self.w_gateup(x)))
@@ -114,7 +124,7 @@ class Llama(nn.Module):
         self.tok_embeddings = nn.Embedding(cfg.vocab_size, cfg.dim)
         self.layers = nn.ModuleList(Block(cfg) for _ in range(cfg.n_layers))
         self.norm = RMSNorm(cfg.dim, cfg.norm_eps)
-        self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+        self.lm_head = PrimeLinear(cfg.dim, cfg.vocab_size, bias=False)
         if cfg.tie_embeddings:
             self.lm_head.weight = self.tok_embeddings.weight
         cos, sin = ops.reference.rope_tables(cfg.head_dim, cfg.max_seq, cfg.rope_theta)

@@ -7,6 +7,9 @@ torch.nn.functional.linear (library GEMMs; everything fused is ours).
 """
 from .functional import (
     apply_rope,
+    invalidate_wt_cache,
+    set_linear_tuned,
+    tuned_linear,
     attn_decode,
     fused_add_rmsnorm,
     cross_entropy,

@@ -152,6 +152,7 @@ class DilocoOptimizer:
                 self._outer_step_resident()
             else:
                 self._outer_step_streamed()
+        ops.invalidate_wt_cache()  # nesterov_outer rewrote flat_w in place
         self.outer_step_count += 1
 
     def _elastic_boundary(self) -> bool:

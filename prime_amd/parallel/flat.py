@@ -96,6 +96,7 @@ class FlatParamSpace:
     def load_flat_(self, master32: torch.Tensor) -> None:
         self.master32.copy_(master32)
         self.flat_w.copy_(self.master32.to(self.dtype))
+        ops.invalidate_wt_cache()
 
 
 class FusedAdamW:
@@ -114,6 +115,7 @@ class FusedAdamW:
 
     def step(self, gscale=None) -> None:
         self.step_count += 1
+        ops.invalidate_wt_cache()  # in-place weight update (raw kernel)
         f = self.flat
         if f.device.type == "cuda":
             ops.fused_adamw(

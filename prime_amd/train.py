@@ -103,6 +103,12 @@ class Trainer:
                 dist.broadcast(self.flat.flat_w, src=0)
                 self.flat.master32.copy_(self.flat.flat_w.float())
 
+        # layout-tuned linear backward (cached W^T) needs stable param
+        # storage: FSDP re-materializes block weights into rotating pool
+        # buffers, so only the non-sharded path may cache by tensor id
+        from . import ops as _ops
+
+        _ops.set_linear_tuned(self.device.type == "cuda" and not self.fsdp)
         self.inner = FusedAdamW(
             self.flat, lr=cfg.optim.lr, betas=tuple(cfg.optim.betas),
             eps=cfg.optim.eps, weight_decay=cfg.optim.weight_decay,

@@ -63,5 +63,64 @@ def main():
     bench_mm("lm_head dX", gl, wl)
 
 
-if __name__ == "__main__":
+if __name__ == "__main__" and "--custom" not in sys.argv:
     main()
+
+
+def bench_custom_nt(name, m, n, k, iters=20):
+    """Custom gemm_nt kernel vs hipBLASLt on the same NT shape."""
+    from prime_amd.ops._lib import check, lib, ptr, stream_of
+
+    a = torch.randn(m, k, device="cuda:0").bfloat16()
+    w = torch.randn(n, k, device="cuda:0").bfloat16()
+    c = torch.empty(m, n, device="cuda:0", dtype=torch.bfloat16)
+
+    def custom():
+        check(lib().prime_gemm_nt(stream_of(a), ptr(a), ptr(w), ptr(c),
+                                  m, n, k, 0), "gemm_nt")
+
+    # refcheck vs library
+    custom()
+    torch.cuda.synchronize()
+    refc = a @ w.T
+    rel = (c.float() - refc.float()).abs().max() / refc.float().abs().max()
+    custom()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        custom()
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / iters
+    tf = 2 * m * k * n / dt / 1e12
+
+    def libmm():
+        torch.matmul(a, w.T, out=c)
+    libmm()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        libmm()
+    torch.cuda.synchronize()
+    dtl = (time.perf_counter() - t0) / iters
+    tfl = 2 * m * k * n / dtl / 1e12
+    print(f"{name:24s} M={m:6d} N={n:6d} K={k:6d}  custom {tf:7.0f} TF/s"
+          f"  lib {tfl:7.0f} TF/s  ratio {tf/tfl:5.2f}  relerr {rel:.2e}")
+
+
+def main_custom():
+    torch.manual_seed(1)
+    M, dim, inter, vocab = 16384, 4096, 14336, 128256
+    print("== custom NT kernel vs hipBLASLt ==")
+    bench_custom_nt("qkv fwd", M, 6144, dim)
+    bench_custom_nt("gateup fwd", M, 2 * inter, dim)
+    bench_custom_nt("down fwd", M, dim, inter)
+    bench_custom_nt("o fwd", M, dim, dim)
+    bench_custom_nt("lm_head fwd", M, vocab, dim)
+    bench_custom_nt("dX-as-NT (qkv)", M, dim, 6144)
+    bench_custom_nt("dW-as-NT (qkv)", 6144, dim, M)
+    bench_custom_nt("square 4k", 4096, 4096, 4096)
+
+
+if __name__ == "__main__" and "--custom" in sys.argv:
+    main_custom()
+    sys.exit(0)
