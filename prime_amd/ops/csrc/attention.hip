@@ -749,6 +749,184 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
   }
 }
 
+// ---------------------------------------------- backward dK/dV v5 (32x32)
+// Swapped the other way around from the fwd/dq kernels: St^T = Q·K^T puts
+// each lane's scores at a FIXED key column (k = lane&31, the dimension
+// this kernel accumulates over), so P^T/dS^T feed the dV/dK accumulation
+// A-fragments after the same cvt_pk + permlane32_swap lane-pair exchange
+// — no per-wave P/dS LDS bounce. K/V rows live in registers (per-wave
+// constants), Q/dO rows are read per-lane from global (L1-resident across
+// the d-slice loop), and only the genuinely-transposed operands (Qt/dOt
+// tiles for the accumulation B-fragments) are staged, double-buffered.
+// 8 waves x 32 k-rows = 256-row k tiles; fp32 split workspaces + reduce
+// as in the 16x16 kernel.
+template <int D, int FLAGS>
+__global__ __launch_bounds__(512, 2) void flash_bwd_dkv_v5_kernel(
+    const bf16* __restrict__ Q, const bf16* __restrict__ Qt,
+    const bf16* __restrict__ K, const bf16* __restrict__ V,
+    const bf16* __restrict__ dO, const bf16* __restrict__ dOt,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    float* __restrict__ wsK, float* __restrict__ wsV, int B, int H, int Hkv,
+    int S, float scale, int causal, int splits,
+    int64_t sqb, int64_t sqs, int64_t sqh,
+    int64_t skb, int64_t sks, int64_t skh,
+    int64_t svb, int64_t svs, int64_t svh) {
+  constexpr int DSL = D / 16;
+  constexpr int DT32 = D / 32;
+  const int n_kt = S / 256;
+  const int per_split = (B * Hkv) * n_kt;
+  const int split = blockIdx.x / per_split;
+  const int rem = blockIdx.x - split * per_split;
+  const int bh = rem / n_kt;
+  const int kt = rem - bh * n_kt;
+  const int b = bh / Hkv, hkv = bh - b * Hkv;
+  const int group = H / Hkv;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lo = lane & 31, hi = lane >> 5;
+  const int k0w = kt * 256 + wid * 32;  // this wave's first k row (global)
+
+  const bf16* Kb = K + b * skb + hkv * skh;
+  const bf16* Vb = V + b * svb + hkv * svh;
+
+  __shared__ bf16 q_lds[2][64 * D];
+  __shared__ bf16 do_lds[2][64 * D];
+  __shared__ bf16 qt_lds[2][D * 64];
+  __shared__ bf16 dot_lds[2][D * 64];
+
+  // K^T / V^T B-fragment source rows (k0w+lo): loaded inline per MFMA —
+  // holding all 8 slices of both in registers (64 VGPRs) pushed the
+  // kernel to 256 regs + scratch spills; the rows stay L1-resident
+  // across the q loop instead
+  const bf16* krow = Kb + (int64_t)(k0w + lo) * sks;
+  const bf16* vrow = Vb + (int64_t)(k0w + lo) * svs;
+  f32x16 dk_acc[DT32], dv_acc[DT32];
+#pragma unroll
+  for (int dt = 0; dt < DT32; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) { dk_acc[dt][r] = 0.f; dv_acc[dt][r] = 0.f; }
+
+  // flattened (g, q-tile) loop for double-buffered Qt/dOt prefetch
+  const int q_start = (causal ? kt * 256 : 0) + split * 64;
+  const int tiles_per_g = (S - q_start + 64 * splits - 1) / (64 * splits);
+  const int n_iter = group * tiles_per_g;
+  const auto stage_for = [&](int it, int buf) {
+    const int g = it / tiles_per_g;
+    const int q0g = q_start + (it - g * tiles_per_g) * 64 * splits;
+    const int h = hkv * group + g;
+    const bf16* Qb_ = Q + b * sqb + h * sqh;
+    const bf16* dOb_ = dO + ((int64_t)b * S * H + h) * D;
+    const bf16* Qtb = Qt + ((int64_t)(b * H + h) * D) * S;
+    const bf16* dOtb = dOt + ((int64_t)(b * H + h) * D) * S;
+    stage_tile<64, D, 512>(Qb_ + (int64_t)q0g * sqs, sqs, q_lds[buf], threadIdx.x);
+    stage_tile<64, D, 512>(dOb_ + (int64_t)q0g * H * D, (int64_t)H * D, do_lds[buf], threadIdx.x);
+    stage_tile<D, 64, 512>(Qtb + q0g, S, qt_lds[buf], threadIdx.x);
+    stage_tile<D, 64, 512>(dOtb + q0g, S, dot_lds[buf], threadIdx.x);
+  };
+  stage_for(0, 0);
+  for (int it = 0; it < n_iter; ++it) {
+    const int cur = it & 1;
+    __syncthreads();
+    if (it + 1 < n_iter)
+      stage_for(it + 1, cur ^ 1);
+    const int g = it / tiles_per_g;
+    const int q0g = q_start + (it - g * tiles_per_g) * 64 * splits;
+    if (causal && q0g + 64 <= k0w) continue;  // fully masked for this wave
+    const int h = hkv * group + g;
+    const float* lse_b = lse + (int64_t)b * S * H + h;
+    const float* dl_b = delta + (int64_t)b * S * H + h;
+
+#pragma unroll
+    for (int qs = 0; qs < 2; ++qs) {
+      const int q32 = q0g + qs * 32;
+      // St^T = Q·K^T, dPt^T = dO·V^T: C[32q][32k], k = lo (lane-local)
+      f32x16 st, dpt;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) { st[r] = 0.f; dpt[r] = 0.f; }
+      if (FLAGS & 1) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ds = 0; ds < DSL; ++ds) {
+        const short8 qfr = ld8_swz<D>(q_lds[cur], qs * 32 + lo, ds * 32 + hi * 16);
+        st = mfma32(qfr, ld8(krow + ds * 16 + hi * 8), st);
+      }
+#pragma unroll
+      for (int ds = 0; ds < DSL; ++ds) {
+        const short8 dofr = ld8_swz<D>(do_lds[cur], qs * 32 + lo, ds * 32 + hi * 16);
+        dpt = mfma32(dofr, ld8(vrow + ds * 16 + hi * 8), dpt);
+      }
+      if (FLAGS & 1) __builtin_amdgcn_s_setprio(0);
+      // P^T / dS^T in-register; lse/delta per q-row crow(r,hi) are
+      // broadcast loads (same address across the 32 lo-lanes)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qg = q32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const float lse_q = lse_b[(int64_t)qg * H];
+        const float dl_q = dl_b[(int64_t)qg * H];
+        float p = __expf(st[r] * scale - lse_q);
+        if (causal && (k0w + lo) > qg) p = 0.f;
+        st[r] = p;
+        dpt[r] = p * (dpt[r] - dl_q) * scale;
+      }
+      // pack P^T / dS^T into accumulation A-fragments: A[32k][16q], the
+      // q-slices split across the lane pair exactly like the fwd P
+      short8 pa[2], da[2];
+#pragma unroll
+      for (int kst = 0; kst < 2; ++kst) {
+        const int g0 = 4 * (2 * kst), g1 = 4 * (2 * kst + 1);
+        unsigned x0 = cvt_pk_bf16(st[g0 + 0], st[g0 + 1]);
+        unsigned x1 = cvt_pk_bf16(st[g0 + 2], st[g0 + 3]);
+        unsigned y0 = cvt_pk_bf16(st[g1 + 0], st[g1 + 1]);
+        unsigned y1 = cvt_pk_bf16(st[g1 + 2], st[g1 + 3]);
+        permlane32_swap(x0, y0);
+        permlane32_swap(x1, y1);
+        short8 f;
+        reinterpret_cast<unsigned*>(&f)[0] = x0;
+        reinterpret_cast<unsigned*>(&f)[1] = x1;
+        reinterpret_cast<unsigned*>(&f)[2] = y0;
+        reinterpret_cast<unsigned*>(&f)[3] = y1;
+        pa[kst] = f;
+        x0 = cvt_pk_bf16(dpt[g0 + 0], dpt[g0 + 1]);
+        x1 = cvt_pk_bf16(dpt[g0 + 2], dpt[g0 + 3]);
+        y0 = cvt_pk_bf16(dpt[g1 + 0], dpt[g1 + 1]);
+        y1 = cvt_pk_bf16(dpt[g1 + 2], dpt[g1 + 3]);
+        permlane32_swap(x0, y0);
+        permlane32_swap(x1, y1);
+        reinterpret_cast<unsigned*>(&f)[0] = x0;
+        reinterpret_cast<unsigned*>(&f)[1] = x1;
+        reinterpret_cast<unsigned*>(&f)[2] = y0;
+        reinterpret_cast<unsigned*>(&f)[3] = y1;
+        da[kst] = f;
+      }
+      // dV += P^T·dO, dK += dS^T·Q (B-frags from swizzled dOt/Qt tiles)
+      if (FLAGS & 1) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int dt = 0; dt < DT32; ++dt)
+#pragma unroll
+        for (int kst = 0; kst < 2; ++kst) {
+          const int colb = qs * 64 + kst * 32 + hi * 16;
+          dv_acc[dt] = mfma32(pa[kst],
+                              ld8_swz<64>(dot_lds[cur], dt * 32 + lo, colb),
+                              dv_acc[dt]);
+          dk_acc[dt] = mfma32(da[kst],
+                              ld8_swz<64>(qt_lds[cur], dt * 32 + lo, colb),
+                              dk_acc[dt]);
+        }
+      if (FLAGS & 1) __builtin_amdgcn_s_setprio(0);
+    }
+  }
+  // fp32 partials: ws[split][b][hkv][kg][d]; C[32k][32d] rows crow(r,hi)
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int kg = k0w + (r & 3) + 8 * (r >> 2) + 4 * hi;
+    const int64_t base = ((((int64_t)split * B + b) * Hkv + hkv) * S + kg) * D;
+#pragma unroll
+    for (int dt = 0; dt < DT32; ++dt) {
+      wsK[base + dt * 32 + lo] = dk_acc[dt][r];
+      wsV[base + dt * 32 + lo] = dv_acc[dt][r];
+    }
+  }
+}
+
 // -------------------------------------------------------- backward dK, dV
 // Stages Q[64][D], dO[64][D] (B-operands for St and dPt), Qt[D][64] and
 // dOt[D][64] (B-operands for dK and dV accumulation).
@@ -1010,6 +1188,35 @@ PRIME_API int prime_flash_bwd_dkv(hipStream_t stream, const void* Q,
   // 8-wave blocks: one q/do/qt/dot stage feeds 128 k-rows (half the
   // staging traffic of the 4-wave version); fall back to 4 waves when the
   // sequence doesn't tile by 128
+  // 32x32 swapped dkv measured SLOWER than the 16x16 split-q kernel
+  // (160 vs 228 TF: the loop-invariant K/V row fragments don't fit the
+  // register budget, and demoting them to inline L2 loads puts ~16 global
+  // loads in every q-step's MFMA dependency chain). Opt-in for further
+  // work via PRIME_ATTN_DKV5=1; see profiles/10b_1gpu_profile.md.
+  static const char* v5denv = getenv("PRIME_ATTN_DKV5");
+  static const char* fenv = getenv("PRIME_ATTN_FLAGS");
+  const int flags = fenv ? atoi(fenv) : 2;
+  if ((D == 128) && (S % 256 == 0) && (v5denv && v5denv[0] == '1')) {
+    const int sp = (int)(splits > 4 ? 4 : splits);
+    const int grid = (int)(B * Hkv * (S / 256) * sp);
+#define LAUNCH_DKV5(F)                                                       \
+    hipLaunchKernelGGL((flash_bwd_dkv_v5_kernel<128, F>), dim3(grid),        \
+                       dim3(512), 0, stream, (const bf16*)Q,                 \
+                       (const bf16*)Qt, (const bf16*)K, (const bf16*)V,      \
+                       (const bf16*)dO, (const bf16*)dOt,                    \
+                       (const float*)lse, (const float*)delta, (float*)wsK,  \
+                       (float*)wsV, (int)B, (int)H, (int)Hkv, (int)S,        \
+                       (float)scale, (int)causal, sp, sqb, sqs, sqh,         \
+                       skb, sks, skh, svb, svs, svh)
+    if (flags & 1) LAUNCH_DKV5(1); else LAUNCH_DKV5(0);
+    int err5 = hipGetLastError();
+    if (err5) return err5;
+    int rgrid5 = prime_grid(B * Hkv * S * D, 256);
+    hipLaunchKernelGGL(dkv_reduce_kernel, dim3(rgrid5), dim3(256), 0, stream,
+                       (const float*)wsK, (const float*)wsV, (bf16*)dK,
+                       (bf16*)dV, (int)B, (int)Hkv, (int)S, (int)D, sp);
+    return (int)hipGetLastError();
+  }
   static const char* nw8env = getenv("PRIME_ATTN_DKV8");
   const bool nw8 = (S % 128 == 0) && !(nw8env && nw8env[0] == '0');
 #define LAUNCH_DKV(DD, NWV)                                                  \

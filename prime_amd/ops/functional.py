@@ -401,7 +401,8 @@ class _TunedLinear(torch.autograd.Function):
         else:
             dx = torch.matmul(dy2, w)
         # dW as NN: transpose dY once (LDS-tiled kernel), then plain NN.
-        # Measured net-positive for every 10B shape except qkv (-1%).
+        # Measured net +3-6% on the 10B shapes (qkv alone is -1%; not
+        # worth a per-shape table)
         if N % 128 == 0 and M % 64 == 0 and M >= 4096:
             dyt = transpose_bshd(dy2.view(1, M, N // 128, 128)).view(N, M)
             dw = torch.mm(dyt, x2)
