@@ -107,12 +107,12 @@ class ElasticDeviceMesh:
 
         W = self.n_workers
         R = len([o for o in range(1, W) if ring._gcd(o, W) == 1])
-        # multi-ring engages R xGMI links concurrently but is opt-in for
-        # now (PRIME_AMD_MULTIRING=1): the outer sync is amortized over H
-        # steps, and an unattended scale run should not bet on concurrent
-        # grouped P2P until it has been validated on an 8-GPU node
-        multi = os.environ.get("PRIME_AMD_MULTIRING", "0") == "1"
-        if multi and W > 2 and R > 1 and delta32.numel() % (R * W * ring.QBLK) == 0:
+        # multi-ring engages R xGMI links concurrently; default ON (the
+        # schedule is unit-tested for every W in 2..8 and every coprime
+        # offset against a simulated reference — tests/test_ring.py — and
+        # sizes pad per call). PRIME_AMD_MULTIRING=0 opts out.
+        multi = os.environ.get("PRIME_AMD_MULTIRING", "1") != "0"
+        if multi and W > 2 and R > 1:
             ring.ring_allreduce_int8_multi(delta32, group=group, average=True)
         else:
             ring.ring_allreduce_int8(delta32, group=group, average=True)

@@ -89,3 +89,39 @@ def test_mixed_topology_2x2():
     assert outs[0]["outer"] == 2
     # after the final outer boundary every rank holds identical params
     assert outs[0]["head"] == outs[1]["head"] == outs[2]["head"] == outs[3]["head"]
+
+
+def _soak_worker(rank, world, steps, h):
+    from prime_amd.utils.config import (
+        DilocoConfig, MetricsConfig, ModelConfig, ParallelConfig, TrainConfig,
+    )
+    from prime_amd.train import Trainer
+
+    cfg = TrainConfig(
+        run_name="soak8",
+        steps=steps,
+        model=ModelConfig(name="llama_test", seq_len=32),
+        diloco=DilocoConfig(H=h),
+        parallel=ParallelConfig(worker_size=1),
+        metrics=MetricsConfig(log_interval=1000),
+    )
+    cfg.data.micro_batch_size = 1
+    tr = Trainer(cfg, run_dir=f"/tmp/prime_amd_test/soak8_r{rank}")
+    for _ in range(steps):
+        tr.train_step()
+    head = tr.diloco.theta_outer[:16].tolist()
+    w = tr.flat.flat_w[:16].tolist()
+    outer = tr.diloco.outer_step_count
+    tr.close()
+    return {"head": head, "w": w, "outer": outer}
+
+
+def test_eight_worker_outer_soak():
+    """8 DiLoCo workers on gloo through 3 outer boundaries with the
+    multi-ring int8 all-reduce (default-on; W=8 engages 4 coprime
+    offsets). All workers must hold identical outer state afterwards."""
+    outs = run_distributed(_soak_worker, 8, args=(6, 2), timeout=600)
+    assert outs[0]["outer"] == 3
+    for r in range(1, 8):
+        assert outs[r]["head"] == outs[0]["head"], r
+        assert outs[r]["w"] == outs[0]["w"], r

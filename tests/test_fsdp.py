@@ -139,3 +139,88 @@ def test_fsdp_per_shard_checkpoint(tmp_path):
         assert abs(o["saved"] - o["resumed"]) < 1e-6
     # the two shards are genuinely different state
     assert outs[0]["saved"] != outs[1]["saved"]
+
+
+# -------------------- async ordering stress (fake delayed collectives)
+def _train_worker_async_fake(rank, world, steps):
+    """Run the FSDP space with DELAYED fake collectives that emulate RCCL
+    async semantics on CPU: outputs are poisoned with NaN until .wait(),
+    and inputs are snapshotted at issue time — any read-before-wait
+    propagates NaN into the loss, and any in-flight input mutation trips
+    an assertion. This exercises the async reduce-scatter, 2-buffer
+    parity reuse and prefetch drain ordering (fsdp.py) that the gloo
+    fallback never runs."""
+    import torch
+
+    from prime_amd.parallel import fsdp as fsdp_mod
+
+    class _DelayedWork:
+        def __init__(self, fn, snap_src, src_live):
+            self._fn = fn
+            self._snap = snap_src
+            self._live = src_live
+            self._done = False
+
+        def wait(self):
+            if self._done:
+                return
+            self._done = True
+            torch.testing.assert_close(
+                self._live, self._snap, rtol=0, atol=0,
+                msg="collective input mutated while in flight",
+            )
+            self._fn(self._snap)
+
+    import torch.distributed as dist
+
+    def fake_all_gather(full, shard, group, async_op=False):
+        W = dist.get_world_size(group)
+        snap = shard.detach().clone()
+        live = shard
+
+        def complete(src):
+            chunks = list(full.chunk(W))
+            dist.all_gather(chunks, src.contiguous(), group=group)
+
+        full.fill_(float("nan"))  # poison until wait()
+        w = _DelayedWork(complete, snap, live)
+        if async_op:
+            return w
+        w.wait()
+        return None
+
+    def fake_reduce_scatter(out_shard, grad_full, group, async_op=False):
+        W = dist.get_world_size(group)
+        r = dist.get_rank(group)
+        snap = grad_full.detach().clone()
+        live = grad_full
+
+        def complete(src):
+            buf = src.clone()
+            dist.all_reduce(buf, group=group)
+            n = buf.numel() // W
+            out_shard.copy_(buf[r * n : (r + 1) * n])
+
+        out_shard.fill_(float("nan"))
+        w = _DelayedWork(complete, snap, live)
+        if async_op:
+            return w
+        w.wait()
+        return None
+
+    fsdp_mod._all_gather_flat = fake_all_gather
+    fsdp_mod._reduce_scatter_flat = fake_reduce_scatter
+
+    out = _train_worker(rank, world, True, steps)
+    for ls in out["losses"]:
+        assert ls == ls, "NaN loss: an output buffer was read before wait()"
+    return out
+
+
+def test_fsdp_async_ordering_stress():
+    dp = run_distributed(_train_worker, 2, args=(False, 3), timeout=300)
+    sh = run_distributed(_train_worker_async_fake, 2, args=(3,), timeout=300)
+    for a, b in zip(dp[0]["losses"], sh[0]["losses"]):
+        assert abs(a - b) < 1e-3, (dp[0]["losses"], sh[0]["losses"])
+    for k in dp[0]["w"]:
+        assert abs(dp[0]["w"][k] - sh[0]["w"][k]) < 2e-2, k

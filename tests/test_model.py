@@ -156,3 +156,89 @@ def test_evaluate_perplexity_token_file(tmp_path):
     )
     assert res["tokens"] == 3 * 2 * 64
     assert 1 < res["perplexity"] < 1e4
+
+
+def _build_big_cheap(name, **over):
+    """Construct a big-dim model without paying two full random-init
+    passes (kaiming + normal over billions of params): meta-device
+    construction + constant fill. Layout/shape tests only."""
+    import torch
+
+    from prime_amd.models import build_model
+
+    with torch.device("meta"):
+        m = build_model(name, **over)
+    m = m.to_empty(device="cpu")
+    with torch.no_grad():
+        for prm in m.parameters():
+            prm.fill_(0.01)
+    m.reset_rope(torch.device("cpu"))
+    return m
+
+
+def test_70b_dims_block_shapes_cpu():
+    """Full-dim Llama-70B block (dim 8192, 64 heads, GQA 8, inter 28672)
+    at reduced layer count: fwd/bwd/optimizer shapes must hold. The full
+    80-layer config is exercised on GPU via tools/sizing_70b.py."""
+    import torch
+
+    from prime_amd.models import build_model
+    from prime_amd.parallel.flat import FlatParamSpace, FusedAdamW
+
+    m = _build_big_cheap("llama_70b", n_layers=1, vocab_size=512, max_seq=128)
+    assert m.cfg.dim == 8192 and m.cfg.intermediate == 28672
+    assert m.cfg.n_heads == 64 and m.cfg.n_kv_heads == 8
+    flat = FlatParamSpace(m)
+    x = torch.randint(0, 512, (1, 32))
+    y = torch.randint(0, 512, (1, 32))
+    loss = m.loss(x, y)
+    flat.zero_grad()
+    loss.backward()
+    assert float(loss) > 0
+    assert float(flat.flat_grad.abs().sum()) > 0
+    # optimizer pass over the full-dim flat space, on a shard-sized
+    # slice (a full 1.8B-elem CPU reference AdamW takes minutes)
+    from prime_amd import ops
+
+    k = 1 << 20
+    ops.reference.adamw_step(flat.master32[:k], flat.flat_grad[:k].float(),
+                             torch.zeros(k), torch.zeros(k),
+                             1e-4, 0.9, 0.95, 1e-8, 0.1, 1)
+
+
+def _fsdp70_worker(rank, world):
+    from prime_amd.models import build_model
+    from prime_amd.parallel.fsdp import ShardedParamSpace
+    from prime_amd.parallel.mesh import ElasticDeviceMesh, MeshConfig
+
+    import torch
+
+    torch.manual_seed(0)
+    mesh = ElasticDeviceMesh(MeshConfig(worker_size=world))
+    m = _build_big_cheap("llama_70b", n_layers=2, vocab_size=512, max_seq=128,
+                          activation_checkpointing=True)
+    flat = ShardedParamSpace(m, mesh)
+    # shard layout: unit sizes match the 70B block param count
+    blk_params = 8192 * (64 + 2 * 8) * 128 + 8192 * 8192 + 3 * 8192 * 28672 + 2 * 8192
+    for u in flat.units:
+        assert u.numel >= blk_params  # >=: per-param 64-elem alignment
+        assert u.shard_len * world == u.numel_padded
+    # layout-only: a single 70B-unit all_gather is ~7.3 GB over gloo
+    # loopback and blows the CPU-lane budget; the gather path itself is
+    # covered at test dims by test_fsdp.py and on GPU by the sizing run
+    val = float(flat.flat_w[0])
+    total = flat.numel_padded
+    assert total * world * 2 / 1e9 > 3.0  # two 70B blocks of bf16 sharded
+    import torch.distributed as dist
+
+    dist.barrier()
+    dist.destroy_process_group()
+    return {"val": val, "shard_len": flat.units[0].shard_len}
+
+
+def test_70b_dims_fsdp_shard_layout_cpu():
+    from tests.conftest import run_distributed
+
+    outs = run_distributed(_fsdp70_worker, 2, timeout=600)
+    assert abs(outs[0]["val"] - 0.01) < 1e-3  # shard holds the fill value
+    assert outs[0]["shard_len"] == outs[1]["shard_len"]
