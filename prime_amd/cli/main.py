@@ -24,6 +24,7 @@ import typer
 from typer.core import TyperGroup
 
 from ..utils.config import ConfigError, default_config_toml, load_config
+from ..utils.display import emit_json, secho, set_plain, table
 from ..utils.logging import render_log_line
 
 
@@ -38,6 +39,14 @@ class DefaultRunGroup(TyperGroup):
 
 
 app = typer.Typer(help="MI355X-native DiLoCo training engine", no_args_is_help=True)
+
+
+@app.callback()
+def _root(plain: bool = typer.Option(
+        False, "--plain", help="no colors/styling (or PRIME_AMD_PLAIN=1)")):
+    if plain:
+        set_plain(True)
+
 train_app = typer.Typer(help="Launch and manage training runs",
                         no_args_is_help=True, cls=DefaultRunGroup)
 config_app = typer.Typer(help="CLI configuration", no_args_is_help=True)
@@ -62,9 +71,9 @@ def _find_run(ref: str) -> Path:
     if len(matches) == 1:
         return matches[0]
     if not matches:
-        typer.secho(f"no run matching '{ref}'", fg="red")
+        secho(f"no run matching '{ref}'", fg="red")
         raise typer.Exit(1)
-    typer.secho(f"ambiguous run '{ref}': {[m.name for m in matches]}", fg="red")
+    secho(f"ambiguous run '{ref}': {[m.name for m in matches]}", fg="red")
     raise typer.Exit(1)
 
 
@@ -100,11 +109,23 @@ def run(
     config: str = typer.Argument(..., help="TOML run config"),
     nproc: int = typer.Option(0),
     detach: bool = typer.Option(False, "--detach", "-d"),
+    env_file: list[str] = typer.Option([], "--env-file", "-e",
+                                       help="dotenv file(s) with ${VAR} "
+                                            "expansion (e.g. WANDB_API_KEY)"),
+    env: list[str] = typer.Option([], "--env",
+                                  help="extra KEY=VALUE for the run env"),
 ):
+    from ..utils.env_vars import EnvFileError, collect_env
+
     try:
         cfg = load_config(config)
     except ConfigError as e:
-        typer.secho(str(e), fg="red")
+        secho(str(e), fg="red")
+        raise typer.Exit(2)
+    try:
+        run_env = collect_env(env_file, env)
+    except (EnvFileError, OSError) as e:
+        secho(str(e), fg="red")
         raise typer.Exit(2)
     run_id = f"{cfg.run_name}-{uuid.uuid4().hex[:6]}"
     run_dir = runs_root() / run_id
@@ -128,10 +149,11 @@ def run(
             "--master-addr", "127.0.0.1", "--master-port", str(_free_port()),
             "-m", "prime_amd.cli.runner", str(run_dir / "config.toml"), str(run_dir),
         ]
-    typer.secho(f"run {run_id}: {' '.join(cmd)}", fg="cyan")
+    secho(f"run {run_id}: {' '.join(cmd)}", fg="cyan")
     log = open(run_dir / "launcher.log", "w")
+    child_env = {**os.environ, **run_env}
     proc = subprocess.Popen(cmd, stdout=log, stderr=subprocess.STDOUT,
-                            start_new_session=True)
+                            start_new_session=True, env=child_env)
     (run_dir / "launcher.pid").write_text(str(proc.pid))
     (run_dir / "status.json").write_text(json.dumps(
         {"status": "STARTING", "pid": proc.pid, "started": time.time()}))
@@ -141,12 +163,12 @@ def run(
     try:
         rc = proc.wait()
     except KeyboardInterrupt:
-        typer.secho("interrupt: stopping run", fg="yellow")
+        secho("interrupt: stopping run", fg="yellow")
         os.killpg(proc.pid, signal.SIGTERM)
         rc = proc.wait()
     st = _status(run_dir)
     color = "green" if st.get("status") == "COMPLETED" else "red"
-    typer.secho(f"run {run_id}: {st.get('status')} (rc={rc})", fg=color)
+    secho(f"run {run_id}: {st.get('status')} (rc={rc})", fg=color)
     if st.get("status") == "COMPLETED":
         typer.echo(json.dumps(st.get("result", {}), indent=2))
 
@@ -159,10 +181,10 @@ def init(
     """Write a config template (reference: config template generator)."""
     p = Path(path)
     if p.exists():
-        typer.secho(f"{p} exists; not overwriting", fg="red")
+        secho(f"{p} exists; not overwriting", fg="red")
         raise typer.Exit(1)
     p.write_text(default_config_toml(model))
-    typer.secho(f"wrote {p}", fg="green")
+    secho(f"wrote {p}", fg="green")
 
 
 @train_app.command("list")
@@ -211,18 +233,26 @@ def logs(
     follow: bool = typer.Option(False, "--follow", "-f"),
     rank: int = typer.Option(0),
     raw: bool = typer.Option(False, help="raw JSON lines"),
+    component: Optional[str] = typer.Option(
+        None, help="only lines of this type (progress/checkpoint/result/...)"),
 ):
     d = _find_run(run)
     f = d / f"rank{rank}.log"
     if not f.exists():
         f = d / "launcher.log"
     if not f.exists():
-        typer.secho("no logs yet", fg="yellow")
+        secho("no logs yet", fg="yellow")
         raise typer.Exit(1)
     with open(f) as fh:
         while True:
             line = fh.readline()
             if line:
+                if component:
+                    try:
+                        if json.loads(line).get("type") != component:
+                            continue
+                    except (ValueError, AttributeError):
+                        continue
                 if raw:
                     typer.echo(line.rstrip())
                 else:
@@ -256,17 +286,25 @@ def metrics(run: str = typer.Argument(...), last: int = typer.Option(10),
 
 
 @train_app.command("checkpoints")
-def checkpoints(run: str = typer.Argument(...)):
+def checkpoints(run: str = typer.Argument(...),
+                json_out: bool = typer.Option(False, "--json")):
     d = _find_run(run)
     ck = d / "ckpt"
-    if not ck.exists():
+    rows = []
+    if ck.exists():
+        for tag in sorted(ck.iterdir()):
+            if tag.is_dir() and not tag.is_symlink():
+                files = list(tag.glob("*.pt"))
+                rows.append({"tag": tag.name, "files": len(files),
+                             "bytes": sum(f.stat().st_size for f in files)})
+    if json_out:
+        emit_json(rows)
+        return
+    if not rows:
         typer.echo("no checkpoints")
         return
-    for tag in sorted(ck.iterdir()):
-        if tag.is_dir() and not tag.is_symlink():
-            files = list(tag.glob("*.pt"))
-            size = sum(f.stat().st_size for f in files)
-            typer.echo(f"{tag.name:16s} {len(files)} file(s) {size/1e9:8.2f} GB")
+    for r in rows:
+        typer.echo(f"{r['tag']:16s} {r['files']} file(s) {r['bytes']/1e9:8.2f} GB")
 
 
 @train_app.command("stop")
@@ -275,7 +313,7 @@ def stop(run: str = typer.Argument(...)):
     st = _status(d)
     pid = st.get("pid")
     if not pid or not _pid_alive(pid):
-        typer.secho("not running", fg="yellow")
+        secho("not running", fg="yellow")
         return
     # exact recorded pid (process group) — never pattern-based. The runner
     # handles SIGTERM gracefully: checkpoint at the next step boundary,
@@ -284,7 +322,7 @@ def stop(run: str = typer.Argument(...)):
         os.killpg(pid, signal.SIGTERM)
     except OSError:
         os.kill(pid, signal.SIGTERM)
-    typer.secho(f"stop requested for {d.name} (graceful: checkpoints first)", fg="green")
+    secho(f"stop requested for {d.name} (graceful: checkpoints first)", fg="green")
 
 
 @train_app.command("delete")
@@ -296,10 +334,10 @@ def delete(run: str = typer.Argument(...),
     d = _find_run(run)
     st = _status(d)
     if st.get("status") in ("RUNNING", "STARTING") and not force:
-        typer.secho("run appears to be RUNNING; stop it first or use --force", fg="red")
+        secho("run appears to be RUNNING; stop it first or use --force", fg="red")
         raise typer.Exit(1)
     shutil.rmtree(d)
-    typer.secho(f"deleted {d.name}", fg="green")
+    secho(f"deleted {d.name}", fg="green")
 
 
 @train_app.command("restart")
@@ -308,7 +346,7 @@ def restart(run: str = typer.Argument(...), detach: bool = typer.Option(False, "
     d = _find_run(run)
     cfg_path = d / "config.toml"
     if not cfg_path.exists():
-        typer.secho("run has no config.toml", fg="red")
+        secho("run has no config.toml", fg="red")
         raise typer.Exit(1)
     text = cfg_path.read_text()
     if "[checkpoint]" in text and "resume" not in text:
@@ -321,21 +359,36 @@ def restart(run: str = typer.Argument(...), detach: bool = typer.Option(False, "
 
 
 def run_cmd(config: str, detach: bool) -> None:
-    run(config, nproc=0, detach=detach)
+    run(config, nproc=0, detach=detach, env_file=[], env=[])
 
 
 @train_app.command("report")
-def report(run: str = typer.Argument(...)):
+def report(run: str = typer.Argument(...),
+           json_out: bool = typer.Option(False, "--json")):
     """Summarize a run: status, loss trajectory, throughput, failures."""
     from ..utils.metrics import read_metrics
 
     d = _find_run(run)
     st = _status(d)
     rows = read_metrics(d / "metrics.jsonl")
-    typer.secho(f"run {d.name}: {st.get('status', '?')}", bold=True)
+    if json_out:
+        losses = [r["loss"] for r in rows if "loss" in r]
+        tps = [r["tokens_per_sec"] for r in rows if "tokens_per_sec" in r]
+        emit_json({
+            "run": d.name, "status": st.get("status"),
+            "failure_analysis": st.get("failure_analysis"),
+            "steps_logged": [rows[0]["step"], rows[-1]["step"]] if rows else None,
+            "loss": {"first": losses[0], "min": min(losses),
+                     "last": losses[-1]} if losses else None,
+            "tokens_per_sec": {"mean": sum(tps) / len(tps),
+                               "last": tps[-1]} if tps else None,
+            "result": st.get("result"),
+        })
+        return
+    secho(f"run {d.name}: {st.get('status', '?')}", bold=True)
     fa = st.get("failure_analysis")
     if fa:
-        typer.secho(f"  failure: {fa['category']} — {fa['hint']}", fg="red")
+        secho(f"  failure: {fa['category']} — {fa['hint']}", fg="red")
     if not rows:
         typer.echo("  no metrics recorded")
         return
@@ -355,10 +408,13 @@ def report(run: str = typer.Argument(...)):
 
 
 @train_app.command("models")
-def models():
+def models(json_out: bool = typer.Option(False, "--json")):
     """List model presets (reference: `prime train models`)."""
     from ..models import CONFIGS
 
+    if json_out:
+        emit_json({name: c.to_dict() for name, c in CONFIGS.items()})
+        return
     for name, c in CONFIGS.items():
         typer.echo(f"{name:16s} {c.n_params()/1e9:7.2f}B params  dim={c.dim} "
                    f"layers={c.n_layers} heads={c.n_heads}/{c.n_kv_heads} "
@@ -379,7 +435,7 @@ def config_view():
 @config_app.command("set")
 def config_set(key: str, value: str):
     Contexts().set(key, value)
-    typer.secho(f"{key} = {value}", fg="green")
+    secho(f"{key} = {value}", fg="green")
 
 
 @config_app.command("save")
@@ -388,9 +444,9 @@ def config_save(name: str):
     try:
         p = Contexts().save(name)
     except ValueError as e:
-        typer.secho(str(e), fg="red")
+        secho(str(e), fg="red")
         raise typer.Exit(1)
-    typer.secho(f"saved context '{name}' -> {p}", fg="green")
+    secho(f"saved context '{name}' -> {p}", fg="green")
 
 
 @config_app.command("use")
@@ -399,9 +455,9 @@ def config_use(name: str):
     try:
         Contexts().use(name)
     except (ValueError, FileNotFoundError) as e:
-        typer.secho(str(e), fg="red")
+        secho(str(e), fg="red")
         raise typer.Exit(1)
-    typer.secho(f"now using context '{name}'", fg="green")
+    secho(f"now using context '{name}'", fg="green")
 
 
 @config_app.command("envs")
@@ -416,9 +472,9 @@ def config_delete(name: str):
     try:
         Contexts().delete(name)
     except ValueError as e:
-        typer.secho(str(e), fg="red")
+        secho(str(e), fg="red")
         raise typer.Exit(1)
-    typer.secho(f"deleted context '{name}'", fg="green")
+    secho(f"deleted context '{name}'", fg="green")
 
 
 # ----------------------------------------------------------- prepare-data
@@ -442,7 +498,7 @@ def prepare_data(
     dtype = np.uint16 if vocab <= vocab_threshold else np.uint32
     arr = np.asarray(ids, dtype=dtype)
     arr.tofile(out)
-    typer.secho(f"wrote {len(arr):,} tokens (vocab {vocab}, {dtype.__name__}) to {out}",
+    secho(f"wrote {len(arr):,} tokens (vocab {vocab}, {dtype.__name__}) to {out}",
                 fg="green")
 
 
@@ -477,7 +533,7 @@ def eval_cmd(
         flat = FlatParamSpace(m)
         payload = CheckpointManager(checkpoint).load(map_location=dev)
         if payload is None:
-            typer.secho("no checkpoint found", fg="red")
+            secho("no checkpoint found", fg="red")
             raise typer.Exit(1)
         flat.load_flat_(payload["tensors"]["master32"].to(dev))
     dc = DataConfig(kind="token_file" if data else "synthetic", path=data,
@@ -527,12 +583,12 @@ def generate_cmd(
         flat = FlatParamSpace(m)
         payload = CheckpointManager(checkpoint).load(map_location=dev)
         if payload is None:
-            typer.secho("no checkpoint found", fg="red")
+            secho("no checkpoint found", fg="red")
             raise typer.Exit(1)
         flat.load_flat_(payload["tensors"]["master32"].to(dev))
     if prompt is not None:
         if tok is None:
-            typer.secho("--prompt requires --tokenizer", fg="red")
+            secho("--prompt requires --tokenizer", fg="red")
             raise typer.Exit(2)
         from ..utils.tokenizer import encode
 
@@ -563,9 +619,9 @@ def export_cmd(
     try:
         n = export_safetensors(checkpoint, model, out)
     except FileNotFoundError as e:
-        typer.secho(str(e), fg="red")
+        secho(str(e), fg="red")
         raise typer.Exit(1)
-    typer.secho(f"wrote {n} tensors to {out}", fg="green")
+    secho(f"wrote {n} tensors to {out}", fg="green")
 
 
 # ------------------------------------------------------------------ store
@@ -578,7 +634,7 @@ def store_cmd(port: int = typer.Option(29777), addr: str = typer.Option("0.0.0.0
     from torch.distributed import TCPStore
 
     TCPStore(addr, port, is_master=True, wait_for_workers=False)
-    typer.secho(f"elastic store listening on {addr}:{port} (ctrl-c to stop)", fg="green")
+    secho(f"elastic store listening on {addr}:{port} (ctrl-c to stop)", fg="green")
     try:
         while True:
             _time.sleep(3600)

@@ -192,3 +192,77 @@ def test_report_command(tmp_path, runs_dir):
     r = runner.invoke(app, ["train", "report", run_id])
     assert r.exit_code == 0
     assert "loss" in r.output and "tokens/sec" in r.output
+
+
+def test_env_file_parsing(tmp_path):
+    from prime_amd.utils.env_vars import EnvFileError, collect_env, parse_env_file
+
+    f = tmp_path / ".env"
+    f.write_text(
+        "# comment\n"
+        "export WANDB_API_KEY=abc123\n"
+        "BASE=/data\n"
+        'OUT="${BASE}/runs"\n'
+        "LITERAL='${BASE}/raw'\n"
+        "FROM_PARENT=${PRIME_TEST_PARENT}\n"
+    )
+    import os
+
+    os.environ["PRIME_TEST_PARENT"] = "hello"
+    try:
+        env = parse_env_file(f)
+    finally:
+        del os.environ["PRIME_TEST_PARENT"]
+    assert env["WANDB_API_KEY"] == "abc123"
+    assert env["OUT"] == "/data/runs"
+    assert env["LITERAL"] == "${BASE}/raw"  # single quotes: no expansion
+    assert env["FROM_PARENT"] == "hello"
+    merged = collect_env([str(f)], ["EXTRA=1"])
+    assert merged["EXTRA"] == "1"
+    bad = tmp_path / "bad.env"
+    bad.write_text("NOT A LINE\n")
+    import pytest
+
+    with pytest.raises(EnvFileError):
+        parse_env_file(bad)
+
+
+def test_models_json_and_plain(tmp_path, monkeypatch):
+    import json as _json
+
+    from typer.testing import CliRunner
+
+    from prime_amd.cli.main import app
+
+    monkeypatch.setenv("PRIME_AMD_RUNS_DIR", str(tmp_path))
+    r = CliRunner().invoke(app, ["train", "models", "--json"])
+    assert r.exit_code == 0, r.output
+    d = _json.loads(r.output)
+    assert "intellect_10b" in d and d["intellect_10b"]["dim"] == 4096
+    r = CliRunner().invoke(app, ["--plain", "train", "models"])
+    assert r.exit_code == 0
+    assert "\x1b[" not in r.output  # no ANSI styling in plain mode
+
+
+def test_report_and_checkpoints_json(tmp_path, monkeypatch):
+    import json as _json
+
+    from typer.testing import CliRunner
+
+    from prime_amd.cli.main import app
+
+    monkeypatch.setenv("PRIME_AMD_RUNS_DIR", str(tmp_path))
+    d = tmp_path / "myrun-abc123"
+    d.mkdir(parents=True)
+    (d / "status.json").write_text('{"status": "COMPLETED"}')
+    (d / "metrics.jsonl").write_text(
+        '{"step": 1, "loss": 2.0, "tokens_per_sec": 100.0}\n'
+        '{"step": 2, "loss": 1.5, "tokens_per_sec": 110.0}\n'
+    )
+    r = CliRunner().invoke(app, ["train", "report", "myrun", "--json"])
+    assert r.exit_code == 0, r.output
+    rep = _json.loads(r.output)
+    assert rep["status"] == "COMPLETED" and rep["loss"]["last"] == 1.5
+    r = CliRunner().invoke(app, ["train", "checkpoints", "myrun", "--json"])
+    assert r.exit_code == 0
+    assert _json.loads(r.output) == []

@@ -5,7 +5,7 @@ import json
 import sys
 
 import numpy as np
-import torch
+
 
 sys.path.insert(0, "/root/repo")
 from prime_amd.utils.config import (DilocoConfig, DataSection, MetricsConfig,

@@ -12,7 +12,7 @@ sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 import torch
 
 from prime_amd import ops
-from prime_amd.ops.functional import _FlashAttention, transpose_bshd
+from prime_amd.ops.functional import _FlashAttention
 
 
 def main():
