@@ -480,26 +480,34 @@ def config_delete(name: str):
 # ----------------------------------------------------------- prepare-data
 @app.command("prepare-data")
 def prepare_data(
-    text_file: str = typer.Argument(..., help="input UTF-8 text file"),
-    out: str = typer.Argument(..., help="output token .bin (uint16/uint32)"),
+    inputs: list[str] = typer.Argument(..., help="input UTF-8 text file(s)"),
+    out: str = typer.Option(..., "--out", "-o",
+                            help="output .bin, or a directory for shards"),
     tokenizer: str = typer.Option(..., help="local tokenizer.json"),
     vocab_threshold: int = typer.Option(65535, help="uint16 if vocab fits"),
+    shard_tokens: int = typer.Option(512 * 1024 * 1024,
+                                     help="tokens per output shard"),
+    workers: int = typer.Option(0, help="parallel tokenizer processes"),
+    json_out: bool = typer.Option(False, "--json"),
 ):
-    """Tokenize a text file into the flat token .bin the token_file
-    dataloader consumes."""
-    import numpy as np
+    """Tokenize text into token shards for the token_file dataloader.
+    Streams inputs in bounded blocks (multi-GB corpora never load into
+    memory), tokenizes blocks in parallel workers, and rolls output
+    shards every --shard-tokens."""
+    from ..data.prepare import prepare_corpus
 
-    from ..utils.tokenizer import encode, load_tokenizer
-
-    tok = load_tokenizer(tokenizer)
-    text = Path(text_file).read_text()
-    ids = encode(tok, text)
-    vocab = tok.get_vocab_size()
-    dtype = np.uint16 if vocab <= vocab_threshold else np.uint32
-    arr = np.asarray(ids, dtype=dtype)
-    arr.tofile(out)
-    secho(f"wrote {len(arr):,} tokens (vocab {vocab}, {dtype.__name__}) to {out}",
-                fg="green")
+    try:
+        res = prepare_corpus(inputs, out, tokenizer,
+                             vocab_threshold=vocab_threshold,
+                             shard_tokens=shard_tokens, workers=workers)
+    except FileNotFoundError as e:
+        secho(f"input not found: {e}", fg="red")
+        raise typer.Exit(1)
+    if json_out:
+        emit_json(res)
+    else:
+        secho(f"wrote {res['tokens']:,} tokens (vocab {res['vocab']}, "
+              f"{res['dtype']}) to {len(res['shards'])} shard(s)", fg="green")
 
 
 # ------------------------------------------------------------------- eval

@@ -61,13 +61,24 @@ class TokenFileDataset:
             raise ValueError("token_file dataset requires data.path")
         p = Path(cfg.path)
         dtype = np.uint16 if vocab_size <= 65535 else np.uint32
-        self.tokens = np.memmap(p, dtype=dtype, mode="r")
+        # single flat .bin, or a directory of part_*.bin shards (the
+        # streaming prepare-data writer's layout)
+        if p.is_dir():
+            parts = sorted(p.glob("*.bin"))
+            if not parts:
+                raise ValueError(f"no .bin shards under {p}")
+            self.files = [np.memmap(f, dtype=dtype, mode="r") for f in parts]
+        else:
+            self.files = [np.memmap(p, dtype=dtype, mode="r")]
         self.cfg = cfg
         self.shard = shard
         self.n_shards = n_shards
         self.batch_idx = 0
         self.epoch = 0
-        self.windows = (len(self.tokens) - 1) // cfg.seq_len
+        # per-file window table (windows never straddle shard files)
+        self._file_windows = [max(0, (len(f) - 1) // cfg.seq_len) for f in self.files]
+        self._file_base = np.cumsum([0] + self._file_windows)
+        self.windows = int(self._file_base[-1])
         if self.windows < 1:
             raise ValueError(f"{p} too small for seq_len={cfg.seq_len}")
         self._perm: np.ndarray | None = None
@@ -96,8 +107,13 @@ class TokenFileDataset:
         for i in range(b):
             gi = self.batch_idx * self.n_shards * b + self.shard * b + i
             w = self._window(gi)
-            start = w * s
-            chunk = torch.from_numpy(self.tokens[start : start + s + 1].astype(np.int64))
+            fi = int(np.searchsorted(self._file_base, w, side="right")) - 1
+            start = (w - int(self._file_base[fi])) * s
+            f = self.files[fi]
+            chunk = torch.from_numpy(f[start : start + s + 1].astype(np.int64))
+            if chunk.numel() < s + 1:  # last window of a shard: wrap pad
+                pad = torch.from_numpy(f[: s + 1 - chunk.numel()].astype(np.int64))
+                chunk = torch.cat([chunk, pad])
             xs.append(chunk[:-1])
             ys.append(chunk[1:])
         self.batch_idx += 1

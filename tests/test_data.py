@@ -56,7 +56,7 @@ def test_token_file_shuffle_covers_epoch(tmp_path):
         x, _ = dl.next_batch(torch.device("cpu"))
         seen.append(int(x[0, 0]))
     # one epoch visits every window exactly once (permutation)
-    starts = {int(t) for t in np.asarray(dl.tokens[: 64 * 20 : 64])}
+    starts = {int(t) for t in np.asarray(dl.files[0][: 64 * 20 : 64])}
     assert len(seen) == dl.windows
     assert set(seen) == starts or len(set(seen)) == len(seen)
 
@@ -97,11 +97,58 @@ def test_prepare_data_roundtrip(tmp_path):
     from prime_amd.cli.main import app
 
     out_bin = tmp_path / "toks.bin"
-    r = CliRunner().invoke(app, ["prepare-data", str(src), str(out_bin),
-                                 "--tokenizer", str(tok_path)])
+    r = CliRunner().invoke(app, ["prepare-data", str(src), "--out",
+                                 str(out_bin), "--tokenizer", str(tok_path)])
     assert r.exit_code == 0, r.output
     cfg = DataConfig(kind="token_file", path=str(out_bin), seq_len=64,
                      micro_batch_size=2, shuffle=True)
     dl = build_dataloader(cfg, 1000, 0, 1)
     x, y = dl.next_batch(torch.device("cpu"))
     assert x.shape == (2, 64) and int(x.max()) < tok.get_vocab_size()
+
+
+def test_prepare_data_sharded_streaming(tmp_path):
+    """Multi-file input, parallel workers, small shard size -> multiple
+    output shards consumed as a directory by TokenFileDataset."""
+    tokenizers = __import__("tokenizers")
+    from tokenizers.models import WordLevel
+    from tokenizers.pre_tokenizers import Whitespace
+    from tokenizers.trainers import WordLevelTrainer
+
+    text = "alpha beta gamma delta epsilon zeta eta theta " * 400
+    srcs = []
+    for i in range(3):
+        f = tmp_path / f"part{i}.txt"
+        f.write_text(text)
+        srcs.append(f)
+    tok = tokenizers.Tokenizer(WordLevel(unk_token="<unk>"))
+    tok.pre_tokenizer = Whitespace()
+    tok.train_from_iterator([text], WordLevelTrainer(special_tokens=["<unk>"]))
+    tok_path = tmp_path / "tok.json"
+    tok.save(str(tok_path))
+
+    from prime_amd.data.prepare import prepare_corpus
+
+    out_dir = tmp_path / "shards"
+    res = prepare_corpus([str(s) for s in srcs], out_dir, str(tok_path),
+                         shard_tokens=4000, workers=2)
+    assert res["tokens"] == 3 * 8 * 400
+    assert len(res["shards"]) == (res["tokens"] + 3999) // 4000
+    # shard directory consumed directly
+    cfg = DataConfig(kind="token_file", path=str(out_dir), seq_len=64,
+                     micro_batch_size=2, shuffle=True)
+    dl = build_dataloader(cfg, 1000, 0, 1)
+    seen_windows = dl.windows
+    assert seen_windows == sum(
+        max(0, (4000 if i < len(res["shards"]) - 1 else
+                res["tokens"] % 4000 or 4000) - 1) // 64
+        for i in range(len(res["shards"])))
+    x, y = dl.next_batch(torch.device("cpu"))
+    assert x.shape == (2, 64)
+    # resume across shard boundaries
+    st = dl.state_dict()
+    xa, _ = dl.next_batch(torch.device("cpu"))
+    dl2 = build_dataloader(cfg, 1000, 0, 1)
+    dl2.load_state_dict(st)
+    xb, _ = dl2.next_batch(torch.device("cpu"))
+    assert torch.equal(xa, xb)
