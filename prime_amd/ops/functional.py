@@ -368,15 +368,19 @@ def invalidate_wt_cache() -> None:
 
 def _wt_of(w: torch.Tensor):
     """Cached W^T [K,N] of a [N,K] weight; None when the shape doesn't
-    tile for the transpose kernel."""
+    tile for the transpose kernel. Keyed by (data_ptr, shape): id(w) is
+    NOT stable — under activation checkpointing the saved-tensor unpack
+    hands backward a fresh transient object per call, and recycled ids
+    collided across different weights (caught by the 70B sizing run)."""
     N, K = w.shape
     if K % 128 or N % 64:
         return None
-    hit = _WT_CACHE.get(id(w))
+    key = (w.data_ptr(), N, K)
+    hit = _WT_CACHE.get(key)
     if hit is not None and hit[0] == _WT_EPOCH:
         return hit[1]
     wt = transpose_bshd(w.view(1, N, K // 128, 128)).view(K, N)
-    _WT_CACHE[id(w)] = (_WT_EPOCH, wt)
+    _WT_CACHE[key] = (_WT_EPOCH, wt)
     return wt
 
 
