@@ -44,7 +44,15 @@ class LlamaConfig:
         return 6.0 * n_mat
 
     def attn_flops_per_token(self, seq_len: int, causal: bool = True) -> float:
-        # fwd QK^T + PV = 4*S*D_head*H flops/token; x3 for fwd+bwd; /2 causal
+        """MODEL attention FLOPs/token: fwd QK^T+PV = 4*S*hd*H, x3 for
+        fwd+bwd (the standard 1+2 convention), /2 causal.
+
+        Convention note: the flash backward IMPLEMENTATION performs ~2.5x
+        the forward's matmuls (dq/dkv recompute S and dP), so the MFU
+        numerator deliberately under-counts what the kernels execute —
+        recomputation is not useful model work. Combined with the 2.5 PF
+        DENSE peak denominator (AMD's 5 PF figure is 2:1-sparse), every
+        MFU printed by this repo is a conservative bound."""
         f = 4.0 * seq_len * self.head_dim * self.n_heads * 3.0
         return f / 2.0 if causal else f
 

@@ -314,7 +314,21 @@ class DilocoOptimizer:
                 if self._scale != 1.0:
                     delta_chunk.mul_(self._scale)  # weighted outer average
                 if delta_chunk.is_cuda:
-                    host = delta_chunk.to("cpu")
+                    # reusable pinned bounce buffer: the gloo ring reads
+                    # host memory; a fresh pageable .to("cpu") per chunk
+                    # both allocates and slows the D2H path
+                    buf = getattr(self, "_elastic_host_buf", None)
+                    if buf is None or buf.numel() < delta_chunk.numel():
+                        try:
+                            buf = torch.empty(delta_chunk.numel(),
+                                              dtype=delta_chunk.dtype,
+                                              pin_memory=True)
+                        except RuntimeError:
+                            buf = torch.empty(delta_chunk.numel(),
+                                              dtype=delta_chunk.dtype)
+                        self._elastic_host_buf = buf
+                    host = buf[: delta_chunk.numel()]
+                    host.copy_(delta_chunk)  # sync D2H (ring needs it now)
                     ring_allreduce_int8_pg(host, v.pg, v.my_index, v.world)
                     delta_chunk.copy_(host, non_blocking=True)
                 else:
