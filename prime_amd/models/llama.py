@@ -50,6 +50,10 @@ class Attention(nn.Module):
         self.wqkv = PrimeLinear(cfg.dim, (cfg.n_heads + 2 * cfg.n_kv_heads) * hd, bias=False)
         self.wo = PrimeLinear(cfg.n_heads * hd, cfg.dim, bias=False)
 
+    # set by the Trainer for sequence-parallel training: attention runs
+    # over the full context via two all-to-alls (parallel/seqpar.py)
+    sp_group = None
+
     def forward(self, x, cos, sin, cache=None, pos: int = 0, pos_dev=None):
         B, S, _ = x.shape
         cfg = self.cfg
@@ -80,7 +84,12 @@ class Attention(nn.Module):
             # prefill (pos == 0): causal flash over the prompt
             o = ops.flash_attention(q, k, v, causal=True)
             return self.wo(o.reshape(B, S, cfg.n_heads * hd))
-        o = ops.flash_attention(q, k, v, causal=True)
+        if self.sp_group is not None:
+            from ..parallel.seqpar import ulysses_attention
+
+            o = ulysses_attention(q, k, v, causal=True, group=self.sp_group)
+        else:
+            o = ops.flash_attention(q, k, v, causal=True)
         return self.wo(o.reshape(B, S, cfg.n_heads * hd))
 
 
@@ -130,6 +139,7 @@ class Llama(nn.Module):
         cos, sin = ops.reference.rope_tables(cfg.head_dim, cfg.max_seq, cfg.rope_theta)
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
+        self.sp_pos_offset = 0  # sequence-parallel rank's token offset
         self.apply(self._init)
         # scaled init for residual-out projections (GPT-2 style)
         std = 0.02 / math.sqrt(2 * cfg.n_layers)
@@ -160,6 +170,8 @@ class Llama(nn.Module):
         device-scalar position dict for hipGraph-captured decode."""
         x = self.tok_embeddings(tokens)
         res = None
+        if pos == 0 and self.sp_pos_offset:
+            pos = self.sp_pos_offset  # SP rank's slice starts mid-context
         cos, sin = self.rope_cos, self.rope_sin
         for i, blk in enumerate(self.layers):
             if self.activation_checkpointing and self.training:

@@ -79,6 +79,24 @@ class Trainer:
         self.model.reset_rope(self.device)  # tables stay fp32
 
         self.fsdp = cfg.parallel.fsdp and cfg.parallel.worker_size > 1
+        self.seq_par = cfg.parallel.seq_parallel and cfg.parallel.worker_size > 1
+        if self.seq_par:
+            if self.fsdp:
+                raise ValueError("seq_parallel and fsdp are mutually "
+                                 "exclusive inner-parallelism modes")
+            W = cfg.parallel.worker_size
+            if self.model_cfg.n_heads % W or self.model_cfg.n_kv_heads % W:
+                raise ValueError(
+                    f"seq_parallel needs n_heads ({self.model_cfg.n_heads}) and "
+                    f"n_kv_heads ({self.model_cfg.n_kv_heads}) divisible by "
+                    f"worker_size ({W})")
+            if self.model_cfg.max_seq < cfg.model.seq_len * W:
+                raise ValueError(
+                    f"seq_parallel full context {cfg.model.seq_len * W} exceeds "
+                    f"model max_seq {self.model_cfg.max_seq}")
+            for blk in self.model.layers:
+                blk.attn.sp_group = self.mesh.local_group
+            self.model.sp_pos_offset = self.mesh.worker_rank * cfg.model.seq_len
         if self.fsdp:
             from .parallel.fsdp import ShardedParamSpace
 
@@ -166,15 +184,21 @@ class Trainer:
                 # elastic: this whole torchrun job is one worker, leader = 0
                 dist.broadcast(t, src=0, group=self.mesh.local_group)
                 widx = int(t[0])
+        sp_W = cfg.parallel.worker_size if self.seq_par else 1
         data_cfg = DataConfig(
-            kind=cfg.data.kind, path=cfg.data.path, seq_len=cfg.model.seq_len,
+            kind=cfg.data.kind, path=cfg.data.path,
+            # SP: every rank of the worker loads the SAME full-context
+            # rows and trains on its own seq_len-token slice
+            seq_len=cfg.model.seq_len * sp_W,
             micro_batch_size=cfg.data.micro_batch_size,
             seed=cfg.data.seed + 100003 * (widx or 0),
             shuffle=cfg.data.shuffle,
         )
         self.data = build_dataloader(
             data_cfg, self.model_cfg.vocab_size,
-            shard=self.mesh.rank, n_shards=max(1, self.mesh.world_size),
+            shard=self.mesh.worker_id if self.seq_par else self.mesh.rank,
+            n_shards=max(1, self.mesh.n_workers if self.seq_par
+                         else self.mesh.world_size),
         )
 
         self.ckpt = None
@@ -205,6 +229,11 @@ class Trainer:
         loss_acc = None
         for _ in range(cfg.data.grad_accum):
             x, y = self.data.next_batch(self.device)
+            if self.seq_par:
+                sl = cfg.model.seq_len
+                r = self.mesh.worker_rank
+                x = x[:, r * sl : (r + 1) * sl]
+                y = y[:, r * sl : (r + 1) * sl]
             loss = self.model.loss(x, y) / cfg.data.grad_accum
             loss.backward()
             loss_acc = loss.detach() if loss_acc is None else loss_acc + loss.detach()

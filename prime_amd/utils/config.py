@@ -61,6 +61,10 @@ class DilocoConfig(_Strict):
 class ParallelConfig(_Strict):
     worker_size: int = 1              # GPUs per DiLoCo worker
     fsdp: bool = False                # shard params across the worker
+    # sequence (context) parallelism: the worker's ranks each hold
+    # seq_len tokens of one (worker_size x seq_len)-token context;
+    # attention spans the full context via all-to-alls (Ulysses style)
+    seq_parallel: bool = False
     backend: str | None = None        # nccl | gloo (default: auto)
     elastic: bool = False
     heartbeat_interval: float = 5.0

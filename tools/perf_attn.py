@@ -16,7 +16,9 @@ from prime_amd.ops.functional import _FlashAttention
 
 
 def main():
-    iters = int(sys.argv[1]) if len(sys.argv) > 1 else 20
+    nums = [a for a in sys.argv[1:] if a.isdigit()]
+    iters = int(nums[0]) if nums else 20
+    causal = "--non-causal" not in sys.argv
     B, S, H, Hkv, D = 8, 2048, 32, 8, 128
     torch.manual_seed(0)
     dev = "cuda:0"
@@ -24,8 +26,8 @@ def main():
     k = torch.randn(B, S, Hkv, D, device=dev).bfloat16().requires_grad_(True)
     v = torch.randn(B, S, Hkv, D, device=dev).bfloat16().requires_grad_(True)
 
-    # causal attention FLOPs (fwd): 2 matmuls * S^2/2 * D * H * B * 2
-    fwd_flops = 2 * 2 * (S * S / 2) * D * H * B
+    # attention FLOPs (fwd): 2 matmuls * S^2(/2 causal) * D * H * B * 2
+    fwd_flops = 2 * 2 * (S * S / (2 if causal else 1)) * D * H * B
     bwd_flops = 2.5 * fwd_flops  # 5 matmuls in bwd vs 2 in fwd
 
     def timeit(fn, n):
@@ -38,15 +40,15 @@ def main():
         return (time.perf_counter() - t0) / n
 
     # forward only
-    t_fwd = timeit(lambda: _FlashAttention.apply(q, k, v, True, D**-0.5), iters)
+    t_fwd = timeit(lambda: _FlashAttention.apply(q, k, v, causal, D**-0.5), iters)
     print(f"fwd : {t_fwd*1e3:7.3f} ms  {fwd_flops/t_fwd/1e12:7.1f} TF/s")
 
     # fwd+bwd (isolates bwd by subtraction)
-    o = _FlashAttention.apply(q, k, v, True, D**-0.5)
+    o = _FlashAttention.apply(q, k, v, causal, D**-0.5)
     do = torch.randn_like(o)
 
     def fb():
-        out = _FlashAttention.apply(q, k, v, True, D**-0.5)
+        out = _FlashAttention.apply(q, k, v, causal, D**-0.5)
         out.backward(do)
         q.grad = k.grad = v.grad = None
 
