@@ -650,6 +650,37 @@ def store_cmd(port: int = typer.Option(29777), addr: str = typer.Option("0.0.0.0
         pass
 
 
+# ------------------------------------------------------------------ serve
+@app.command("serve")
+def serve_cmd(
+    model: str = typer.Option("llama_150m"),
+    checkpoint: Optional[str] = typer.Option(None, help="checkpoint dir"),
+    tokenizer: Optional[str] = typer.Option(None, help="local tokenizer.json"),
+    host: str = typer.Option("127.0.0.1"),
+    port: int = typer.Option(8392),
+):
+    """Serve the model over an OpenAI-style HTTP API (engine-side
+    counterpart of the reference's hosted inference surface):
+    GET /v1/models, POST /v1/completions, POST /v1/chat/completions."""
+    import uvicorn
+
+    from ..serve import create_app, load_model_for_serving
+
+    tok = None
+    if tokenizer:
+        from ..utils.tokenizer import load_tokenizer
+
+        tok = load_tokenizer(tokenizer)
+    try:
+        m = load_model_for_serving(model, checkpoint)
+    except FileNotFoundError as e:
+        secho(str(e), fg="red")
+        raise typer.Exit(1)
+    secho(f"serving {model} on http://{host}:{port} (ctrl-c to stop)", fg="green")
+    uvicorn.run(create_app(m, model, tok), host=host, port=port,
+                log_level="warning")
+
+
 # ------------------------------------------------------------------ bench
 @app.command("bench")
 def bench(steps: int = 10, warmup: int = 3, model: str = "intellect_10b"):
