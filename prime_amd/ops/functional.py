@@ -260,16 +260,17 @@ class _FlashAttention(torch.autograd.Function):
         dv = torch.empty(B, S, Hkv, D, device=q.device, dtype=q.dtype)
         # split the causal q loop across the grid: the unsplit grid is only
         # B*Hkv*S/64 blocks with 1..S/64 trip-count imbalance
-        base_blocks = B * Hkv * (S // 64)
         import os
 
         env = os.environ.get("PRIME_AMD_DKV_SPLITS")
         if env:
             splits = max(1, min(int(env), S // 64))
         else:
-            # A/B on MI355X @10B shapes: splits=2 (167 TF) > 4 (161) > 8 (153)
-            # — workspace traffic outweighs balance beyond ~4096 blocks
-            splits = max(1, min(8, 4096 // max(1, base_blocks), S // 64))
+            # measured on MI355X (10B shapes): S=2048 wants splits=2
+            # (bench 14.6k at 2 vs 14.5k at 4 — workspace traffic), S=8192
+            # wants 8 (12,690 vs 12,593 tok/s — causal trip imbalance
+            # dominates at long context)
+            splits = max(1, min(8, S // 1024, S // 64))
         ws = torch.empty(2, splits, B, Hkv, S, D, device=q.device,
                          dtype=torch.float32)
         check(
