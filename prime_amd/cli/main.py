@@ -421,6 +421,115 @@ def models(json_out: bool = typer.Option(False, "--json")):
                    f"vocab={c.vocab_size}")
 
 
+@train_app.command("usage")
+def usage(run: str = typer.Argument(...),
+          json_out: bool = typer.Option(False, "--json")):
+    """Resource usage for a run: tokens, wall time, GPU-seconds, FLOPs
+    (engine-side counterpart of the reference's billing/usage views)."""
+    from ..models import CONFIGS
+    from ..utils.metrics import model_flops_per_token, read_metrics
+
+    d = _find_run(run)
+    st = _status(d)
+    rows = read_metrics(d / "metrics.jsonl")
+    cfg_txt = (d / "config.toml")
+    n_gpus, model_name, seq_len = 1, None, 2048
+    if cfg_txt.exists():
+        from ..utils.config import load_config
+
+        try:
+            c = load_config(cfg_txt)
+            n_gpus = c.parallel.worker_size
+            model_name, seq_len = c.model.name, c.model.seq_len
+        except Exception:  # noqa: BLE001 — usage stays best-effort
+            pass
+    if not rows:
+        secho("no metrics recorded", fg="yellow")
+        raise typer.Exit(0)
+    steps = rows[-1].get("step", 0)
+    tps = [r["tokens_per_sec"] for r in rows if "tokens_per_sec" in r]
+    ms = [r["ms_per_step"] for r in rows if "ms_per_step" in r]
+    mfus = [r["mfu"] for r in rows if "mfu" in r]
+    wall_s = sum(msv for msv in ms) / 1000.0 * (steps / max(1, len(ms)))
+    tokens = int(sum(tps) / max(1, len(tps)) * wall_s) if tps else 0
+    flops = None
+    if model_name in CONFIGS:
+        flops = model_flops_per_token(CONFIGS[model_name], seq_len) * tokens
+    data = {
+        "run": d.name, "status": st.get("status"), "model": model_name,
+        "steps": steps, "est_tokens": tokens,
+        "est_wall_seconds": round(wall_s, 1),
+        "gpu_seconds": round(wall_s * n_gpus, 1),
+        "est_pflops": round(flops / 1e15, 1) if flops else None,
+        "mean_mfu": round(sum(mfus) / len(mfus), 4) if mfus else None,
+    }
+    if json_out:
+        emit_json(data)
+        return
+    for k, v in data.items():
+        typer.echo(f"  {k:18s}: {v}")
+
+
+@app.command("top")
+def top(interval: float = typer.Option(2.0, help="refresh seconds"),
+        once: bool = typer.Option(False, help="render once and exit")):
+    """Live dashboard over all runs (an engine-side sliver of the
+    reference's Lab TUI: statuses, loss, throughput, MFU, refreshed)."""
+    from ..utils.display import plain_mode
+
+    def rows_now():
+        out = []
+        for d in sorted(runs_root().iterdir()):
+            if not d.is_dir():
+                continue
+            st = _status(d)
+            last = _last_metrics(d)
+            out.append({
+                "run": d.name, "status": st.get("status", "?"),
+                "step": last.get("step", "-"),
+                "loss": last.get("loss"), "tok_s": last.get("tokens_per_sec"),
+                "mfu": last.get("mfu"), "outer": last.get("outer_steps", "-"),
+            })
+        return out
+
+    def fmt(v, spec):
+        return format(v, spec) if isinstance(v, (int, float)) else str(v or "-")
+
+    use_rich = not plain_mode() and not once
+    if use_rich:
+        try:
+            from rich.live import Live
+            from rich.table import Table
+        except ImportError:
+            use_rich = False
+    if not use_rich:
+        for r in rows_now():
+            typer.echo(f"{r['run']:40s} {r['status']:12s} "
+                       f"{fmt(r['step'], '>8')} {fmt(r['loss'], '.4f'):>10} "
+                       f"{fmt(r['tok_s'], ',.0f'):>12} {fmt(r['mfu'], '.3f'):>7}")
+        return
+
+    def build():
+        t = Table(title="prime-amd runs")
+        for col in ("run", "status", "step", "loss", "tok/s", "MFU", "outer"):
+            t.add_column(col)
+        for r in rows_now():
+            style = {"RUNNING": "green", "COMPLETED": "cyan",
+                     "FAILED": "red", "DIED": "red"}.get(r["status"].split(":")[0])
+            t.add_row(r["run"], r["status"], str(r["step"]),
+                      fmt(r["loss"], ".4f"), fmt(r["tok_s"], ",.0f"),
+                      fmt(r["mfu"], ".3f"), str(r["outer"]), style=style)
+        return t
+
+    with Live(build(), refresh_per_second=4) as live:
+        try:
+            while True:
+                time.sleep(interval)
+                live.update(build())
+        except KeyboardInterrupt:
+            pass
+
+
 # ----------------------------------------------------------------- config
 # named contexts with env-var precedence (reference: core/config.py)
 from ..utils.contexts import Contexts  # noqa: E402

@@ -266,3 +266,27 @@ def test_report_and_checkpoints_json(tmp_path, monkeypatch):
     r = CliRunner().invoke(app, ["train", "checkpoints", "myrun", "--json"])
     assert r.exit_code == 0
     assert _json.loads(r.output) == []
+
+
+def test_usage_and_top(tmp_path, monkeypatch):
+    import json as _json
+
+    from typer.testing import CliRunner
+
+    from prime_amd.cli.main import app
+
+    monkeypatch.setenv("PRIME_AMD_RUNS_DIR", str(tmp_path))
+    d = tmp_path / "urun-xyz"
+    d.mkdir(parents=True)
+    (d / "status.json").write_text('{"status": "COMPLETED"}')
+    (d / "metrics.jsonl").write_text(
+        '{"step": 10, "loss": 2.0, "tokens_per_sec": 1000.0, '
+        '"ms_per_step": 100.0, "mfu": 0.3}\n'
+    )
+    r = CliRunner().invoke(app, ["train", "usage", "urun", "--json"])
+    assert r.exit_code == 0, r.output
+    u = _json.loads(r.output)
+    assert u["steps"] == 10 and u["gpu_seconds"] > 0
+    r = CliRunner().invoke(app, ["--plain", "top", "--once"])
+    assert r.exit_code == 0, r.output
+    assert "urun-xyz" in r.output
