@@ -39,6 +39,70 @@ def _mem_available_bytes() -> int | None:
     return None
 
 
+def _disk_backed_dir() -> str | None:
+    """First candidate temp dir NOT on tmpfs/ramfs (a RAM-backed 'file'
+    would defeat the low-RAM fallback)."""
+    import os
+
+    mounts = []
+    try:
+        for line in open("/proc/mounts"):
+            parts = line.split()
+            if len(parts) >= 3:
+                mounts.append((parts[1], parts[2]))
+    except OSError:
+        return None
+    mounts.sort(key=lambda m: -len(m[0]))  # longest prefix first
+
+    def fstype(path: str) -> str:
+        rp = os.path.realpath(path)
+        for mp, t in mounts:
+            if rp == mp or rp.startswith(mp.rstrip("/") + "/") or mp == "/":
+                return t
+        return "?"
+
+    for cand in (os.environ.get("TMPDIR"), "/var/tmp", "/tmp", "."):
+        if cand and os.path.isdir(cand) and os.access(cand, os.W_OK):
+            if fstype(cand) not in ("tmpfs", "ramfs"):
+                return cand
+    return None
+
+
+def host_outer_buffer(k: int, world: int) -> torch.Tensor:
+    """Allocate fp32 host outer state (theta/momentum), tiered by what
+    the HOST can hold — N ranks x 84 GB can exceed a node's RAM, and an
+    OOM-killed rank ends a scale run:
+      1. pinned pages (full PCIe rate) when RAM is plentiful
+      2. pageable RAM when it fits
+      3. a disk-file-backed mapping (OS pages it) as the floor"""
+    import os
+    import tempfile
+
+    want_pin = os.environ.get("PRIME_AMD_OUTER_PIN", "auto")
+    need = 2 * k * 4 * world  # theta+buf across all local ranks
+    avail = _mem_available_bytes()
+    forced_file = os.environ.get("PRIME_AMD_OUTER_FILEBACKED", "0") == "1"
+    if want_pin != "0" and not forced_file:
+        if want_pin == "1" or (avail is not None and need < 0.5 * avail):
+            try:
+                return torch.zeros(k, dtype=torch.float32, pin_memory=True)
+            except RuntimeError:
+                pass
+    if not forced_file and (avail is None or need < 0.7 * avail):
+        return torch.zeros(k, dtype=torch.float32)
+    d = _disk_backed_dir()
+    if d is None:
+        return torch.zeros(k, dtype=torch.float32)  # best effort
+    f = tempfile.NamedTemporaryFile(prefix="prime_outer_", dir=d, delete=False)
+    path = f.name
+    f.truncate(k * 4)
+    f.close()
+    t = torch.from_file(path, shared=True, size=k, dtype=torch.float32)
+    os.unlink(path)  # inode lives until the mapping drops; pages to disk
+    t.zero_()
+    return t
+
+
 class DilocoOptimizer:
     def __init__(
         self,
@@ -93,23 +157,10 @@ class DilocoOptimizer:
             self._n_comm = (n + self.chunk - 1) // self.chunk * self.chunk
 
             def _host_buf(k: int) -> torch.Tensor:
-                # pinning is a throughput optimization (PCIe at full rate);
-                # skip it when the node can't afford it: 8 ranks x 84 GB of
-                # pinned pages would exhaust a typical host. Never risk the
-                # OOM-killer for an every-H-steps transfer.
                 import os
 
                 world = int(os.environ.get("WORLD_SIZE", 1))
-                want_pin = os.environ.get("PRIME_AMD_OUTER_PIN", "auto")
-                if want_pin != "0":
-                    need = 2 * k * 4 * world  # theta+buf across all local ranks
-                    avail = _mem_available_bytes()
-                    if want_pin == "1" or (avail is not None and need < 0.5 * avail):
-                        try:
-                            return torch.zeros(k, dtype=torch.float32, pin_memory=True)
-                        except RuntimeError:
-                            pass
-                return torch.zeros(k, dtype=torch.float32)
+                return host_outer_buffer(k, world)
 
             self.theta_outer = _host_buf(self._n_comm)
             # chunked D2H init (no 40 GB host temp from master32.to("cpu"))

@@ -125,3 +125,17 @@ def test_eight_worker_outer_soak():
     for r in range(1, 8):
         assert outs[r]["head"] == outs[0]["head"], r
         assert outs[r]["w"] == outs[0]["w"], r
+
+
+def test_host_outer_buffer_tiers(monkeypatch):
+    """Low-RAM hosts fall back to a disk-file-backed mapping instead of
+    risking the OOM killer at N ranks x 84 GB of outer state."""
+    from prime_amd.parallel.diloco import host_outer_buffer
+
+    t = host_outer_buffer(1024, 1)
+    assert t.numel() == 1024 and float(t.abs().sum()) == 0.0
+    monkeypatch.setenv("PRIME_AMD_OUTER_FILEBACKED", "1")
+    t2 = host_outer_buffer(1024, 8)
+    assert t2.numel() == 1024 and float(t2.abs().sum()) == 0.0
+    t2.fill_(3.0)  # mapping is writable
+    assert float(t2[123]) == 3.0
