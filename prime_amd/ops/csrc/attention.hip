@@ -302,7 +302,11 @@ __device__ __forceinline__ void permlane32_swap(unsigned& a, unsigned& b) {
   asm volatile("v_permlane32_swap_b32 %0, %1" : "+v"(a), "+v"(b));
 }
 
-// FLAGS bits: 1 = setprio around MFMA clusters, 2 = defer-max (THR=8)
+// FLAGS bits: 1 = setprio around MFMA clusters, 2 = defer-max (THR=8),
+// 4 = 3-buffer deep prefetch with counted vmcnt at the tile boundary
+// (guide T4: __syncthreads' vmcnt(0) drain waits on the JUST-issued
+// stage; with 2 tiles of lookahead the boundary only needs the oldest
+// in-flight tile, so vmcnt(4) leaves the newest 4 loads flying)
 template <int D, int FLAGS>
 __global__ __launch_bounds__(512, 2) void flash_fwd_v5_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
@@ -331,8 +335,9 @@ __global__ __launch_bounds__(512, 2) void flash_fwd_v5_kernel(
   const bf16* Kb = K + b * skb + hkv * skh;
   const bf16* Vtb = Vt + ((int64_t)(b * Hkv + hkv) * D) * S;
 
-  __shared__ bf16 k_lds[2][64 * D];
-  __shared__ bf16 vt_lds[2][D * 64];
+  constexpr int NBUF = (FLAGS & 4) ? 3 : 2;
+  __shared__ bf16 k_lds[NBUF][64 * D];
+  __shared__ bf16 vt_lds[NBUF][D * 64];
   __shared__ float bcast_all[8][32];  // per-wave alpha / inv-l broadcast
   float* bcast = bcast_all[wid];
 
@@ -364,15 +369,38 @@ __global__ __launch_bounds__(512, 2) void flash_fwd_v5_kernel(
   const int my_kv_end = causal ? (q0w + 32) : S;
   stage_tile<64, D, 512>(Kb, sks, k_lds[0], threadIdx.x);
   stage_tile<D, 64, 512>(Vtb, S, vt_lds[0], threadIdx.x);
+  if (FLAGS & 4) {  // 2-deep prologue
+    if (64 < kv_end) {
+      stage_tile<64, D, 512>(Kb + (int64_t)64 * sks, sks, k_lds[1], threadIdx.x);
+      stage_tile<D, 64, 512>(Vtb + 64, S, vt_lds[1], threadIdx.x);
+    }
+  }
   int idx = 0;
-  for (int kv = 0; kv < kv_end; kv += 64, idx ^= 1) {
-    __syncthreads();  // prev tile consumed + this tile's stage drained
-    if (kv + 64 < kv_end) {
-      // prefetch-behind-barrier: these loads get the whole tile's compute
-      // to land (the NEXT loop iteration's syncthreads drains them)
-      stage_tile<64, D, 512>(Kb + (int64_t)(kv + 64) * sks, sks,
-                             k_lds[idx ^ 1], threadIdx.x);
-      stage_tile<D, 64, 512>(Vtb + kv + 64, S, vt_lds[idx ^ 1], threadIdx.x);
+  for (int kv = 0, kt = 0; kv < kv_end; kv += 64, ++kt) {
+    idx = kt % NBUF;
+    if (FLAGS & 4) {
+      // tile kv's stages are >=1 full tile old: only the newest tile's
+      // 4 loads may still fly across this rendezvous
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      __builtin_amdgcn_sched_barrier(0);
+      const int pre = kv + 2 * 64;
+      if (pre < kv_end) {
+        stage_tile<64, D, 512>(Kb + (int64_t)pre * sks, sks,
+                               k_lds[(kt + 2) % NBUF], threadIdx.x);
+        stage_tile<D, 64, 512>(Vtb + pre, S, vt_lds[(kt + 2) % NBUF],
+                               threadIdx.x);
+      }
+    } else {
+      __syncthreads();  // prev tile consumed + this tile's stage drained
+      if (kv + 64 < kv_end) {
+        // prefetch-behind-barrier: these loads get the whole tile's
+        // compute to land (the next syncthreads drains them)
+        stage_tile<64, D, 512>(Kb + (int64_t)(kv + 64) * sks, sks,
+                               k_lds[(kt + 1) % NBUF], threadIdx.x);
+        stage_tile<D, 64, 512>(Vtb + kv + 64, S, vt_lds[(kt + 1) % NBUF],
+                               threadIdx.x);
+      }
     }
     if (kv >= my_kv_end) continue;  // fully-masked tile for this wave
 
@@ -1092,7 +1120,11 @@ PRIME_API int prime_flash_fwd(hipStream_t stream, const void* Q, const void* K,
   static const char* v5env = getenv("PRIME_ATTN_V5");
   static const char* flagenv = getenv("PRIME_ATTN_FLAGS");
   // measured on MI355X (15-iter sweep): FLAGS=2 417 TF > 3 (406) > 0 (387)
-  // > 1 (382) — defer-max pays, setprio slightly negative here
+  // > 1 (382) — defer-max pays, setprio slightly negative here. The
+  // 3-buffer counted-vmcnt variant (FLAGS=6) measured 401-402 vs 417-418
+  // in interleaved A/B: the prefetch-behind-barrier schedule already
+  // covers the stage latency, and the third buffer only adds LDS +
+  // addressing cost.
   const int flags = flagenv ? atoi(flagenv) : 2;
   const bool use_v5 =
       (D == 128) && (S % 256 == 0) && !(v5env && v5env[0] == '0');
@@ -1104,11 +1136,14 @@ PRIME_API int prime_flash_fwd(hipStream_t stream, const void* Q, const void* K,
                        (const bf16*)Vt, (bf16*)O, (float*)lse, (int)B,       \
                        (int)H, (int)Hkv, (int)S, (float)scale, (int)causal,  \
                        sqb, sqs, sqh, skb, sks, skh)
-    switch (flags & 3) {
+    switch (flags & 7) {
       case 0: LAUNCH_V5(0); break;
       case 1: LAUNCH_V5(1); break;
       case 2: LAUNCH_V5(2); break;
-      default: LAUNCH_V5(3); break;
+      case 3: LAUNCH_V5(3); break;
+      case 6: LAUNCH_V5(6); break;  // defer-max + deep prefetch
+      case 7: LAUNCH_V5(7); break;
+      default: LAUNCH_V5(2); break;
     }
     return (int)hipGetLastError();
   }

@@ -321,3 +321,39 @@ def test_adamw_fused_clip_matches_explicit():
                     eps=1e-8, wd=0.0, step=1, gscale=gs)
     # bf16 rounding of the pre-scaled grad differs slightly; compare loosely
     torch.testing.assert_close(p32a.cpu(), p32b.cpu(), atol=2e-4, rtol=1e-3)
+
+
+def test_tuned_linear_matches_reference():
+    """Layout-tuned linear (NT fwd, cached-W^T dX, dY^T-NN dW) must match
+    plain F.linear numerics, including after an in-place weight update
+    (cache invalidation)."""
+    import torch.nn.functional as F
+
+    from prime_amd import ops
+
+    ops.set_linear_tuned(True)
+    try:
+        M, K, N = 4096, 512, 1024
+        x = _bf(torch.randn(M, K)).requires_grad_(True)
+        w = _bf(torch.randn(N, K)).requires_grad_(True)
+        y = ops.tuned_linear(x, w)
+        dy = _bf(torch.randn_like(y.detach()))
+        y.backward(dy)
+        xr = x.detach().clone().requires_grad_(True)
+        wr = w.detach().clone().requires_grad_(True)
+        yr = F.linear(xr, wr)
+        yr.backward(dy)
+        torch.testing.assert_close(y.detach(), yr.detach(), atol=3e-2, rtol=3e-2)
+        torch.testing.assert_close(x.grad, xr.grad, atol=5e-2, rtol=5e-2)
+        torch.testing.assert_close(w.grad, wr.grad, atol=5e-1, rtol=3e-2)
+        # in-place weight update must invalidate the cached W^T
+        with torch.no_grad():
+            w.mul_(2.0)
+        ops.invalidate_wt_cache()
+        x.grad = None
+        ops.tuned_linear(x, w).backward(dy)
+        xr.grad = None
+        F.linear(xr, wr.detach().mul(2.0).requires_grad_(True)).backward(dy)
+        torch.testing.assert_close(x.grad, xr.grad, atol=1e-1, rtol=5e-2)
+    finally:
+        ops.set_linear_tuned(False)
