@@ -86,6 +86,7 @@ def main() -> None:
         torch.cuda.synchronize()
 
     # ---- timed region
+    outer_before = trainer.diloco.outer_step_count
     t0 = time.perf_counter()
     for _ in range(args.steps):
         trainer.train_step()
@@ -93,6 +94,7 @@ def main() -> None:
         torch.cuda.synchronize()
     trainer.mesh.barrier()
     elapsed = time.perf_counter() - t0
+    outer_in_window = trainer.diloco.outer_step_count - outer_before
 
     # max over ranks
     if trainer.mesh.initialized:
@@ -113,9 +115,12 @@ def main() -> None:
     tokens_per_step_per_gpu = args.micro_batch * args.seq_len
     total_tokens = tokens_per_step_per_gpu * args.steps * n_gpus
     tps_inner = total_tokens / elapsed
-    # headline includes the amortized DiLoCo outer sync: steps/H boundaries
-    # would fall inside a window of this many inner steps
-    elapsed_incl = elapsed + t_outer * args.steps / max(1, args.h)
+    # headline includes the amortized DiLoCo outer sync: steps/H
+    # boundaries belong to a window of this many inner steps, minus any
+    # boundary the window ALREADY contained (K >= H runs would otherwise
+    # be charged twice)
+    owed = max(0.0, args.steps / max(1, args.h) - outer_in_window)
+    elapsed_incl = elapsed + t_outer * owed
     tps = total_tokens / elapsed_incl
     tps_gpu = tps / n_gpus
     mfu = mfu_of(tps_gpu, trainer.flops_per_token)
