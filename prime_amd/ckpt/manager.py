@@ -226,3 +226,53 @@ def export_safetensors(ckpt_dir, model_name: str, out_path,
         tensors[n] = flat.flat_w[o : o + k].view(shp).to(torch.bfloat16).contiguous()
     save_file(tensors, str(out_path), metadata={"model": model_name})
     return len(tensors)
+
+
+def import_safetensors(in_path, model_name: str, ckpt_dir,
+                       overrides: dict | None = None,
+                       strict: bool = True) -> int:
+    """Inverse of export_safetensors: pack per-parameter tensors from a
+    safetensors file (e.g. published pretrained weights) into a flat-
+    master checkpoint this engine resumes/serves from. Tensors must match
+    the named config's parameter names/shapes (tied lm_head may be
+    absent). Returns the number of tensors consumed."""
+    from safetensors.torch import load_file
+
+    from ..models import build_model
+    from ..parallel.flat import FlatParamSpace
+
+    tensors = load_file(str(in_path))
+    model = build_model(model_name, **(overrides or {}))
+    flat = FlatParamSpace(model)
+    seen = 0
+    missing = []
+    for n, _ in flat.params:
+        o, k, shp = flat.offsets[n]
+        src = tensors.get(n)
+        if src is None and n == "lm_head.weight" and model.cfg.tie_embeddings:
+            src = tensors.get("tok_embeddings.weight")
+        if src is None:
+            missing.append(n)
+            continue
+        if tuple(src.shape) != tuple(shp):
+            raise ValueError(f"{n}: shape {tuple(src.shape)} != {tuple(shp)}")
+        flat.master32[o : o + k].copy_(src.flatten().float())
+        seen += 1
+    if missing and strict:
+        raise ValueError(f"missing tensors for {len(missing)} params "
+                         f"(first: {missing[:3]}); use strict=False to "
+                         "keep random init for them")
+    flat.flat_w.copy_(flat.master32.to(flat.flat_w.dtype))
+    mgr = CheckpointManager(ckpt_dir, async_save=False)
+    mgr.save(0, {
+        "master32": flat.master32,
+        "theta_outer": flat.master32.clone(),
+        "outer_buf": torch.zeros_like(flat.master32),
+        "adam_m": torch.zeros_like(flat.master32),
+        "adam_v": torch.zeros_like(flat.master32),
+    }, {
+        "inner_step": 0, "outer_step": 0, "adam_step": 0, "step_count": 0,
+        "data_state": {"batch_idx": 0}, "model": model_name,
+        "imported_from": str(in_path),
+    })
+    return seen

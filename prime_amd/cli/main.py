@@ -741,6 +741,26 @@ def export_cmd(
     secho(f"wrote {n} tensors to {out}", fg="green")
 
 
+@app.command("import")
+def import_cmd(
+    weights: str = typer.Argument(..., help="input .safetensors"),
+    ckpt_dir: str = typer.Argument(..., help="output checkpoint dir"),
+    model: str = typer.Option(..., help="model preset the tensors match"),
+    lenient: bool = typer.Option(False, help="keep random init for params "
+                                             "missing from the file"),
+):
+    """Pack pretrained safetensors weights into a resumable/servable
+    checkpoint (inverse of `export`)."""
+    from ..ckpt.manager import import_safetensors
+
+    try:
+        n = import_safetensors(weights, model, ckpt_dir, strict=not lenient)
+    except (ValueError, FileNotFoundError) as e:
+        secho(str(e), fg="red")
+        raise typer.Exit(1)
+    secho(f"imported {n} tensors into {ckpt_dir}", fg="green")
+
+
 # ------------------------------------------------------------------ store
 @app.command("store")
 def store_cmd(port: int = typer.Option(29777), addr: str = typer.Option("0.0.0.0")):

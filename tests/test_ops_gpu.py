@@ -357,3 +357,24 @@ def test_tuned_linear_matches_reference():
         torch.testing.assert_close(x.grad, xr.grad, atol=1e-1, rtol=5e-2)
     finally:
         ops.set_linear_tuned(False)
+
+
+def test_gemm_nt_vs_library():
+    """Custom 256^2-tile MFMA GEMM (the measured in-repo baseline the
+    GEMM study in profiles/ cites): exact-shape NT, bf16, fp32 accum —
+    must match hipBLASLt numerically."""
+    from prime_amd.ops._lib import check, lib, ptr, stream_of
+
+    M, N, K = 512, 768, 256
+    a = _bf(torch.randn(M, K))
+    w = _bf(torch.randn(N, K))
+    c = torch.empty(M, N, device=a.device, dtype=torch.bfloat16)
+    check(lib().prime_gemm_nt(stream_of(a), ptr(a), ptr(w), ptr(c),
+                              M, N, K, 0), "gemm_nt")
+    torch.cuda.synchronize()
+    ref = a @ w.T
+    torch.testing.assert_close(c, ref, atol=3e-2, rtol=3e-2)
+    # unsupported shape must refuse loudly, not corrupt
+    rc = lib().prime_gemm_nt(stream_of(a), ptr(a), ptr(w), ptr(c),
+                             M + 1, N, K, 0)
+    assert rc != 0
