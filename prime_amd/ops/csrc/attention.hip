@@ -958,7 +958,12 @@ __global__ __launch_bounds__(512, 2) void flash_bwd_dkv_v5_kernel(
 // -------------------------------------------------------- backward dK, dV
 // Stages Q[64][D], dO[64][D] (B-operands for St and dPt), Qt[D][64] and
 // dOt[D][64] (B-operands for dK and dV accumulation).
-template <int D, int NW = 4>  // NW waves x 16 k-rows per block
+// DBUF=2: double-buffered q/do/qt/dot stages with prefetch-behind-barrier
+// (one barrier per q-tile instead of the serial stage-between-two-
+// barriers, whose zero-gap vmcnt drain shows as this kernel's 9.0
+// wait/busy in the PMC profile). At NW=8 this is exactly the 160 KiB
+// LDS limit: 2x4x16 KiB stages + 8x2x2 KiB P/dS wave buffers.
+template <int D, int NW = 4, int DBUF = 1>  // NW waves x 16 k-rows
 __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ Qt,
     const bf16* __restrict__ K, const bf16* __restrict__ V,
@@ -987,10 +992,10 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
   const bf16* Kb = K + b * skb + hkv * skh;
   const bf16* Vb = V + b * svb + hkv * svh;
 
-  __shared__ bf16 q_lds[64 * D];
-  __shared__ bf16 do_lds[64 * D];
-  __shared__ bf16 qt_lds[D * 64];
-  __shared__ bf16 dot_lds[D * 64];
+  __shared__ bf16 q_lds[DBUF][64 * D];
+  __shared__ bf16 do_lds[DBUF][64 * D];
+  __shared__ bf16 qt_lds[DBUF][D * 64];
+  __shared__ bf16 dot_lds[DBUF][D * 64];
   __shared__ bf16 pt_lds_all[NW][16 * 64];
   __shared__ bf16 dst_lds_all[NW][16 * 64];
   bf16* pt_lds = pt_lds_all[wid];
@@ -1009,29 +1014,48 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
     dv_acc[dt] = f32x4{0.f, 0.f, 0.f, 0.f};
   }
 
-  for (int g = 0; g < group; ++g) {
+  const int q_start = (causal ? (kt * (NW * 16)) / 64 * 64 : 0) + split * 64;
+  const int tiles_per_g =
+      (S > q_start) ? (S - q_start + 64 * splits - 1) / (64 * splits) : 0;
+  const int n_iter = group * tiles_per_g;
+  const auto stage_it = [&](int it, int buf) {
+    const int g_ = it / tiles_per_g;
+    const int q0g_ = q_start + (it - g_ * tiles_per_g) * 64 * splits;
+    const int h_ = hkv * group + g_;
+    const bf16* Qb_ = Q + b * sqb + h_ * sqh;
+    const bf16* dOb_ = dO + ((int64_t)b * S * H + h_) * D;
+    const bf16* Qtb_ = Qt + ((int64_t)(b * H + h_) * D) * S;
+    const bf16* dOtb_ = dOt + ((int64_t)(b * H + h_) * D) * S;
+    stage_tile<64, D, NW * 64>(Qb_ + (int64_t)q0g_ * sqs, sqs, q_lds[buf], threadIdx.x);
+    stage_tile<64, D, NW * 64>(dOb_ + (int64_t)q0g_ * H * D, (int64_t)H * D, do_lds[buf], threadIdx.x);
+    stage_tile<D, 64, NW * 64>(Qtb_ + q0g_, S, qt_lds[buf], threadIdx.x);
+    stage_tile<D, 64, NW * 64>(dOtb_ + q0g_, S, dot_lds[buf], threadIdx.x);
+  };
+  if (DBUF == 2 && n_iter > 0) stage_it(0, 0);
+  for (int it = 0; it < n_iter; ++it) {
+    const int g = it / tiles_per_g;
+    const int q0g = q_start + (it - g * tiles_per_g) * 64 * splits;
     const int h = hkv * group + g;
-    const bf16* Qb = Q + b * sqb + h * sqh;
-    const bf16* Qtb = Qt + ((int64_t)(b * H + h) * D) * S;
-    const bf16* dOb = dO + ((int64_t)b * S * H + h) * D;
-    const bf16* dOtb = dOt + ((int64_t)(b * H + h) * D) * S;
     const float* lse_b = lse + (int64_t)b * S * H + h;
     const float* dl_b = delta + (int64_t)b * S * H + h;
-    const int q_start = (causal ? (kt * (NW * 16)) / 64 * 64 : 0) + split * 64;
-    for (int q0g = q_start; q0g < S; q0g += 64 * splits) {
+    const int cur = (DBUF == 2) ? (it & 1) : 0;
+    {
       __syncthreads();
-      stage_tile<64, D, NW * 64>(Qb + (int64_t)q0g * sqs, sqs, q_lds, threadIdx.x);
-      stage_tile<64, D, NW * 64>(dOb + (int64_t)q0g * H * D, (int64_t)H * D, do_lds, threadIdx.x);
-      stage_tile<D, 64, NW * 64>(Qtb + q0g, S, qt_lds, threadIdx.x);
-      stage_tile<D, 64, NW * 64>(dOtb + q0g, S, dot_lds, threadIdx.x);
-      __syncthreads();
+      if (DBUF == 2) {
+        // prefetch-behind-barrier: next tile's loads land under this
+        // tile's compute; the next barrier drains them
+        if (it + 1 < n_iter) stage_it(it + 1, cur ^ 1);
+      } else {
+        stage_it(it, 0);
+        __syncthreads();
+      }
 #pragma unroll
       for (int sub = 0; sub < 4; ++sub) {
         f32x4 st_acc{0.f, 0.f, 0.f, 0.f}, dpt_acc{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int ds = 0; ds < DS; ++ds) {
-          const short8 qb = ld8_swz<D>(q_lds, sub * 16 + li, ds * 64 + lg * 16);
-          const short8 dob = ld8_swz<D>(do_lds, sub * 16 + li, ds * 64 + lg * 16);
+          const short8 qb = ld8_swz<D>(q_lds[cur], sub * 16 + li, ds * 64 + lg * 16);
+          const short8 dob = ld8_swz<D>(do_lds[cur], sub * 16 + li, ds * 64 + lg * 16);
           st_acc = mfma16(kf[ds], qb, st_acc);
           dpt_acc = mfma16(vf[ds], dob, dpt_acc);
         }
@@ -1055,8 +1079,8 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
         for (int ks = 0; ks < 2; ++ks) {
           const short8 pa = ld8_swz<64>(pt_lds, li, ks * 64 + lg * 16);
           const short8 da = ld8_swz<64>(dst_lds, li, ks * 64 + lg * 16);
-          dv_acc[dt] = mfma16(pa, ld8_swz<64>(dot_lds, dt * 16 + li, ks * 64 + lg * 16), dv_acc[dt]);
-          dk_acc[dt] = mfma16(da, ld8_swz<64>(qt_lds, dt * 16 + li, ks * 64 + lg * 16), dk_acc[dt]);
+          dv_acc[dt] = mfma16(pa, ld8_swz<64>(dot_lds[cur], dt * 16 + li, ks * 64 + lg * 16), dv_acc[dt]);
+          dk_acc[dt] = mfma16(da, ld8_swz<64>(qt_lds[cur], dt * 16 + li, ks * 64 + lg * 16), dk_acc[dt]);
         }
     }
   }
@@ -1254,8 +1278,8 @@ PRIME_API int prime_flash_bwd_dkv(hipStream_t stream, const void* Q,
   }
   static const char* nw8env = getenv("PRIME_ATTN_DKV8");
   const bool nw8 = (S % 128 == 0) && !(nw8env && nw8env[0] == '0');
-#define LAUNCH_DKV(DD, NWV)                                                  \
-    hipLaunchKernelGGL((flash_bwd_dkv_kernel<DD, NWV>),                      \
+#define LAUNCH_DKV(DD, NWV, DB)                                              \
+    hipLaunchKernelGGL((flash_bwd_dkv_kernel<DD, NWV, DB>),                  \
                        dim3((int)(B * Hkv * (S / (NWV * 16)) * splits)),     \
                        dim3(NWV * 64), 0, stream, (const bf16*)Q,            \
                        (const bf16*)Qt, (const bf16*)K, (const bf16*)V,      \
@@ -1264,10 +1288,14 @@ PRIME_API int prime_flash_bwd_dkv(hipStream_t stream, const void* Q,
                        (float*)wsV, (int)B, (int)H, (int)Hkv, (int)S,        \
                        (float)scale, (int)causal, (int)splits, sqb, sqs,     \
                        sqh, skb, sks, skh, svb, svs, svh)
-  if (nw8) {
-    if (D == 128) LAUNCH_DKV(128, 8); else LAUNCH_DKV(64, 8);
+  static const char* dbenv = getenv("PRIME_ATTN_DKV_DBUF");
+  const bool dbuf = !(dbenv && dbenv[0] == '0');
+  if (nw8 && dbuf) {
+    if (D == 128) LAUNCH_DKV(128, 8, 2); else LAUNCH_DKV(64, 8, 2);
+  } else if (nw8) {
+    if (D == 128) LAUNCH_DKV(128, 8, 1); else LAUNCH_DKV(64, 8, 1);
   } else {
-    if (D == 128) LAUNCH_DKV(128, 4); else LAUNCH_DKV(64, 4);
+    if (D == 128) LAUNCH_DKV(128, 4, 1); else LAUNCH_DKV(64, 4, 1);
   }
   int err = hipGetLastError();
   if (err) return err;
