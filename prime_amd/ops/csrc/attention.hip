@@ -1151,7 +1151,29 @@ PRIME_API int prime_flash_fwd(hipStream_t stream, const void* Q, const void* K,
   // addressing cost.
   const int flags = flagenv ? atoi(flagenv) : 2;
   const bool use_v5 =
-      (D == 128) && (S % 256 == 0) && !(v5env && v5env[0] == '0');
+      (D == 128 || D == 64) && (S % 256 == 0) && !(v5env && v5env[0] == '0');
+  if (use_v5 && D == 64) {
+    const int grid = (int)(B * H * (S / 256));
+    switch (flags & 7) {
+      case 0:
+        hipLaunchKernelGGL((flash_fwd_v5_kernel<64, 0>), dim3(grid),
+                           dim3(512), 0, stream, (const bf16*)Q,
+                           (const bf16*)K, (const bf16*)Vt, (bf16*)O,
+                           (float*)lse, (int)B, (int)H, (int)Hkv, (int)S,
+                           (float)scale, (int)causal, sqb, sqs, sqh, skb,
+                           sks, skh);
+        break;
+      default:
+        hipLaunchKernelGGL((flash_fwd_v5_kernel<64, 2>), dim3(grid),
+                           dim3(512), 0, stream, (const bf16*)Q,
+                           (const bf16*)K, (const bf16*)Vt, (bf16*)O,
+                           (float*)lse, (int)B, (int)H, (int)Hkv, (int)S,
+                           (float)scale, (int)causal, sqb, sqs, sqh, skb,
+                           sks, skh);
+        break;
+    }
+    return (int)hipGetLastError();
+  }
   if (use_v5) {
     const int grid = (int)(B * H * (S / 256));
 #define LAUNCH_V5(F)                                                         \
@@ -1210,6 +1232,17 @@ PRIME_API int prime_flash_bwd_dq(hipStream_t stream, const void* Q,
   static const char* v5env = getenv("PRIME_ATTN_V5");
   static const char* flagenv = getenv("PRIME_ATTN_FLAGS");
   const int flags = flagenv ? atoi(flagenv) : 2;
+  if ((D == 64) && (S % 256 == 0) && !(v5env && v5env[0] == '0')) {
+    const int grid = (int)(B * H * (S / 256));
+    hipLaunchKernelGGL((flash_bwd_dq_v5_kernel<64, 0>), dim3(grid),
+                       dim3(512), 0, stream, (const bf16*)Q, (const bf16*)K,
+                       (const bf16*)V, (const bf16*)Kt, (const bf16*)dO,
+                       (const float*)lse, (const float*)delta, (bf16*)dQ,
+                       (int)B, (int)H, (int)Hkv, (int)S, (float)scale,
+                       (int)causal, sqb, sqs, sqh, skb, sks, skh,
+                       svb, svs, svh);
+    return (int)hipGetLastError();
+  }
   if ((D == 128) && (S % 256 == 0) && !(v5env && v5env[0] == '0')) {
     const int grid = (int)(B * H * (S / 256));
 #define LAUNCH_DQ5(F)                                                       \
