@@ -32,9 +32,9 @@ class _PinnedStager:
 
     def __init__(self):
         self.buffers: dict[str, torch.Tensor] = {}
+        self._pinned_bytes = 0  # cumulative: per-tensor checks overcommit
 
-    @staticmethod
-    def _alloc(t: torch.Tensor) -> torch.Tensor:
+    def _alloc(self, t: torch.Tensor) -> torch.Tensor:
         try:
             avail = None
             for line in open("/proc/meminfo"):
@@ -42,8 +42,10 @@ class _PinnedStager:
                     avail = int(line.split()[1]) * 1024
                     break
             need = t.numel() * t.element_size()
-            if avail is None or need < 0.3 * avail:
-                return torch.empty_like(t, device="cpu", pin_memory=True)
+            if avail is None or self._pinned_bytes + need < 0.3 * avail:
+                buf = torch.empty_like(t, device="cpu", pin_memory=True)
+                self._pinned_bytes += need
+                return buf
         except (OSError, RuntimeError):
             pass
         return torch.empty_like(t, device="cpu")
@@ -93,6 +95,52 @@ class CheckpointManager:
         self.root.mkdir(parents=True, exist_ok=True)
 
     # ------------------------------------------------------------- save
+    @staticmethod
+    def _fs_is_ram(path: Path) -> bool:
+        try:
+            rp = os.path.realpath(path)
+            best, best_t = "", ""
+            for line in open("/proc/mounts"):
+                parts = line.split()
+                if len(parts) >= 3 and (rp == parts[1] or
+                                        rp.startswith(parts[1].rstrip("/") + "/")):
+                    if len(parts[1]) > len(best):
+                        best, best_t = parts[1], parts[2]
+            return best_t in ("tmpfs", "ramfs")
+        except OSError:
+            return False
+
+    def _check_capacity(self, tag_dir: Path, need: int) -> None:
+        """Refuse loudly instead of killing the node: a 10B checkpoint is
+        ~200 GB — written to a tmpfs path it silently eats host RAM until
+        the OOM killer takes the machine down (observed on a soak run)."""
+        try:
+            free = shutil.disk_usage(tag_dir).free
+        except OSError:
+            return
+        kind = "tmpfs (RAM-backed!)" if self._fs_is_ram(tag_dir) else "disk"
+        if need * 1.05 > free:
+            raise RuntimeError(
+                f"checkpoint needs ~{need / 2**30:.1f} GiB but {tag_dir} "
+                f"({kind}) has {free / 2**30:.1f} GiB free — refusing to "
+                "write; point checkpoint.path at a filesystem with room"
+            )
+        if self._fs_is_ram(tag_dir):
+            avail = None
+            try:
+                for line in open("/proc/meminfo"):
+                    if line.startswith("MemAvailable:"):
+                        avail = int(line.split()[1]) * 1024
+                        break
+            except OSError:
+                pass
+            if avail is not None and need * 1.2 > avail:
+                raise RuntimeError(
+                    f"checkpoint path {tag_dir} is RAM-backed (tmpfs) and "
+                    f"~{need / 2**30:.1f} GiB would exhaust MemAvailable "
+                    f"({avail / 2**30:.1f} GiB) — refusing; use a disk path"
+                )
+
     def save(self, outer_step: int, tensors: dict[str, torch.Tensor],
              meta: dict) -> None:
         """Snapshot `tensors` (+ JSON-serializable `meta`)."""
@@ -100,6 +148,8 @@ class CheckpointManager:
         tag_dir = self.root / f"step_{outer_step}"
         tag_dir.mkdir(parents=True, exist_ok=True)
         fname = tag_dir / self._fname()
+        self._check_capacity(
+            tag_dir, sum(t.numel() * t.element_size() for t in tensors.values()))
 
         on_gpu = any(t.is_cuda for t in tensors.values())
         if on_gpu and self.async_save:

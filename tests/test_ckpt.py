@@ -128,3 +128,21 @@ def test_import_export_roundtrip(tmp_path):
     save_file(t, str(partial))
     with pytest.raises(ValueError, match="missing"):
         import_safetensors(partial, "llama_test", tmp_path / "ck3")
+
+
+def test_capacity_guard_refuses_oversized(tmp_path, monkeypatch):
+    """A checkpoint bigger than the target filesystem must refuse loudly
+    (a tmpfs target otherwise eats host RAM until the OOM killer takes
+    the node - observed on a GPU soak run)."""
+    import shutil as _shutil
+
+    import pytest
+
+    mgr = CheckpointManager(tmp_path / "ck", async_save=False)
+
+    class FakeUsage:
+        free = 1 << 20  # 1 MiB free
+
+    monkeypatch.setattr(_shutil, "disk_usage", lambda p: FakeUsage)
+    with pytest.raises(RuntimeError, match="refusing"):
+        mgr.save(1, {"x": torch.zeros(1 << 22)}, {})  # 16 MiB > 1 MiB free
