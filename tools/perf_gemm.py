@@ -67,17 +67,18 @@ if __name__ == "__main__" and "--custom" not in sys.argv:
     main()
 
 
-def bench_custom_nt(name, m, n, k, iters=20):
+def bench_custom_nt(name, m, n, k, iters=20, v2=False, setprio=0):
     """Custom gemm_nt kernel vs hipBLASLt on the same NT shape."""
     from prime_amd.ops._lib import check, lib, ptr, stream_of
 
     a = torch.randn(m, k, device="cuda:0").bfloat16()
     w = torch.randn(n, k, device="cuda:0").bfloat16()
     c = torch.empty(m, n, device="cuda:0", dtype=torch.bfloat16)
+    fn = lib().prime_gemm_nt8 if v2 else lib().prime_gemm_nt
 
     def custom():
-        check(lib().prime_gemm_nt(stream_of(a), ptr(a), ptr(w), ptr(c),
-                                  m, n, k, 0), "gemm_nt")
+        check(fn(stream_of(a), ptr(a), ptr(w), ptr(c), m, n, k, setprio),
+              "gemm_nt")
 
     # refcheck vs library
     custom()
@@ -110,6 +111,16 @@ def bench_custom_nt(name, m, n, k, iters=20):
 def main_custom():
     torch.manual_seed(1)
     M, dim, inter, vocab = 16384, 4096, 14336, 128256
+    v2 = "--v2" in sys.argv
+    sp = 1 if "--setprio" in sys.argv else 0
+    if v2:
+        print(f"== custom NT v2 (counted-vmcnt, setprio={sp}) vs hipBLASLt ==")
+        for name, m, n, k in [("qkv fwd", M, 6144, dim),
+                              ("gateup fwd", M, 2 * inter, dim),
+                              ("down fwd", M, dim, inter),
+                              ("square 4k", 4096, 4096, 4096)]:
+            bench_custom_nt(name, m, n, k, v2=True, setprio=sp)
+        return
     print("== custom NT kernel vs hipBLASLt ==")
     bench_custom_nt("qkv fwd", M, 6144, dim)
     bench_custom_nt("gateup fwd", M, 2 * inter, dim)
