@@ -770,13 +770,18 @@ def store_cmd(port: int = typer.Option(29777), addr: str = typer.Option("0.0.0.0
 
     from torch.distributed import TCPStore
 
-    TCPStore(addr, port, is_master=True, wait_for_workers=False)
+    # the reference MUST be held: an unbound TCPStore master is garbage-
+    # collected immediately and the listening socket closes (caught by
+    # tests/test_elastic_fsdp.py::test_store_host_failover)
+    store = TCPStore(addr, port, is_master=True, wait_for_workers=False)
     secho(f"elastic store listening on {addr}:{port} (ctrl-c to stop)", fg="green")
     try:
         while True:
             _time.sleep(3600)
     except KeyboardInterrupt:
         pass
+    finally:
+        del store
 
 
 # ------------------------------------------------------------------ serve

@@ -141,7 +141,12 @@ class ElasticWorker:
     def members(self) -> list[str]:
         """Sorted registry (join order == lexicographic by construction)."""
         # TCPStore has no key listing; keep a mirror index
-        idx = self._index_cas(lambda i: sorted(set(i) | {self.wid}))
+        try:
+            idx = self._index_cas(lambda i: sorted(set(i) | {self.wid}))
+        except Exception:  # noqa: BLE001 — store host gone: reconnect once
+            if not self.reconnect():
+                raise
+            idx = self._index_cas(lambda i: sorted(set(i) | {self.wid}))
         live = [w for w in idx if self.store.check([f"members/{w}"])]
         if live != idx:
             dead = set(idx) - set(live)
@@ -149,14 +154,46 @@ class ElasticWorker:
             live = [w for w in live if self.store.check([f"members/{w}"])]
         return live
 
+    # ----------------------------------------------------------- reconnect
+    def reconnect(self, attempts: int = 5, backoff: float = 2.0) -> bool:
+        """The TCPStore host went away (e.g. a standalone `prime-amd
+        store` was restarted): build a fresh client against the same
+        address and re-register under a new wid. The restarted registry
+        starts empty; membership reconverges at the next boundary as
+        every surviving worker re-registers."""
+        import time as _time
+
+        for i in range(attempts):
+            try:
+                store = TCPStore(self.addr, self.port, is_master=False,
+                                 wait_for_workers=False,
+                                 timeout=timedelta(seconds=10))
+                seq = int(store.add("join_seq", 1))
+                self.store = store
+                self.join_seq = seq
+                self.wid = f"{seq:06d}-re-{self.wid.split('-', 1)[1]}"
+                self._peer_tracks.clear()
+                self.epoch = int(self._get_str("epoch", "0"))
+                self._register()
+                if self._ckpt_srv is not None:
+                    host, port = self._ckpt_srv.getsockname()
+                    self.store.set(f"ckptsrv/{self.wid}", f"{host}:{port}")
+                    self.store.set(f"ckptsrv/{self.wid}/r0", f"{host}:{port}")
+                return True
+            except Exception:  # noqa: BLE001 — host still down; back off
+                _time.sleep(backoff * (i + 1))
+        return False
+
     # ------------------------------------------------------------ heartbeat
     def _hb_loop(self) -> None:
         while not self._stop.wait(self.hb_interval):
             self._hb_seq += 1
             try:
                 self.store.set(f"hb/{self.wid}", str(self._hb_seq))
-            except Exception:  # noqa: BLE001 — store host gone; trainer will notice
-                return
+            except Exception:  # noqa: BLE001 — store host gone; the sync
+                # path owns reconnection; just keep trying the (possibly
+                # replaced) store handle
+                continue
 
     def _is_stale(self, wid: str) -> bool:
         try:
@@ -191,24 +228,29 @@ class ElasticWorker:
             try:
                 self.store.set(f"ready/{n}/{self.wid}",
                                f"{max(0.0, contribution):.6f}")
+                view_key = f"view/{n}"
+                t0 = _now()
+                members: list[str] = []
+                while True:
+                    if self.store.check([view_key]):
+                        members = json.loads(self.store.get(view_key).decode())
+                        break
+                    if self._try_arbiter(n, t0):
+                        members = json.loads(self.store.get(view_key).decode())
+                        break
+                    time.sleep(0.05)
             except Exception as e:  # noqa: BLE001
+                # store host gone (possibly mid-wait): reconnect to a
+                # restarted `prime-amd store` and redo the boundary from
+                # the new registry's epoch
+                if self.reconnect():
+                    continue
                 raise RuntimeError(
-                    "elastic store unreachable — the TCPStore host is gone. "
-                    "Run a standalone registry (`prime-amd store`) so worker "
-                    "churn cannot take the store down, then restart workers "
-                    f"against it. ({e})"
+                    "elastic store unreachable and reconnect failed — "
+                    "run a standalone registry (`prime-amd store`) so "
+                    "worker churn cannot take the store down, then "
+                    f"restart workers against it. ({e})"
                 ) from e
-            view_key = f"view/{n}"
-            t0 = _now()
-            members: list[str] = []
-            while True:
-                if self.store.check([view_key]):
-                    members = json.loads(self.store.get(view_key).decode())
-                    break
-                if self._try_arbiter(n, t0):
-                    members = json.loads(self.store.get(view_key).decode())
-                    break
-                time.sleep(0.05)
             self.epoch = n
             self.store.set("epoch", str(n))
             if self.wid not in members:

@@ -242,3 +242,93 @@ def test_concurrent_join_stress():
     out = _spawn(jobs, timeout=120)
     for i in range(8):
         assert out[i]["n"] == 8
+
+
+# --------------------------------------------- store-host failover
+def _store_proc(port, ready_evt, stop_evt):
+    from torch.distributed import TCPStore
+
+    store = TCPStore("127.0.0.1", port, is_master=True, wait_for_workers=False)
+    ready_evt.set()
+    stop_evt.wait(120)
+    del store
+
+
+def _failover_worker(gport, idx):
+    from prime_amd.parallel.elastic import ElasticWorker
+
+    w = ElasticWorker(port=gport, host_store=False, worker_name=f"f{idx}",
+                      heartbeat_interval=0.3, heartbeat_timeout=10.0)
+    while len(w.members()) < 2:
+        time.sleep(0.05)
+    v1 = w.sync()
+    # coordinate the store restart through files (the store dies)
+    import pathlib
+
+    flagdir = pathlib.Path("/tmp/prime_amd_test/failover")
+    flagdir.mkdir(parents=True, exist_ok=True)
+    (flagdir / f"synced{idx}").write_text("1")
+    while not (flagdir / "store_restarted").exists():
+        time.sleep(0.1)
+    # next boundary: the old store is gone -> reconnect path
+    t0 = time.monotonic()
+    while len(w.members()) < 2:  # both re-registered on the new store
+        if time.monotonic() - t0 > 60:
+            raise AssertionError("peers did not reconverge")
+        time.sleep(0.05)
+    v2 = w.sync()
+    w.close(leaving=False)
+    return {"w1": v1.world, "w2": v2.world, "rewid": w.wid.split("-")[1]}
+
+
+def test_store_host_failover():
+    """Kill the standalone registry mid-run, restart it empty on the same
+    port: workers reconnect, re-register and reconverge to world=2."""
+    import multiprocessing as mp
+    import pathlib
+    import shutil
+
+    shutil.rmtree("/tmp/prime_amd_test/failover", ignore_errors=True)
+    gport = free_port()
+    ctx = mp.get_context("spawn")
+    ready, stop = ctx.Event(), ctx.Event()
+    store_p = ctx.Process(target=_store_proc, args=(gport, ready, stop))
+    store_p.start()
+    assert ready.wait(30)
+
+    q = ctx.Queue()
+    procs = []
+    for i in range(2):
+        pr = ctx.Process(target=_entry, args=(_failover_worker, i, {}, (gport, i), q))
+        pr.start()
+        procs.append(pr)
+    flagdir = pathlib.Path("/tmp/prime_amd_test/failover")
+    t0 = time.monotonic()
+    while not all((flagdir / f"synced{i}").exists() for i in range(2)):
+        assert time.monotonic() - t0 < 60
+        time.sleep(0.1)
+    store_p.terminate()
+    store_p.join(10)
+    ready2, stop2 = ctx.Event(), ctx.Event()
+    store_p2 = ctx.Process(target=_store_proc, args=(gport, ready2, stop2))
+    store_p2.start()
+    assert ready2.wait(30)
+    (flagdir / "store_restarted").write_text("1")
+
+    results, errs = {}, []
+    for _ in range(2):
+        kind, i, payload = q.get(timeout=180)
+        if kind == "ok":
+            results[i] = payload
+        else:
+            errs.append((i, payload))
+    for pr in procs:
+        pr.join(timeout=30)
+        if pr.is_alive():
+            pr.terminate()
+    stop2.set()
+    store_p2.terminate()
+    assert not errs, errs
+    for i in (0, 1):
+        assert results[i]["w1"] == 2 and results[i]["w2"] == 2
+        assert results[i]["rewid"] == "re"  # went through reconnect()
