@@ -784,6 +784,71 @@ def store_cmd(port: int = typer.Option(29777), addr: str = typer.Option("0.0.0.0
         del store
 
 
+# ----------------------------------------------------------------- doctor
+@app.command("doctor")
+def doctor(json_out: bool = typer.Option(False, "--json")):
+    """Validate the environment (reference: `prime lab doctor`): toolchain,
+    kernel library freshness, GPU/RCCL readiness, elastic store reach."""
+    import shutil as _shutil
+
+    checks = []
+
+    def add(name, ok, detail):
+        checks.append({"check": name, "ok": bool(ok), "detail": detail})
+
+    hipcc = _shutil.which("hipcc")
+    add("hipcc", hipcc is not None, hipcc or "not on PATH (kernel rebuilds impossible)")
+    try:
+        from ..ops.build import LIB_PATH, needs_build
+
+        stale = needs_build()
+        add("kernel .so", LIB_PATH.exists() and not stale,
+            f"{LIB_PATH} " + ("fresh (source-hash match)" if not stale
+                              else "missing/stale - will rebuild on first use"))
+    except Exception as e:  # noqa: BLE001
+        add("kernel .so", False, str(e))
+    try:
+        import torch
+
+        add("torch", True, f"{torch.__version__} (HIP {torch.version.hip})")
+        cuda = torch.cuda.is_available()
+        detail = "no GPU visible (CPU/gloo plumbing only)"
+        if cuda:
+            prop = torch.cuda.get_device_properties(0)
+            detail = (f"{torch.cuda.device_count()}x {prop.name} "
+                      f"({prop.total_memory / 2**30:.0f} GiB, {prop.gcnArchName})")
+            ok_arch = "gfx950" in prop.gcnArchName
+            add("gfx950", ok_arch, prop.gcnArchName)
+        add("gpu", cuda, detail)
+        import torch.distributed as dist
+
+        add("rccl backend", dist.is_nccl_available(),
+            "torch.distributed nccl(=RCCL) " +
+            ("available" if dist.is_nccl_available() else "MISSING"))
+        add("ipc mode", os.environ.get("HSA_ENABLE_IPC_MODE_LEGACY") == "0",
+            "HSA_ENABLE_IPC_MODE_LEGACY=" +
+            os.environ.get("HSA_ENABLE_IPC_MODE_LEGACY", "<unset>") +
+            " (0 required for multi-process GPU work on dmabuf-only hosts)")
+    except Exception as e:  # noqa: BLE001
+        add("torch", False, str(e))
+    addr = os.environ.get("PRIME_GLOBAL_ADDR")
+    if addr:
+        port = int(os.environ.get("PRIME_GLOBAL_PORT", 29777))
+        try:
+            with socket.create_connection((addr, port), timeout=3):
+                add("elastic store", True, f"{addr}:{port} reachable")
+        except OSError as e:
+            add("elastic store", False, f"{addr}:{port} unreachable ({e})")
+    if json_out:
+        emit_json(checks)
+    else:
+        for c in checks:
+            mark, color = ("ok", "green") if c["ok"] else ("!!", "yellow")
+            secho(f"  [{mark}] {c['check']:14s} {c['detail']}", fg=color)
+    if not all(c["ok"] for c in checks if c["check"] in ("hipcc", "torch")):
+        raise typer.Exit(1)
+
+
 # ------------------------------------------------------------------ serve
 @app.command("serve")
 def serve_cmd(

@@ -290,3 +290,19 @@ def test_usage_and_top(tmp_path, monkeypatch):
     r = CliRunner().invoke(app, ["--plain", "top", "--once"])
     assert r.exit_code == 0, r.output
     assert "urun-xyz" in r.output
+
+
+def test_doctor(tmp_path, monkeypatch):
+    import json as _json
+
+    from typer.testing import CliRunner
+
+    from prime_amd.cli.main import app
+
+    r = CliRunner().invoke(app, ["doctor", "--json"])
+    assert r.exit_code == 0, r.output
+    checks = {c["check"]: c for c in _json.loads(r.output)}
+    assert checks["hipcc"]["ok"]          # build image ships hipcc
+    assert checks["torch"]["ok"]
+    assert checks["kernel .so"]["ok"]     # fresh source-hash stamp
+    assert not checks["gpu"]["ok"]        # no GPU in the build container
