@@ -32,6 +32,9 @@ def main() -> None:
     ap.add_argument("--micro-batch", type=int, default=8)
     ap.add_argument("--h", type=int, default=100, help="DiLoCo inner steps per outer sync")
     ap.add_argument("--no-outer-warmup", action="store_true")
+    ap.add_argument("--fp8", action="store_true",
+                    help="opt-in fp8 forward linears (NOT the headline "
+                         "contract dtype; reported as bf16+fp8fwd)")
     ap.add_argument("--ckpt", choices=["on", "off"], default="off",
                     help="activation checkpointing (off: 288 GB HBM fits the "
                          "10B config's activations; recompute costs ~25%%)")
@@ -56,6 +59,7 @@ def main() -> None:
         model=ModelConfig(
             name=args.model, seq_len=args.seq_len,
             activation_checkpointing=(args.ckpt == "on"),
+            fp8=args.fp8,
         ),
         data=DataSection(kind="synthetic", micro_batch_size=args.micro_batch),
         diloco=DilocoConfig(H=args.h, quant_int8=True, outer_device="auto"),
@@ -139,7 +143,7 @@ def main() -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": "bf16+fp8fwd" if args.fp8 else "bf16",
             "data": "synthetic",
             "mfu": mfu,
             # BASELINE.md convention: record the peak used for MFU (DENSE

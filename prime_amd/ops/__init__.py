@@ -8,6 +8,7 @@ torch.nn.functional.linear (library GEMMs; everything fused is ours).
 from .functional import (
     apply_rope,
     invalidate_wt_cache,
+    set_linear_fp8,
     set_linear_tuned,
     tuned_linear,
     attn_decode,

@@ -127,6 +127,7 @@ class Trainer:
         from . import ops as _ops
 
         _ops.set_linear_tuned(self.device.type == "cuda" and not self.fsdp)
+        _ops.set_linear_fp8(cfg.model.fp8 and self.device.type == "cuda")
         self.inner = FusedAdamW(
             self.flat, lr=cfg.optim.lr, betas=tuple(cfg.optim.betas),
             eps=cfg.optim.eps, weight_decay=cfg.optim.weight_decay,

@@ -378,3 +378,32 @@ def test_gemm_nt_vs_library():
     rc = lib().prime_gemm_nt(stream_of(a), ptr(a), ptr(w), ptr(c),
                              M + 1, N, K, 0)
     assert rc != 0
+
+
+def test_fp8_linear_matches_bf16_loosely():
+    """Opt-in fp8 forward: same math as the bf16 linear within fp8's
+    dynamic-scaled precision; backward stays the bf16 layout-tuned path
+    and must match tightly."""
+    import torch.nn.functional as F
+
+    from prime_amd import ops
+
+    ops.set_linear_fp8(True)
+    try:
+        M, K, N = 4096, 512, 1024
+        x = _bf(0.5 * torch.randn(M, K)).requires_grad_(True)
+        w = _bf(0.5 * torch.randn(N, K)).requires_grad_(True)
+        y = ops.tuned_linear(x, w)
+        yr = F.linear(x.detach(), w.detach())
+        # fp8 e4m3: ~2 decimal digits; compare relative to the output scale
+        rel = (y.detach() - yr).float().norm() / yr.float().norm()
+        assert float(rel) < 0.06, float(rel)  # e4m3 per-tensor-scale noise
+        dy = _bf(torch.randn_like(y.detach()))
+        y.backward(dy)
+        xr = x.detach().clone().requires_grad_(True)
+        wr = w.detach().clone().requires_grad_(True)
+        F.linear(xr, wr).backward(dy)
+        torch.testing.assert_close(x.grad, xr.grad, atol=5e-2, rtol=5e-2)
+        torch.testing.assert_close(w.grad, wr.grad, atol=5e-1, rtol=5e-2)
+    finally:
+        ops.set_linear_fp8(False)

@@ -2,6 +2,7 @@
 bigram corpus (learnable structure, unlike uniform random) and log the loss
 curve to gpurun_out/."""
 import json
+import os
 import sys
 
 import numpy as np
@@ -31,6 +32,7 @@ toks.tofile("/tmp/zipf.bin")
 cfg = TrainConfig(
     run_name="zipf150m", steps=300,
     model=ModelConfig(name="llama_150m", seq_len=512,
+                      fp8=os.environ.get("PRIME_AMD_FP8", "0") == "1",
                       overrides={"vocab_size": V, "max_seq": 1024}),
     data=DataSection(kind="token_file", path="/tmp/zipf.bin",
                      micro_batch_size=16),
