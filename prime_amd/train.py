@@ -37,6 +37,13 @@ def lr_at(step: int, cfg) -> float:
 class Trainer:
     def __init__(self, cfg: TrainConfig, run_dir: str | Path | None = None):
         self.cfg = cfg
+        # fp8's weight cache is (data_ptr, shape)-keyed like the W^T
+        # cache: FSDP's rotating gather pools would alias it (same pool
+        # pointer and shape for different layers within one step)
+        if cfg.model.fp8 and cfg.parallel.fsdp and cfg.parallel.worker_size > 1:
+            raise ValueError("model.fp8 is not supported with parallel.fsdp "
+                             "yet (per-step weight caches need stable "
+                             "parameter storage)")
         self.stop_requested = False
         self.run_dir = Path(run_dir) if run_dir else Path("runs") / cfg.run_name
         self.run_dir.mkdir(parents=True, exist_ok=True)

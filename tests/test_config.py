@@ -53,3 +53,20 @@ def test_shipped_configs_load():
     for f in files:
         cfg = load_config(f)
         assert cfg.steps > 0, f
+
+
+def test_fp8_fsdp_rejected(tmp_path):
+    from prime_amd.train import Trainer
+    from prime_amd.utils.config import (ModelConfig, ParallelConfig,
+                                        TrainConfig)
+
+    import pytest
+
+    cfg = TrainConfig(
+        run_name="bad",
+        model=ModelConfig(name="llama_test", fp8=True,
+                          activation_checkpointing=True),
+        parallel=ParallelConfig(fsdp=True, worker_size=2),
+    )
+    with pytest.raises(ValueError, match="fp8.*fsdp|fsdp.*fp8"):
+        Trainer(cfg, run_dir=tmp_path)
