@@ -32,6 +32,10 @@ def main() -> None:
     ap.add_argument("--micro-batch", type=int, default=8)
     ap.add_argument("--h", type=int, default=100, help="DiLoCo inner steps per outer sync")
     ap.add_argument("--no-outer-warmup", action="store_true")
+    ap.add_argument("--fp8-dgrad", action="store_true",
+                    help="with --fp8: dX GEMMs in e5m2 too")
+    ap.add_argument("--fp8-wgrad", action="store_true",
+                    help="with --fp8 --fp8-dgrad: dW GEMMs in fp8 too")
     ap.add_argument("--fp8", action="store_true",
                     help="opt-in fp8 forward linears (NOT the headline "
                          "contract dtype; reported as bf16+fp8fwd)")
@@ -59,7 +63,7 @@ def main() -> None:
         model=ModelConfig(
             name=args.model, seq_len=args.seq_len,
             activation_checkpointing=(args.ckpt == "on"),
-            fp8=args.fp8,
+            fp8=args.fp8, fp8_dgrad=args.fp8_dgrad, fp8_wgrad=args.fp8_wgrad,
         ),
         data=DataSection(kind="synthetic", micro_batch_size=args.micro_batch),
         diloco=DilocoConfig(H=args.h, quant_int8=True, outer_device="auto"),

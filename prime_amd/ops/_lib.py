@@ -43,6 +43,7 @@ _SIGS = {
     "prime_attn_decode": [ctypes.c_void_p] * 5 + [ctypes.c_int64] * 6 + [ctypes.c_double, ctypes.c_void_p],
     "prime_gemm_nt": [ctypes.c_void_p] * 4 + [ctypes.c_int64] * 4,
     "prime_gemm_nt8": [ctypes.c_void_p] * 4 + [ctypes.c_int64] * 4,
+    "prime_quant_fp8": [ctypes.c_void_p] * 6 + [ctypes.c_int64] * 2,
 }
 
 

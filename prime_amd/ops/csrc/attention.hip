@@ -1049,6 +1049,10 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
         stage_it(it, 0);
         __syncthreads();
       }
+      // per-wave causal clip: this wave's k-rows start at k0, so q-tiles
+      // entirely below the diagonal are fully masked — skip their
+      // compute (the wave still co-staged and hits the next barrier)
+      if (causal && q0g + 64 <= k0) continue;
 #pragma unroll
       for (int sub = 0; sub < 4; ++sub) {
         f32x4 st_acc{0.f, 0.f, 0.f, 0.f}, dpt_acc{0.f, 0.f, 0.f, 0.f};

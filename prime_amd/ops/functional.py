@@ -393,7 +393,15 @@ def _linear_backward(x2, w, dy):
     dy2 = dy.reshape(-1, N).contiguous()
     # dX as NT against the per-step transposed weight
     wt = _wt_of(w) if _LINEAR_TUNED else None
-    if wt is not None:
+    if (_LINEAR_FP8_DGRAD and wt is not None and M % 16 == 0
+            and dy2.dtype == torch.bfloat16):
+        # fp8 dgrad: e5m2 gradients (grad dynamic range) x e4m3 weights
+        dy8, dyinv = _quant_act_fp8(dy2, ("dy", w.data_ptr(), *w.shape),
+                                    e5m2=True)
+        wt8, wtinv = _w8_of(wt)
+        dx = torch._scaled_mm(dy8, wt8.t(), scale_a=dyinv, scale_b=wtinv,
+                              out_dtype=dy2.dtype)
+    elif wt is not None:
         dx = torch.matmul(dy2, wt.t())
     else:
         dx = torch.matmul(dy2, w)
@@ -402,7 +410,21 @@ def _linear_backward(x2, w, dy):
     # worth a per-shape table)
     if N % 128 == 0 and M % 64 == 0 and M >= 4096:
         dyt = transpose_bshd(dy2.view(1, M, N // 128, 128)).view(N, M)
-        dw = torch.mm(dyt, x2)
+        if (_LINEAR_FP8_WGRAD and K % 128 == 0 and M % 16 == 0
+                and x2.dtype == torch.bfloat16):
+            # fp8 wgrad (the full TE-style recipe, opt-in behind its own
+            # flag): dW^T = (x^T)_e4m3 @ (dY^T)^T_e5m2, then the LDS-tiled
+            # transpose back to [N,K]. The fp32 master still integrates
+            # every update; the convergence gate is the Zipf-150m run.
+            xt = transpose_bshd(x2.view(1, M, K // 128, 128)).view(K, M)
+            xt8, xtinv = _quant_act_fp8(xt, ("xT", w.data_ptr(), *w.shape))
+            dyt8, dytinv = _quant_act_fp8(
+                dyt, ("dyT", w.data_ptr(), *w.shape), e5m2=True)
+            g = torch._scaled_mm(xt8, dyt8.t(), scale_a=xtinv,
+                                 scale_b=dytinv, out_dtype=x2.dtype)
+            dw = transpose_bshd(g.view(1, K, N // 128, 128)).view(N, K)
+        else:
+            dw = torch.mm(dyt, x2)
     else:
         dw = torch.mm(dy2.t(), x2)
     return dx.view(*dy.shape[:-1], K), dw
@@ -434,10 +456,17 @@ _LINEAR_FP8 = False
 _FP8_MAX = 448.0  # OCP e4m3 finite max
 
 
-def set_linear_fp8(on: bool) -> None:
-    global _LINEAR_FP8
+_LINEAR_FP8_DGRAD = False
+_LINEAR_FP8_WGRAD = False
+
+
+def set_linear_fp8(on: bool, dgrad: bool = False, wgrad: bool = False) -> None:
+    global _LINEAR_FP8, _LINEAR_FP8_DGRAD, _LINEAR_FP8_WGRAD
     _LINEAR_FP8 = bool(on)
+    _LINEAR_FP8_DGRAD = bool(on and dgrad)
+    _LINEAR_FP8_WGRAD = bool(on and dgrad and wgrad)
     _FP8_CACHE.clear()
+    _FP8_ACT.clear()
 
 
 def _w8_of(w: torch.Tensor):
@@ -448,20 +477,53 @@ def _w8_of(w: torch.Tensor):
     amax = w.abs().amax().float().clamp(min=1e-12)
     scale = _FP8_MAX / amax
     w8 = (w * scale.to(w.dtype)).to(torch.float8_e4m3fn)
-    sinv = 1.0 / scale  # dequant scale for _scaled_mm
+    sinv = (1.0 / scale).reshape(1).contiguous()  # dequant scale (device)
     _FP8_CACHE[key] = (_WT_EPOCH, w8, sinv)
     return w8, sinv
+
+
+# per-site delayed-scaling state for activations: the fused quant kernel
+# (csrc/quant_fp8.hip) casts with the PREVIOUS step's amax in ONE pass
+# (outliers saturate for one step, TransformerEngine-style) and records
+# this step's amax for the next — vs the 3-pass amax/mul/cast torch path
+_FP8_ACT: dict = {}
+
+
+def _quant_act_fp8(x2: torch.Tensor, site: tuple, e5m2: bool = False):
+    dt8 = torch.float8_e5m2 if e5m2 else torch.float8_e4m3fn
+    fmax8 = 57344.0 if e5m2 else _FP8_MAX
+    st = _FP8_ACT.get(site)
+    if st is None:
+        # bootstrap: dynamic 2-pass scaling, seed the amax state
+        ax = x2.abs().amax().float().clamp(min=1e-12)
+        sx = fmax8 / ax
+        x8 = (x2 * sx.to(x2.dtype)).to(dt8)
+        _FP8_ACT[site] = {
+            "amax": ax.reshape(1).contiguous(),
+            "next": torch.zeros(1, device=x2.device),
+            "sinv": (1.0 / sx).reshape(1).contiguous(),
+        }
+        return x8, _FP8_ACT[site]["sinv"]
+    x8 = torch.empty(x2.shape, device=x2.device, dtype=dt8)
+    st["next"].zero_()
+    check(
+        lib().prime_quant_fp8(
+            stream_of(x2), ptr(x2), ptr(x8), ptr(st["amax"]), ptr(st["next"]),
+            ptr(st["sinv"]), x2.numel(), 1 if e5m2 else 0,
+        ),
+        "quant_fp8",
+    )
+    st["amax"], st["next"] = st["next"], st["amax"]
+    return x8, st["sinv"]
 
 
 class _Fp8Linear(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w):
-        x2 = x.reshape(-1, x.shape[-1])
-        ax = x2.abs().amax().float().clamp(min=1e-12)
-        sx = _FP8_MAX / ax
-        x8 = (x2 * sx.to(x2.dtype)).to(torch.float8_e4m3fn)
+        x2 = x.reshape(-1, x.shape[-1]).contiguous()
+        x8, sxinv = _quant_act_fp8(x2, (w.data_ptr(), *w.shape))
         w8, swinv = _w8_of(w)
-        y = torch._scaled_mm(x8, w8.t(), scale_a=1.0 / sx, scale_b=swinv,
+        y = torch._scaled_mm(x8, w8.t(), scale_a=sxinv, scale_b=swinv,
                              out_dtype=x.dtype)
         ctx.save_for_backward(x2, w)
         return y.view(*x.shape[:-1], w.shape[0])

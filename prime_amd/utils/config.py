@@ -27,6 +27,12 @@ class ModelConfig(_Strict):
     # dynamic scaling; master/grads/backward stay bf16/fp32) — ~1.8x the
     # bf16 GEMM rate on MI355X. The headline bench stays bf16.
     fp8: bool = False
+    # second fp8 tier: dX (dgrad) GEMMs in e5m2 x e4m3 as well — dW and
+    # the optimizer stay bf16/fp32 (the highest-risk path never drops)
+    fp8_dgrad: bool = False
+    # third tier (full TE-style recipe): dW GEMMs in fp8 too; the fp32
+    # master still integrates every update
+    fp8_wgrad: bool = False
     overrides: dict = Field(default_factory=dict)
 
 

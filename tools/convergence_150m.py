@@ -33,6 +33,8 @@ cfg = TrainConfig(
     run_name="zipf150m", steps=300,
     model=ModelConfig(name="llama_150m", seq_len=512,
                       fp8=os.environ.get("PRIME_AMD_FP8", "0") == "1",
+                      fp8_dgrad=os.environ.get("PRIME_AMD_FP8_DGRAD", "0") == "1",
+                      fp8_wgrad=os.environ.get("PRIME_AMD_FP8_WGRAD", "0") == "1",
                       overrides={"vocab_size": V, "max_seq": 1024}),
     data=DataSection(kind="token_file", path="/tmp/zipf.bin",
                      micro_batch_size=16),
