@@ -147,7 +147,9 @@ def main() -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16+fp8fwd" if args.fp8 else "bf16",
+            "dtype": ("bf16+fp8fwd" + ("+dgrad" if args.fp8_dgrad else "")
+                      + ("+wgrad" if args.fp8_wgrad else "")
+                      if args.fp8 else "bf16"),
             "data": "synthetic",
             "mfu": mfu,
             # BASELINE.md convention: record the peak used for MFU (DENSE
