@@ -29,9 +29,15 @@ def main():
                         activation_checkpointing=True)
     m = m.to(dtype=torch.bfloat16)
     m.reset_rope(torch.device(dev))
+    import os
+
     flat = FlatParamSpace(m)
     opt = FusedAdamW(flat, lr=1e-4)
     ops.set_linear_tuned(True)
+    fp8 = os.environ.get("PRIME_AMD_FP8", "0") == "1"
+    ops.set_linear_fp8(fp8, dgrad=fp8, wgrad=fp8)
+    if fp8:
+        print("fp8 mode: fwd+dgrad+wgrad")
     n_params = flat.numel_padded
     print(f"layers={n_layers} params={n_params/1e9:.2f}B "
           f"(full 80-layer model: ~{(n_params + 0) / n_layers * 80 / 1e9:.0f}B-ish core)")
