@@ -386,6 +386,32 @@ def _wt_of(w: torch.Tensor):
     return wt
 
 
+# Wgrad deferral (Megatron-style dW on a side stream, overlapping the
+# backward's memory-bound kernels) was implemented and measured: the
+# naive version collapsed throughput 3x — side-stream allocations split
+# the caching allocator's pools at 157 GB live and every dW buffer
+# forced cross-stream synchronization. Doing it right needs main-stream-
+# allocated out= buffers for every intermediate (incl. an out= variant
+# of the transpose kernel); parked as a documented negative in
+# profiles/10b_1gpu_profile.md.
+def _compute_dw(x2, w, dy2):
+    N, K = w.shape
+    M = x2.shape[0]
+    if N % 128 == 0 and M % 64 == 0 and M >= 4096:
+        dyt = transpose_bshd(dy2.view(1, M, N // 128, 128)).view(N, M)
+        if (_LINEAR_FP8_WGRAD and K % 128 == 0 and M % 16 == 0
+                and x2.dtype == torch.bfloat16):
+            xt = transpose_bshd(x2.view(1, M, K // 128, 128)).view(K, M)
+            xt8, xtinv = _quant_act_fp8(xt, ("xT", w.data_ptr(), *w.shape))
+            dyt8, dytinv = _quant_act_fp8(
+                dyt, ("dyT", w.data_ptr(), *w.shape), e5m2=True)
+            g = torch._scaled_mm(xt8, dyt8.t(), scale_a=xtinv,
+                                 scale_b=dytinv, out_dtype=x2.dtype)
+            return transpose_bshd(g.view(1, K, N // 128, 128)).view(N, K)
+        return torch.mm(dyt, x2)
+    return torch.mm(dy2.t(), x2)
+
+
 def _linear_backward(x2, w, dy):
     """Shared layout-tuned backward for the bf16 and fp8-forward paths."""
     N, K = w.shape
@@ -405,29 +431,8 @@ def _linear_backward(x2, w, dy):
         dx = torch.matmul(dy2, wt.t())
     else:
         dx = torch.matmul(dy2, w)
-    # dW as NN: transpose dY once (LDS-tiled kernel), then plain NN.
-    # Measured net +3-6% on the 10B shapes (qkv alone is -1%; not
-    # worth a per-shape table)
-    if N % 128 == 0 and M % 64 == 0 and M >= 4096:
-        dyt = transpose_bshd(dy2.view(1, M, N // 128, 128)).view(N, M)
-        if (_LINEAR_FP8_WGRAD and K % 128 == 0 and M % 16 == 0
-                and x2.dtype == torch.bfloat16):
-            # fp8 wgrad (the full TE-style recipe, opt-in behind its own
-            # flag): dW^T = (x^T)_e4m3 @ (dY^T)^T_e5m2, then the LDS-tiled
-            # transpose back to [N,K]. The fp32 master still integrates
-            # every update; the convergence gate is the Zipf-150m run.
-            xt = transpose_bshd(x2.view(1, M, K // 128, 128)).view(K, M)
-            xt8, xtinv = _quant_act_fp8(xt, ("xT", w.data_ptr(), *w.shape))
-            dyt8, dytinv = _quant_act_fp8(
-                dyt, ("dyT", w.data_ptr(), *w.shape), e5m2=True)
-            g = torch._scaled_mm(xt8, dyt8.t(), scale_a=xtinv,
-                                 scale_b=dytinv, out_dtype=x2.dtype)
-            dw = transpose_bshd(g.view(1, K, N // 128, 128)).view(N, K)
-        else:
-            dw = torch.mm(dyt, x2)
-    else:
-        dw = torch.mm(dy2.t(), x2)
-    return dx.view(*dy.shape[:-1], K), dw
+    # dW: NN-via-dY^T rewrite + the fp8 wgrad tier live in _compute_dw
+    return dx.view(*dy.shape[:-1], K), _compute_dw(x2, w, dy2)
 
 
 class _TunedLinear(torch.autograd.Function):

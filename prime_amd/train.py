@@ -137,6 +137,7 @@ class Trainer:
         _ops.set_linear_fp8(cfg.model.fp8 and self.device.type == "cuda",
                             dgrad=cfg.model.fp8_dgrad,
                             wgrad=cfg.model.fp8_wgrad)
+
         self.inner = FusedAdamW(
             self.flat, lr=cfg.optim.lr, betas=tuple(cfg.optim.betas),
             eps=cfg.optim.eps, weight_decay=cfg.optim.weight_decay,

@@ -107,3 +107,4 @@ def test_generate_graphed_matches_eager(tmp_path):
     eager = generate(m, prompt, max_new_tokens=12, temperature=0.0, use_graph=False)
     graphed = generate(m, prompt, max_new_tokens=12, temperature=0.0, use_graph=True)
     assert eager.tolist() == graphed.tolist()
+
