@@ -394,20 +394,52 @@ def _wt_of(w: torch.Tensor):
 # allocated out= buffers for every intermediate (incl. an out= variant
 # of the transpose kernel); parked as a documented negative in
 # profiles/10b_1gpu_profile.md.
+def _transpose_quant_fp8(x2: torch.Tensor, site: tuple, e5m2: bool = False):
+    """[M, C] bf16 -> [C, M] fp8 in ONE pass (csrc/transpose.hip's fused
+    variant): the wgrad operands x^T / dY^T are consumed only as fp8, so
+    the bf16 transposed intermediate was pure HBM traffic. Falls back to
+    transpose + fused quant on the bootstrap call (amax not seeded)."""
+    M, C = x2.shape
+    st = _FP8_ACT.get(site)
+    # the one-pass variant measured 8% SLOWER end-to-end (10B fp8 bench
+    # 20.1k vs 21.9k): its 8-byte fp8 stores halve the store width and
+    # the per-element conversion sits in the store loop — the separate
+    # LDS-tiled transpose + fused quant pair wins. Kept opt-in for
+    # further tuning (PRIME_AMD_FUSED_TQ=1).
+    import os
+
+    fused_ok = os.environ.get("PRIME_AMD_FUSED_TQ", "0") == "1"
+    if st is None or M % 64 or C % 128 or not fused_ok:
+        xt = transpose_bshd(x2.view(1, M, C // 128, 128)).view(C, M)
+        return _quant_act_fp8(xt, site, e5m2=e5m2)
+    dt8 = torch.float8_e5m2 if e5m2 else torch.float8_e4m3fn
+    xt8 = torch.empty(C, M, device=x2.device, dtype=dt8)
+    st["next"].zero_()
+    check(
+        lib().prime_transpose_fp8(
+            stream_of(x2), ptr(x2), ptr(xt8), 1, M, C // 128, 128,
+            x2.stride(0) * M, x2.stride(0), 128,
+            ptr(st["amax"]), ptr(st["next"]), ptr(st["sinv"]), 1 if e5m2 else 0,
+        ),
+        "transpose_fp8",
+    )
+    st["amax"], st["next"] = st["next"], st["amax"]
+    return xt8, st["sinv"]
+
+
 def _compute_dw(x2, w, dy2):
     N, K = w.shape
     M = x2.shape[0]
     if N % 128 == 0 and M % 64 == 0 and M >= 4096:
-        dyt = transpose_bshd(dy2.view(1, M, N // 128, 128)).view(N, M)
         if (_LINEAR_FP8_WGRAD and K % 128 == 0 and M % 16 == 0
                 and x2.dtype == torch.bfloat16):
-            xt = transpose_bshd(x2.view(1, M, K // 128, 128)).view(K, M)
-            xt8, xtinv = _quant_act_fp8(xt, ("xT", w.data_ptr(), *w.shape))
-            dyt8, dytinv = _quant_act_fp8(
-                dyt, ("dyT", w.data_ptr(), *w.shape), e5m2=True)
+            xt8, xtinv = _transpose_quant_fp8(x2, ("xT", w.data_ptr(), *w.shape))
+            dyt8, dytinv = _transpose_quant_fp8(
+                dy2, ("dyT", w.data_ptr(), *w.shape), e5m2=True)
             g = torch._scaled_mm(xt8, dyt8.t(), scale_a=xtinv,
                                  scale_b=dytinv, out_dtype=x2.dtype)
             return transpose_bshd(g.view(1, K, N // 128, 128)).view(N, K)
+        dyt = transpose_bshd(dy2.view(1, M, N // 128, 128)).view(N, M)
         return torch.mm(dyt, x2)
     return torch.mm(dy2.t(), x2)
 
