@@ -70,3 +70,18 @@ def test_fp8_fsdp_rejected(tmp_path):
     )
     with pytest.raises(ValueError, match="fp8.*fsdp|fsdp.*fp8"):
         Trainer(cfg, run_dir=tmp_path)
+
+
+def test_all_shipped_configs_parse():
+    """Every TOML under configs/ must validate against the strict schema
+    (the 8-GPU-node configs are otherwise only exercised by the driver)."""
+    from pathlib import Path
+
+    from prime_amd.utils.config import load_config
+
+    root = Path(__file__).resolve().parent.parent / "configs"
+    tomls = sorted(root.glob("*.toml"))
+    assert len(tomls) >= 6
+    for t in tomls:
+        cfg = load_config(t)
+        assert cfg.model.name
