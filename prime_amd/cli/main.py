@@ -882,12 +882,23 @@ def serve_cmd(
 
 # ------------------------------------------------------------------ bench
 @app.command("bench")
-def bench(steps: int = 10, warmup: int = 3, model: str = "intellect_10b"):
+def bench(steps: int = 10, warmup: int = 3, model: str = "intellect_10b",
+          seq_len: int = 2048, micro_batch: int = 8,
+          fp8: bool = typer.Option(False, help="opt-in fp8 GEMM tiers"),
+          fp8_dgrad: bool = False, fp8_wgrad: bool = False):
     """Run the flagship benchmark (see bench.py for the driver contract)."""
     import bench as bench_mod  # noqa: F401 — repo-root bench
-    subprocess.run([sys.executable, str(Path(bench_mod.__file__)),
-                    "--steps", str(steps), "--warmup", str(warmup),
-                    "--model", model], check=True)
+    cmd = [sys.executable, str(Path(bench_mod.__file__)),
+           "--steps", str(steps), "--warmup", str(warmup),
+           "--model", model, "--seq-len", str(seq_len),
+           "--micro-batch", str(micro_batch)]
+    if fp8:
+        cmd.append("--fp8")
+    if fp8_dgrad:
+        cmd.append("--fp8-dgrad")
+    if fp8_wgrad:
+        cmd.append("--fp8-wgrad")
+    subprocess.run(cmd, check=True)
 
 
 def main() -> None:

@@ -85,3 +85,24 @@ def test_all_shipped_configs_parse():
     for t in tomls:
         cfg = load_config(t)
         assert cfg.model.name
+
+
+def test_fp8_flag_inert_on_cpu(tmp_path):
+    """fp8 config on a CPU run must be a no-op (the plumbing configs keep
+    working on fp8-flagged TOMLs)."""
+    from prime_amd.train import Trainer
+    from prime_amd.utils.config import (DilocoConfig, MetricsConfig,
+                                        ModelConfig, TrainConfig)
+
+    cfg = TrainConfig(
+        run_name="fp8cpu", steps=2,
+        model=ModelConfig(name="llama_test", seq_len=32, fp8=True,
+                          fp8_dgrad=True, fp8_wgrad=True),
+        diloco=DilocoConfig(H=10**6),
+        metrics=MetricsConfig(log_interval=100),
+    )
+    cfg.data.micro_batch_size = 1
+    tr = Trainer(cfg, run_dir=tmp_path)
+    loss = tr.train_step()
+    assert float(loss) > 0
+    tr.close()
