@@ -360,6 +360,13 @@ class Trainer:
                 tps = self.tokens_per_step * window_steps / dt
                 tps_gpu = tps / max(1, self.mesh.world_size)
                 last_loss = float(loss)
+                if last_loss != last_loss or last_loss in (float("inf"),
+                                                           float("-inf")):
+                    from .utils.failures import NonFiniteLossError
+
+                    raise NonFiniteLossError(
+                        f"loss is {last_loss} at step {self.step_count}"
+                    )
                 self.wandb.write(
                     self.step_count, loss=last_loss, tokens_per_sec=tps,
                     mfu=mfu(tps_gpu, self.flops_per_token),

@@ -95,3 +95,32 @@ def test_tied_embeddings_flat_space():
     loss = m.loss(x, x)
     loss.backward()
     assert torch.isfinite(loss)
+
+
+def test_non_finite_loss_detected(tmp_path):
+    """A diverged (NaN) loss must stop the run with the classified error
+    instead of training on silently."""
+    import pytest
+    import torch
+
+    from prime_amd.train import Trainer
+    from prime_amd.utils.config import (DilocoConfig, MetricsConfig,
+                                        ModelConfig, OptimConfig, TrainConfig)
+    from prime_amd.utils.failures import NonFiniteLossError, classify_failure
+
+    cfg = TrainConfig(
+        run_name="nanrun", steps=3,
+        model=ModelConfig(name="llama_test", seq_len=32),
+        diloco=DilocoConfig(H=10**6),
+        optim=OptimConfig(lr=0.0, grad_clip=0),
+        metrics=MetricsConfig(log_interval=1),
+    )
+    cfg.data.micro_batch_size = 1
+    tr = Trainer(cfg, run_dir=tmp_path)
+    # poison the weights -> NaN loss on the next forward
+    with torch.no_grad():
+        tr.flat.flat_w.fill_(float("nan"))
+    with pytest.raises(NonFiniteLossError):
+        tr.run()
+    assert classify_failure(NonFiniteLossError("x"))["category"] == "NON_FINITE_LOSS"
+    tr.close()
