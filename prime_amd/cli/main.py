@@ -315,11 +315,11 @@ def stop(run: str = typer.Argument(...)):
     if not pid or not _pid_alive(pid):
         secho("not running", fg="yellow")
         return
-    # exact recorded pid (process group) — never pattern-based. The runner
-    # handles SIGTERM gracefully: checkpoint at the next step boundary,
-    # then it writes the final STOPPED status itself.
+    # exact recorded process group (never pattern-based): signalling the
+    # pgid reaches EVERY rank, and the trainer's per-step consensus stop
+    # makes them exit at the same boundary with a checkpoint.
     try:
-        os.killpg(pid, signal.SIGTERM)
+        os.killpg(st.get("pgid") or pid, signal.SIGTERM)
     except OSError:
         os.kill(pid, signal.SIGTERM)
     secho(f"stop requested for {d.name} (graceful: checkpoints first)", fg="green")

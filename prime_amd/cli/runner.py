@@ -21,7 +21,8 @@ def main() -> int:
     cfg = load_config(cfg_path)
     if rank == 0:
         (run_dir / "status.json").write_text(json.dumps(
-            {"status": "RUNNING", "pid": os.getpid(), "started": time.time(),
+            {"status": "RUNNING", "pid": os.getpid(),
+             "pgid": os.getpgid(0), "started": time.time(),
              "steps_total": cfg.steps}))
     trainer = Trainer(cfg, run_dir)
 

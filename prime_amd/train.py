@@ -345,8 +345,20 @@ class Trainer:
         for _ in range(cfg.steps):
             if profiler is not None:
                 profiler.step()
-            if self.stop_requested:
+            stop = self.stop_requested
+            if self.mesh.initialized and self.mesh.world_size > 1:
+                # consensus stop: SIGTERM lands on ranks at different
+                # steps; without agreement the early stopper leaves peers
+                # hanging in the next collective
+                import torch.distributed as dist
+
+                t = torch.tensor([1 if stop else 0], device=self.device
+                                 if self.device.type == "cuda" else "cpu")
+                dist.all_reduce(t, op=dist.ReduceOp.MAX)
+                stop = bool(int(t[0]))
+            if stop:
                 self.log.warning("stop requested: saving checkpoint and exiting")
+                self.stop_requested = True
                 if self.ckpt:
                     self.save_checkpoint()
                 break

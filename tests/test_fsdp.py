@@ -224,3 +224,48 @@ def test_fsdp_async_ordering_stress():
         assert abs(a - b) < 1e-3, (dp[0]["losses"], sh[0]["losses"])
     for k in dp[0]["w"]:
         assert abs(dp[0]["w"][k] - sh[0]["w"][k]) < 2e-2, k
+
+
+def _consensus_stop_worker(rank, world):
+    """Only rank 1 receives the 'signal'; both ranks must stop together at
+    the same boundary (the consensus all-reduce) with a final checkpoint."""
+    from prime_amd.train import Trainer
+    from prime_amd.utils.config import (
+        CheckpointConfig, DilocoConfig, MetricsConfig, ModelConfig,
+        ParallelConfig, TrainConfig,
+    )
+
+    cfg = TrainConfig(
+        run_name="stopc",
+        steps=50,
+        model=ModelConfig(name="llama_test", seq_len=32),
+        diloco=DilocoConfig(H=10**6),
+        parallel=ParallelConfig(worker_size=world),
+        checkpoint=CheckpointConfig(interval=1, async_save=False),
+        metrics=MetricsConfig(log_interval=1000),
+    )
+    cfg.data.micro_batch_size = 1
+    tr = Trainer(cfg, run_dir=f"/tmp/prime_amd_test/stopc_r{rank}")
+    steps_done = 0
+    orig = tr.train_step
+
+    def step_and_maybe_stop():
+        nonlocal steps_done
+        out = orig()
+        steps_done += 1
+        if rank == 1 and steps_done == 3:
+            tr.stop_requested = True  # simulated SIGTERM on ONE rank
+        return out
+
+    tr.train_step = step_and_maybe_stop
+    res = tr.run()
+    tr.close()
+    return {"steps": res["steps"], "stopped": tr.stop_requested}
+
+
+def test_consensus_stop_two_ranks():
+    outs = run_distributed(_consensus_stop_worker, 2, args=(), timeout=300)
+    # both ranks agreed to stop at the first boundary after rank 1's
+    # step-3 request (stop is detected at the next loop top)
+    assert outs[0]["steps"] == outs[1]["steps"] == 3
+    assert outs[0]["stopped"] and outs[1]["stopped"]
