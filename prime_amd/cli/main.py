@@ -188,7 +188,15 @@ def init(
 
 
 @train_app.command("list")
-def list_runs(json_out: bool = typer.Option(False, "--json")):
+def list_runs(json_out: bool = typer.Option(False, "--json"),
+              output: str = typer.Option(
+                  "table", "--output", "-o",
+                  help="table | json (reference: display.py format "
+                       "validation)")):
+    from ..utils.display import validate_format
+
+    if validate_format(output) == "json":
+        json_out = True
     rows = []
     for d in sorted(runs_root().iterdir()):
         if not d.is_dir():
@@ -210,13 +218,9 @@ def list_runs(json_out: bool = typer.Option(False, "--json")):
     if not rows:
         typer.echo("no runs")
         return
-    fmt = "{run:40s} {status:22s} {step:>8} {loss:>10} {tok_s:>12}"
-    typer.echo(fmt.format(run="RUN", status="STATUS", step="STEP", loss="LOSS", tok_s="TOK/S"))
-    for r in rows:
-        loss = f"{r['loss']:.4f}" if isinstance(r["loss"], float) else r["loss"]
-        tps = f"{r['tok/s']:,.0f}" if isinstance(r["tok/s"], float) else r["tok/s"]
-        typer.echo(fmt.format(run=r["run"], status=r["status"], step=str(r["step"]),
-                              loss=loss, tok_s=tps))
+    table(rows, [("run", "RUN", -40), ("status", "STATUS", -22),
+                 ("step", "STEP", 8), ("loss", "LOSS", 10),
+                 ("tok/s", "TOK/S", 12)])
 
 
 def _last_metrics(run_dir: Path) -> dict:

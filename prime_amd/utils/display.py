@@ -46,16 +46,6 @@ def emit_json(data: Any) -> None:
     typer.echo(json.dumps(data, indent=2, default=str))
 
 
-def emit(data: Any, json_out: bool, table_fn=None) -> None:
-    """Emit `data` as JSON when requested, else via table_fn (or repr)."""
-    if json_out:
-        emit_json(data)
-    elif table_fn is not None:
-        table_fn(data)
-    else:
-        typer.echo(str(data))
-
-
 def table(rows: Iterable[dict], columns: list[tuple[str, str, int]]) -> None:
     """Minimal fixed-width table: columns = [(key, header, width)]."""
     rows = list(rows)

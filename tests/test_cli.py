@@ -306,3 +306,16 @@ def test_doctor(tmp_path, monkeypatch):
     assert checks["torch"]["ok"]
     assert checks["kernel .so"]["ok"]     # fresh source-hash stamp
     assert not checks["gpu"]["ok"]        # no GPU in the build container
+
+
+def test_list_output_format_validation(tmp_path, monkeypatch):
+    from typer.testing import CliRunner
+
+    from prime_amd.cli.main import app
+
+    monkeypatch.setenv("PRIME_AMD_RUNS_DIR", str(tmp_path))
+    r = CliRunner().invoke(app, ["train", "list", "--output", "yaml"])
+    assert r.exit_code == 2
+    assert "invalid output format" in r.output
+    r = CliRunner().invoke(app, ["train", "list", "--output", "json"])
+    assert r.exit_code == 0
