@@ -57,3 +57,54 @@ def test_softmax_reference_attention_rows_sum(seq, heads):
     # attention over constant V returns constant rows (probabilities sum to 1)
     o = ref.attention(q, k, v, causal=True)
     torch.testing.assert_close(o, torch.ones_like(o), atol=1e-5, rtol=1e-5)
+
+
+@given(st.integers(1, 2048), st.integers(1, 64))
+@settings(max_examples=40, deadline=None)
+def test_pad_run_preserves_content(n, align_units):
+    """The per-call ring padding helper must hand the callee a zero-padded
+    aligned buffer and copy the result back exactly."""
+    import torch
+
+    from prime_amd.parallel.ring import _pad_run
+
+    t = torch.randn(n)
+    orig = t.clone()
+    align = align_units * 8
+    seen = {}
+
+    def fn(buf):
+        seen["len"] = buf.numel()
+        assert buf.numel() % align == 0
+        torch.testing.assert_close(buf[:n], orig)
+        assert float(buf[n:].abs().sum()) == 0.0
+        buf.mul_(2.0)
+
+    ran = _pad_run(t, align, fn)
+    if n % align == 0:
+        assert not ran  # caller handles aligned sizes itself
+        torch.testing.assert_close(t, orig)
+    else:
+        assert ran and seen["len"] >= n
+        torch.testing.assert_close(t, orig * 2.0)
+
+
+@given(st.text(alphabet="ABCdefGHI_jkl012",
+               min_size=1, max_size=12),
+       st.text(alphabet=st.characters(blacklist_characters="\n\"'$",
+                                      blacklist_categories=("Cs", "Cc")),
+               max_size=24))
+@settings(max_examples=50, deadline=None)
+def test_env_file_roundtrip(key, value):
+    import tempfile
+    from pathlib import Path
+
+    from prime_amd.utils.env_vars import EnvFileError, parse_env_file
+
+    if key[0].isdigit():
+        key = "_" + key
+    with tempfile.TemporaryDirectory() as d:
+        f = Path(d) / ".env"
+        f.write_text(f"{key}={value}\n")
+        env = parse_env_file(f)
+        assert env[key] == value.strip()
