@@ -44,6 +44,7 @@ _SIGS = {
     "prime_gemm_nt": [ctypes.c_void_p] * 4 + [ctypes.c_int64] * 4,
     "prime_gemm_nt8": [ctypes.c_void_p] * 4 + [ctypes.c_int64] * 4,
     "prime_quant_fp8": [ctypes.c_void_p] * 6 + [ctypes.c_int64] * 2,
+    "prime_rowwise_quant_fp8": [ctypes.c_void_p] * 4 + [ctypes.c_int64] * 3,
     "prime_transpose_fp8": [ctypes.c_void_p] * 3 + [ctypes.c_int64] * 7 + [ctypes.c_void_p] * 3 + [ctypes.c_int64],
 }
 

@@ -73,3 +73,70 @@ PRIME_API int prime_quant_fp8(hipStream_t stream, const void* x, void* y,
                        (float*)sinv_out, n / 8);
   return (int)hipGetLastError();
 }
+
+// Rowwise variant: per-ROW amax + scale (no delayed-scaling state, no
+// outlier saturation across tokens). One workgroup per row: pass 1 reads
+// the row for its amax (coalesced), pass 2 re-reads (L2-hot — rows are
+// 8-256 KB against 4 MiB per-XCD L2) and casts. Emits the per-row
+// dequant scale vector torch._scaled_mm consumes as scale_a=[M,1] /
+// scale_b=[1,N].
+template <int E5M2>
+__global__ __launch_bounds__(256) void rowwise_quant_fp8_kernel(
+    const bf16* __restrict__ x, unsigned char* __restrict__ y,
+    float* __restrict__ sinv_out, int64_t R, int64_t C8) {
+  const float fmax8 = E5M2 ? 57344.f : 448.f;
+  __shared__ float wmax[4];
+  __shared__ float s_shared;
+  for (int64_t r = blockIdx.x; r < R; r += gridDim.x) {
+    const bf16* row = x + r * C8 * 8;
+    float amax = 0.f;
+    for (int64_t u = threadIdx.x; u < C8; u += 256) {
+      const q8_short8 raw = *reinterpret_cast<const q8_short8*>(row + u * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        amax = fmaxf(amax, fabsf(bf2f(reinterpret_cast<const bf16*>(&raw)[j])));
+    }
+    amax = wave_reduce_max(amax);
+    if ((threadIdx.x & 63) == 0) wmax[threadIdx.x >> 6] = amax;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      float m = fmaxf(fmaxf(wmax[0], wmax[1]), fmaxf(wmax[2], wmax[3]));
+      m = fmaxf(m, 1e-12f);
+      s_shared = fmax8 / m;
+      sinv_out[r] = m / fmax8;
+    }
+    __syncthreads();
+    const float s = s_shared;
+    unsigned char* orow = y + r * C8 * 8;
+    for (int64_t u = threadIdx.x; u < C8; u += 256) {
+      const q8_short8 raw = *reinterpret_cast<const q8_short8*>(row + u * 8);
+      unsigned long long packed = 0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float q = fminf(fmaxf(
+            bf2f(reinterpret_cast<const bf16*>(&raw)[j]) * s, -fmax8), fmax8);
+        const unsigned char b = E5M2 ? __hip_fp8_e5m2(q).__x
+                                     : __hip_fp8_e4m3(q).__x;
+        packed |= (unsigned long long)b << (8 * j);
+      }
+      *reinterpret_cast<unsigned long long*>(orow + u * 8) = packed;
+    }
+    __syncthreads();  // wmax/s_shared reused by the next row
+  }
+}
+
+PRIME_API int prime_rowwise_quant_fp8(hipStream_t stream, const void* x,
+                                      void* y, void* sinv_out, int64_t R,
+                                      int64_t C, int64_t e5m2) {
+  if (C % 8) return hipErrorInvalidValue;
+  const int grid = (int)(R < 2048 ? R : 2048);
+  if (e5m2)
+    hipLaunchKernelGGL((rowwise_quant_fp8_kernel<1>), dim3(grid), dim3(256),
+                       0, stream, (const bf16*)x, (unsigned char*)y,
+                       (float*)sinv_out, R, C / 8);
+  else
+    hipLaunchKernelGGL((rowwise_quant_fp8_kernel<0>), dim3(grid), dim3(256),
+                       0, stream, (const bf16*)x, (unsigned char*)y,
+                       (float*)sinv_out, R, C / 8);
+  return (int)hipGetLastError();
+}
