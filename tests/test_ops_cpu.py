@@ -117,3 +117,16 @@ def test_fused_add_rmsnorm_cpu():
     torch.testing.assert_close(x.grad, x2.grad, atol=1e-5, rtol=1e-4)
     torch.testing.assert_close(res.grad, r2.grad, atol=1e-5, rtol=1e-4)
     torch.testing.assert_close(w.grad, w2.grad, atol=1e-5, rtol=1e-4)
+
+
+def test_fp8_rowwise_env_gate(monkeypatch):
+    """Per-row fp8 scaling is opt-in: default off (delayed scaling wins
+    the 10B A/B by 1.5%), PRIME_AMD_FP8_ROWWISE=1 turns it on."""
+    from prime_amd.ops.functional import _fp8_rowwise_on
+
+    monkeypatch.delenv("PRIME_AMD_FP8_ROWWISE", raising=False)
+    assert not _fp8_rowwise_on()
+    monkeypatch.setenv("PRIME_AMD_FP8_ROWWISE", "1")
+    assert _fp8_rowwise_on()
+    monkeypatch.setenv("PRIME_AMD_FP8_ROWWISE", "0")
+    assert not _fp8_rowwise_on()

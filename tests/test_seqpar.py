@@ -14,7 +14,7 @@ def _a2a_worker(rank, world):
 
     dist.init_process_group("gloo", rank=rank, world_size=world)
     torch.manual_seed(7)  # same full tensors on both ranks
-    B, S, H, Hkv, D = 2, 32, 4, 2, 16
+    B, S, H, Hkv, D = 2, 16 * world, 2 * world, world, 16
     q_full = torch.randn(B, S, H, D)
     k_full = torch.randn(B, S, Hkv, D)
     v_full = torch.randn(B, S, Hkv, D)
@@ -48,6 +48,11 @@ def _a2a_worker(rank, world):
 
 def test_ulysses_a2a_and_attention():
     assert all(run_distributed(_a2a_worker, 2, timeout=120))
+
+
+def test_ulysses_a2a_world4():
+    # SP degree 4 (the intellect10b_sp4_32k.toml shape, tiny dims)
+    assert all(run_distributed(_a2a_worker, 4, timeout=180))
 
 
 def _sp_trainer(rank, world, sp, steps):
