@@ -460,7 +460,7 @@ def _linear_backward(x2, w, dy):
         # a rowwise scale_a with a scalar scale_b is rejected.
         dy8, dyinv = _quant_act_fp8(dy2, ("dy", w.data_ptr(), *w.shape),
                                     e5m2=True)
-        wt8, wtinv = _w8_of(wt)
+        wt8, wtinv = _w8_of(wt, skey=("wt", w.data_ptr(), *w.shape))
         dx = torch._scaled_mm(dy8, wt8.t(), scale_a=dyinv, scale_b=wtinv,
                               out_dtype=dy2.dtype)
     elif wt is not None:
@@ -510,8 +510,14 @@ def set_linear_fp8(on: bool, dgrad: bool = False, wgrad: bool = False) -> None:
     _FP8_ACT.clear()
 
 
-def _w8_of(w: torch.Tensor, rowwise: bool = False):
-    key = (w.data_ptr(), *w.shape, rowwise)
+def _w8_of(w: torch.Tensor, rowwise: bool = False, skey: tuple | None = None):
+    # skey: stable identity for tensors whose storage churns per step
+    # (the cached W^T) — keying state by their data_ptr re-bootstraps the
+    # delayed-scaling amax every step (measured ~237 extra torch
+    # abs+amax launches/step) and leaks stale _FP8_ACT/_FP8_CACHE
+    # entries as the allocator rotates blocks.
+    key = ((skey if skey is not None else (w.data_ptr(), *w.shape))
+           + (rowwise,))
     hit = _FP8_CACHE.get(key)
     if hit is not None and hit[0] == _WT_EPOCH:
         return hit[1], hit[2]
