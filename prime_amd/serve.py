@@ -70,6 +70,10 @@ def create_app(model, model_name: str, tokenizer=None):
         return decode(tokenizer, ids)
 
     def _run(ids: list[int], req) -> dict:
+        if not ids:
+            raise HTTPException(400, "empty prompt")
+        if not 1 <= req.max_tokens <= 8192:
+            raise HTTPException(400, "max_tokens must be in [1, 8192]")
         toks = torch.tensor([ids], device=dev)
         with lock:
             t0 = time.perf_counter()
