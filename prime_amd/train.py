@@ -342,6 +342,7 @@ class Trainer:
         window_t0, window_steps = t_start, 0
         last_loss = float("nan")
         loss = None
+        executed = 0  # steps run THIS session (early stop / resume aware)
         for _ in range(cfg.steps):
             if profiler is not None:
                 profiler.step()
@@ -363,6 +364,7 @@ class Trainer:
                     self.save_checkpoint()
                 break
             loss = self.train_step()
+            executed += 1
             window_steps += 1
             if self.step_count % cfg.metrics.log_interval == 0:
                 if self.device.type == "cuda":
@@ -409,7 +411,7 @@ class Trainer:
             "outer_steps": self.diloco.outer_step_count,
             "loss": last_loss,
             "total_time_s": total_t,
-            "tokens_per_sec": self.tokens_per_step * cfg.steps / total_t,
+            "tokens_per_sec": self.tokens_per_step * executed / total_t,
         }
         self.log.info(f"run complete: {result}", type_="result")
         return result
