@@ -101,3 +101,54 @@ def test_sp_trainer_matches_full_context():
         assert abs(sp[0]["w"][kk] - ref[0]["w"][kk]) < 2e-2, kk
     # both SP ranks hold identical replicated params
     assert sp[0]["w"] == sp[1]["w"]
+
+
+def _sp_diloco_trainer(rank, world, sp, steps):
+    """SP composed with DiLoCo: (sp=True) 2 workers x SP(2) over seq 32
+    slices vs (sp=False) 2 full-context workers at seq 64 — same data
+    per worker, H=2 outer syncs included; losses and final params must
+    match the full-context run."""
+    from prime_amd.train import Trainer
+    from prime_amd.utils.config import (
+        DilocoConfig, MetricsConfig, ModelConfig, ParallelConfig, TrainConfig,
+    )
+
+    ws = 2 if sp else 1
+    cfg = TrainConfig(
+        run_name=f"spdiloco_{sp}",
+        steps=steps,
+        model=ModelConfig(name="llama_test", seq_len=32 if sp else 64),
+        diloco=DilocoConfig(H=2),
+        parallel=ParallelConfig(worker_size=ws, seq_parallel=sp),
+        metrics=MetricsConfig(log_interval=100),
+    )
+    cfg.data.micro_batch_size = 2
+    tr = Trainer(cfg, run_dir=f"/tmp/prime_amd_test/spd{sp}_r{rank}")
+    losses = []
+    for _ in range(steps):
+        loss = tr.train_step()
+        if sp:
+            import torch.distributed as dist
+
+            t = loss.detach().clone()
+            dist.all_reduce(t, group=tr.mesh.local_group)
+            losses.append(float(t) / ws)
+        else:
+            losses.append(float(loss))
+    w = {n: p.detach().sum().item() for n, p in tr.model.named_parameters()}
+    outer = tr.diloco.outer_step_count
+    tr.close()
+    return {"losses": losses, "w": w, "outer": outer}
+
+
+def test_sp_composes_with_diloco():
+    sp = run_distributed(_sp_diloco_trainer, 4, args=(True, 4), timeout=300)
+    ref = run_distributed(_sp_diloco_trainer, 2, args=(False, 4), timeout=300)
+    assert sp[0]["outer"] == ref[0]["outer"] == 2  # H=2, 4 steps
+    # worker 0 of the SP run = ranks 0,1; its loss must track ref worker 0
+    for a, b in zip(sp[0]["losses"], ref[0]["losses"]):
+        assert abs(a - b) < 1e-3, (sp[0]["losses"], ref[0]["losses"])
+    for kk in ref[0]["w"]:
+        assert abs(sp[0]["w"][kk] - ref[0]["w"][kk]) < 2e-2, kk
+    # SP ranks of one worker hold identical replicated params
+    assert sp[0]["w"] == sp[1]["w"]
