@@ -343,7 +343,10 @@ class Trainer:
         last_loss = float("nan")
         loss = None
         executed = 0  # steps run THIS session (early stop / resume aware)
-        for _ in range(cfg.steps):
+        # cfg.steps is the run's TOTAL step target: a resumed run trains
+        # only the remainder, so a crash-restart loop converges on the
+        # configured total instead of extending it every restart
+        for _ in range(max(0, cfg.steps - self.step_count)):
             if profiler is not None:
                 profiler.step()
             stop = self.stop_requested

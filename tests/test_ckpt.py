@@ -51,11 +51,11 @@ def test_trainer_resume(tmp_path):
     inner_end = tr.diloco.inner_step_count
     tr.close()
 
-    tr2 = Trainer(_train_cfg(2, tmp_path, resume="latest"), run_dir=tmp_path / "runB")
+    tr2 = Trainer(_train_cfg(6, tmp_path, resume="latest"), run_dir=tmp_path / "runB")
     assert tr2.diloco.inner_step_count == inner_end
     torch.testing.assert_close(tr2.flat.flat_w, w_end)
-    res = tr2.run()  # continues training
-    assert res["steps"] == inner_end + 2
+    res = tr2.run()  # continues training up to the TOTAL step target
+    assert res["steps"] == 6
     tr2.close()
 
 
@@ -146,3 +146,28 @@ def test_capacity_guard_refuses_oversized(tmp_path, monkeypatch):
     monkeypatch.setattr(_shutil, "disk_usage", lambda p: FakeUsage)
     with pytest.raises(RuntimeError, match="refusing"):
         mgr.save(1, {"x": torch.zeros(1 << 22)}, {})  # 16 MiB > 1 MiB free
+
+
+def test_resume_trajectory_matches_uninterrupted(tmp_path):
+    """Stopping at an outer boundary and resuming must reproduce the
+    uninterrupted run EXACTLY: same data order (loader batch_idx), same
+    LR schedule (step_count), same AdamW moments — so the final weights
+    of 2+2 steps equal those of 4 straight steps."""
+    torch.manual_seed(0)
+    tr = Trainer(_train_cfg(4, tmp_path / "a"), run_dir=tmp_path / "runA")
+    tr.run()
+    w_straight = tr.flat.flat_w.clone()
+    m_straight = tr.inner.m.clone()
+    tr.close()
+
+    torch.manual_seed(0)
+    tr1 = Trainer(_train_cfg(2, tmp_path / "b"), run_dir=tmp_path / "runB1")
+    tr1.run()
+    tr1.close()
+    tr2 = Trainer(_train_cfg(4, tmp_path / "b", resume="latest"),
+                  run_dir=tmp_path / "runB2")
+    res = tr2.run()
+    assert res["steps"] == 4
+    torch.testing.assert_close(tr2.flat.flat_w, w_straight)
+    torch.testing.assert_close(tr2.inner.m, m_straight)
+    tr2.close()
