@@ -352,8 +352,13 @@ def restart(run: str = typer.Argument(...), detach: bool = typer.Option(False, "
     if not cfg_path.exists():
         secho("run has no config.toml", fg="red")
         raise typer.Exit(1)
+    import re
+
     text = cfg_path.read_text()
-    if "[checkpoint]" in text and "resume" not in text:
+    # line-anchored: a substring check would false-positive on e.g.
+    # run_name = "resume_test" and silently restart from scratch
+    has_resume = re.search(r"(?m)^\s*resume\s*=", text)
+    if "[checkpoint]" in text and not has_resume:
         text = text.replace("[checkpoint]", '[checkpoint]\nresume = "latest"')
     elif "[checkpoint]" not in text:
         text += '\n[checkpoint]\nresume = "latest"\n'

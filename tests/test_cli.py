@@ -319,3 +319,34 @@ def test_list_output_format_validation(tmp_path, monkeypatch):
     assert "invalid output format" in r.output
     r = CliRunner().invoke(app, ["train", "list", "--output", "json"])
     assert r.exit_code == 0
+
+
+def test_restart_completes_to_total(tmp_path, runs_dir):
+    """restart resumes from the latest checkpoint and finishes the run's
+    TOTAL step target (not cfg.steps more); a run_name containing the
+    word 'resume' must not defeat the resume injection."""
+    cfg = tmp_path / "run.toml"
+    cfg.write_text(
+        'run_name = "resume_edge"\nsteps = 4\n'
+        '[model]\nname = "llama_test"\nseq_len = 64\n'
+        '[data]\nmicro_batch_size = 2\n'
+        '[diloco]\nH = 2\n'
+        '[checkpoint]\ninterval = 1\nasync_save = false\n'
+        '[metrics]\nlog_interval = 100\n'
+    )
+    r = runner.invoke(app, ["train", "run", str(cfg)])
+    assert r.exit_code == 0, r.output
+    rows = json.loads(runner.invoke(app, ["train", "list", "--json"]).output)
+    run_id = rows[0]["run"]
+    r = runner.invoke(app, ["train", "restart", run_id])
+    assert r.exit_code == 0, r.output
+    # restart of a completed run trains 0 further steps and re-completes
+    rows = json.loads(runner.invoke(app, ["train", "list", "--json"]).output)
+    latest = [x for x in rows if x["run"].startswith("resume_edge")]
+    assert any(x["status"] == "COMPLETED" for x in latest)
+    # the rewritten config must carry resume = "latest"
+    import re as _re
+
+    restarted = sorted(runs_dir.glob("*/config.restart.toml"))
+    assert restarted and _re.search(r"(?m)^resume = \"latest\"",
+                                    restarted[-1].read_text())
