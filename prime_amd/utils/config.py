@@ -21,7 +21,7 @@ class _Strict(BaseModel):
 
 class ModelConfig(_Strict):
     name: str = "llama_150m"
-    seq_len: int = 2048
+    seq_len: int = Field(2048, gt=0)
     activation_checkpointing: bool = False
     # opt-in mixed precision: forward linears in OCP fp8-e4m3 (per-tensor
     # dynamic scaling; master/grads/backward stay bf16/fp32) — ~1.8x the
@@ -39,8 +39,8 @@ class ModelConfig(_Strict):
 class DataSection(_Strict):
     kind: Literal["synthetic", "token_file"] = "synthetic"
     path: str | None = None
-    micro_batch_size: int = 4
-    grad_accum: int = 1
+    micro_batch_size: int = Field(4, gt=0)
+    grad_accum: int = Field(1, gt=0)
     seed: int = 1234
     shuffle: bool = True
     # explicit DiLoCo-worker data index; elastic mode derives one from the
@@ -53,15 +53,15 @@ class OptimConfig(_Strict):
     betas: tuple[float, float] = (0.9, 0.95)
     eps: float = 1e-8
     weight_decay: float = 0.1
-    warmup_steps: int = 100
+    warmup_steps: int = Field(100, ge=0)
     lr_decay_steps: int | None = None
-    min_lr_ratio: float = 0.1
-    grad_clip: float = 1.0
+    min_lr_ratio: float = Field(0.1, ge=0, le=1)
+    grad_clip: float = Field(1.0, ge=0)
 
 
 class DilocoConfig(_Strict):
     enabled: bool = True
-    H: int = 100
+    H: int = Field(100, gt=0)
     outer_lr: float = 0.7
     outer_momentum: float = 0.9
     quant_int8: bool = True
@@ -69,7 +69,7 @@ class DilocoConfig(_Strict):
 
 
 class ParallelConfig(_Strict):
-    worker_size: int = 1              # GPUs per DiLoCo worker
+    worker_size: int = Field(1, gt=0)  # GPUs per DiLoCo worker
     fsdp: bool = False                # shard params across the worker
     # sequence (context) parallelism: the worker's ranks each hold
     # seq_len tokens of one (worker_size x seq_len)-token context;
@@ -77,29 +77,29 @@ class ParallelConfig(_Strict):
     seq_parallel: bool = False
     backend: str | None = None        # nccl | gloo (default: auto)
     elastic: bool = False
-    heartbeat_interval: float = 5.0
-    heartbeat_timeout: float = 30.0
+    heartbeat_interval: float = Field(5.0, gt=0)
+    heartbeat_timeout: float = Field(30.0, gt=0)
 
 
 class CheckpointConfig(_Strict):
-    interval: int = 0                 # outer steps between checkpoints; 0=off
+    interval: int = Field(0, ge=0)    # outer steps between checkpoints; 0=off
     path: str | None = None
     remote_path: str | None = None    # secondary copy (NFS/fuse mount)
-    keep: int = 3
+    keep: int = Field(3, ge=0)
     async_save: bool = True
     resume: str | None = None
 
 
 class MetricsConfig(_Strict):
-    log_interval: int = 10
+    log_interval: int = Field(10, gt=0)
     jsonl: bool = True
     wandb: bool = False
-    torch_profiler_steps: int = 0   # trace steps [3, 3+N) to <run>/trace.json
+    torch_profiler_steps: int = Field(0, ge=0)  # trace steps [3, 3+N) to <run>/trace.json
 
 
 class TrainConfig(_Strict):
     run_name: str = "run"
-    steps: int = 100
+    steps: int = Field(100, ge=0)
     seed: int = 1234
     device: str | None = None         # cuda | cpu (default: auto)
     model: ModelConfig = Field(default_factory=ModelConfig)

@@ -106,3 +106,25 @@ def test_fp8_flag_inert_on_cpu(tmp_path):
     loss = tr.train_step()
     assert float(loss) > 0
     tr.close()
+
+
+def test_numeric_bounds_rejected(tmp_path):
+    """Out-of-range numerics fail at load time with a friendly ConfigError
+    (H=0 or log_interval=0 would otherwise surface as ZeroDivisionError
+    deep inside the step loop)."""
+    import pytest
+
+    from prime_amd.utils.config import ConfigError, load_config
+
+    for body in (
+        "[diloco]\nH = 0\n",
+        "[metrics]\nlog_interval = 0\n",
+        "[data]\nmicro_batch_size = 0\n",
+        "[model]\nseq_len = -1\n",
+        "[parallel]\nworker_size = 0\n",
+        "steps = -5\n",
+    ):
+        f = tmp_path / "bad.toml"
+        f.write_text('run_name = "x"\n' + body)
+        with pytest.raises(ConfigError):
+            load_config(str(f))
