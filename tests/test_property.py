@@ -108,3 +108,44 @@ def test_env_file_roundtrip(key, value):
         f.write_text(f"{key}={value}\n")
         env = parse_env_file(f)
         assert env[key] == value.strip()
+
+
+@settings(deadline=None, max_examples=25)
+@given(st.integers(2, 5), st.integers(1, 4), st.integers(1, 3),
+       st.integers(0, 3))
+def test_token_file_shards_disjoint_and_covering(n_shards, b, files, batches):
+    """For any (shard count, micro-batch, file layout): the global window
+    indices drawn by all shards in one step are pairwise disjoint, and
+    with shuffle every window of the epoch is visited exactly once per
+    epoch across shards (no duplicated or dropped data)."""
+    import tempfile
+    from pathlib import Path
+
+    from prime_amd.data.loader import DataConfig, TokenFileDataset
+
+    seq = 8
+    with tempfile.TemporaryDirectory() as td:
+        root = Path(td)
+        for i in range(files):
+            toks = (np.arange(seq * 5 + 1) % 250).astype(np.uint16)
+            toks.tofile(root / f"part_{i}.bin")
+        cfg = DataConfig(kind="token_file", path=str(root), seq_len=seq,
+                         micro_batch_size=b, shuffle=True, seed=7)
+        loaders = [TokenFileDataset(cfg, 250, shard=s, n_shards=n_shards)
+                   for s in range(n_shards)]
+        windows = loaders[0].windows
+        seen: list[int] = []
+        total_draws = (batches + 1) * n_shards * b
+        for step in range(batches + 1):
+            for ld in loaders:
+                gi0 = ld.batch_idx * n_shards * b + ld.shard * b
+                ws = [ld._window(gi0 + i) for i in range(b)]
+                ld.next_batch(torch.device("cpu"))
+                seen.extend(ws)
+        # one full epoch's worth of draws must hit each window once
+        per_epoch = seen[: (len(seen) // windows) * windows]
+        if per_epoch:
+            for e in range(len(per_epoch) // windows):
+                chunk = per_epoch[e * windows : (e + 1) * windows]
+                assert sorted(chunk) == list(range(windows)), (n_shards, b)
+        assert len(seen) == total_draws
